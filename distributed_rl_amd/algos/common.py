@@ -1,0 +1,115 @@
+"""Shared learner scaffolding: device/dtype handling, telemetry cadence,
+checkpoint/resume, weight publishing.
+
+Telemetry parity: TB scalars + console block every 500
+learner steps (APE_X/Learner.py:219-262), checkpoint ``weight.pth`` under
+``weight/<ALG>/<run>/`` (APE_X/Learner.py:256-262). Unlike the reference
+(write-only checkpoints) we also persist optimizer + step for
+resume.
+"""
+
+from __future__ import annotations
+
+import os
+import time
+from typing import Any, Dict, Optional
+
+import torch
+
+from ..config import Config
+from ..models import BaseAgent, get_optim
+
+
+class LearnerBase:
+    LOG_EVERY = 500
+    CKPT_EVERY = 500
+
+    def __init__(self, cfg: Config, device: Optional[str] = None, rank: int = 0,
+                 world_size: int = 1, run_root: str = ".",
+                 run_name: Optional[str] = None, enable_tb: bool = True):
+        self.cfg = cfg
+        self.rank = rank
+        self.world_size = world_size
+        if device is None:
+            device = cfg.learner_device if torch.cuda.is_available() else "cpu"
+        self.device = torch.device(device)
+        self.autocast_dtype = (
+            torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        )
+        self.step_count = 0
+        self.run_name = run_name or cfg.run_name()
+        self.run_root = run_root
+        self._writer = None
+        self._enable_tb = enable_tb and rank == 0
+        self._timing: Dict[str, float] = {}
+        self._t_block = time.perf_counter()
+
+    # -- model helpers ----------------------------------------------------
+    def build_model(self) -> BaseAgent:
+        return BaseAgent(self.cfg.model_info).to(self.device)
+
+    def build_optim(self, model) -> torch.optim.Optimizer:
+        return get_optim(self.cfg.optim_info, model)
+
+    # -- telemetry --------------------------------------------------------
+    @property
+    def writer(self):
+        if self._writer is None and self._enable_tb:
+            from torch.utils.tensorboard import SummaryWriter
+
+            path = self.cfg.log_dir(self.run_root, self.run_name)
+            os.makedirs(path, exist_ok=True)
+            self._writer = SummaryWriter(path)
+            from ..utils import writeTrainInfo
+
+            self._writer.add_text("configuration", writeTrainInfo(self.cfg.raw).info, 0)
+        return self._writer
+
+    def log_scalar(self, tag: str, value: float, step: Optional[int] = None):
+        w = self.writer
+        if w is not None:
+            w.add_scalar(tag, value, self.step_count if step is None else step)
+
+    def time_block(self, key: str, dt: float):
+        self._timing[key] = self._timing.get(key, 0.0) + dt
+
+    def flush_timing(self) -> Dict[str, float]:
+        out = dict(self._timing)
+        out["wall"] = time.perf_counter() - self._t_block
+        self._timing = {}
+        self._t_block = time.perf_counter()
+        return out
+
+    # -- checkpoint / resume ---------------------------------------------
+    def checkpoint_dir(self) -> str:
+        d = self.cfg.weight_dir(self.run_root, self.run_name)
+        os.makedirs(d, exist_ok=True)
+        return d
+
+    def state_for_checkpoint(self) -> Dict[str, Any]:
+        raise NotImplementedError
+
+    def load_from_checkpoint(self, state: Dict[str, Any]) -> None:
+        raise NotImplementedError
+
+    def save_checkpoint(self) -> str:
+        if self.rank != 0:
+            return ""
+        d = self.checkpoint_dir()
+        state = self.state_for_checkpoint()
+        # reference-format model-only file (contract: weight.pth)
+        torch.save(state["model"], os.path.join(d, "weight.pth"))
+        torch.save(state, os.path.join(d, "resume.pt"))
+        return os.path.join(d, "weight.pth")
+
+    def resume(self, path: str) -> None:
+        """Load either a reference-style weight.pth (model only) or a full
+        resume.pt."""
+        state = torch.load(path, map_location=self.device, weights_only=False)
+        if isinstance(state, dict) and "model" in state and "step" in state:
+            self.load_from_checkpoint(state)
+        else:
+            self.load_model_only(state)
+
+    def load_model_only(self, sd) -> None:
+        raise NotImplementedError

@@ -1,0 +1,197 @@
+"""Op dispatch: HIP extension on device tensors, torch_ref on CPU.
+
+Policy (contract with the round driver): on a GPU box the hand-written HIP
+path MUST be the one that runs — if a device tensor reaches an op and the
+gfx950 extension is missing, we raise instead of silently falling back to
+eager PyTorch. Set DRL_ALLOW_EAGER=1 to override (debug only).
+"""
+
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+import torch
+
+from . import torch_ref  # noqa: F401
+
+_EXT = None
+_EXT_TRIED = False
+
+
+def hip_ext(required: bool = True):
+    """Return the loaded _drl_hip extension module (or None)."""
+    global _EXT, _EXT_TRIED
+    if _EXT is None and not _EXT_TRIED:
+        _EXT_TRIED = True
+        from . import build as _build
+
+        _EXT = _build.load_prebuilt()
+        if _EXT is None and torch.cuda.is_available():
+            # On a GPU box with no prebuilt .so: try building once (hipcc is
+            # present in the image), then give up loudly.
+            try:
+                _EXT = _build.build()
+            except Exception as e:  # pragma: no cover
+                if required:
+                    raise RuntimeError(
+                        "gfx950 HIP extension missing and build failed: %s" % e
+                    )
+    if _EXT is None and required and not _allow_eager():
+        raise RuntimeError(
+            "device tensor reached a fused op but the _drl_hip extension is "
+            "not built; run `python -m distributed_rl_amd.ops.build` "
+            "(or set DRL_ALLOW_EAGER=1 to debug with eager torch)"
+        )
+    return _EXT
+
+
+def _allow_eager() -> bool:
+    return os.environ.get("DRL_ALLOW_EAGER", "0") == "1"
+
+
+def _use_hip(t: torch.Tensor) -> bool:
+    if not t.is_cuda:
+        return False
+    ext = hip_ext(required=not _allow_eager())
+    return ext is not None
+
+
+# ---------------------------------------------------------------------------
+# K1 — dequant
+# ---------------------------------------------------------------------------
+
+
+def dequant_frames(x_u8: torch.Tensor, dtype: torch.dtype = torch.bfloat16) -> torch.Tensor:
+    if _use_hip(x_u8):
+        out = torch.empty(x_u8.shape, dtype=dtype, device=x_u8.device)
+        hip_ext().dequant(x_u8.contiguous(), out)
+        return out
+    return torch_ref.dequant_frames(x_u8, dtype)
+
+
+# ---------------------------------------------------------------------------
+# K4 — fused n-step double-DQN loss as an autograd Function
+# ---------------------------------------------------------------------------
+
+
+class _DQNLossFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q_s, q_sp_on, q_sp_tg, actions, rewards, dones, weights,
+                gamma_n, alpha):
+        ext = hip_ext()
+        B, A = q_s.shape
+        dev = q_s.device
+        loss = torch.zeros((), dtype=torch.float32, device=dev)
+        prio = torch.empty(B, dtype=torch.float32, device=dev)
+        coef = torch.empty(B, dtype=torch.float32, device=dev)
+        ext.dqn_loss_fwd(
+            q_s.contiguous(), q_sp_on.contiguous(), q_sp_tg.contiguous(),
+            actions.contiguous(), rewards.contiguous(), dones.contiguous(),
+            weights.contiguous(), float(gamma_n), float(alpha),
+            loss.view(1), prio, coef,
+        )
+        ctx.save_for_backward(coef, actions)
+        ctx.shape = (B, A)
+        ctx.mark_non_differentiable(prio)
+        return loss, prio
+
+    @staticmethod
+    def backward(ctx, gout, _gprio):
+        coef, actions = ctx.saved_tensors
+        B, A = ctx.shape
+        grad_q = torch.empty(B, A, dtype=torch.float32, device=coef.device)
+        hip_ext().dqn_loss_bwd(coef, actions, gout.reshape(1).contiguous(), grad_q)
+        return grad_q, None, None, None, None, None, None, None, None
+
+
+def nstep_dqn_loss(q_s, q_sp_on, q_sp_tg, actions, rewards, dones, weights,
+                   gamma: float, n_step: int, alpha: float):
+    """Returns (loss, new_priorities). q_s must be fp32 (cast before call)."""
+    if _use_hip(q_s):
+        return _DQNLossFn.apply(
+            q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions.long(),
+            rewards.float(), dones.float(), weights.float(),
+            gamma ** n_step, alpha,
+        )
+    return torch_ref.nstep_dqn_loss(
+        q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions, rewards.float(),
+        dones.float(), weights.float(), gamma, n_step, alpha
+    )
+
+
+# ---------------------------------------------------------------------------
+# K6 / K7
+# ---------------------------------------------------------------------------
+
+
+def value_rescale(x: torch.Tensor, eps: float = 1e-3) -> torch.Tensor:
+    if _use_hip(x) and not x.requires_grad:
+        y = torch.empty_like(x, dtype=torch.float32)
+        hip_ext().value_rescale(x.float().contiguous(), y, eps)
+        return y
+    return torch_ref.value_rescale(x, eps)
+
+
+def inv_value_rescale(x: torch.Tensor, eps: float = 1e-3) -> torch.Tensor:
+    if _use_hip(x) and not x.requires_grad:
+        y = torch.empty_like(x, dtype=torch.float32)
+        hip_ext().inv_value_rescale(x.float().contiguous(), y, eps)
+        return y
+    return torch_ref.inv_value_rescale(x, eps)
+
+
+def sequence_priority(td_abs: torch.Tensor, alpha: float, eta: float = 0.9) -> torch.Tensor:
+    if _use_hip(td_abs):
+        T, B = td_abs.shape
+        out = torch.empty(B, dtype=torch.float32, device=td_abs.device)
+        hip_ext().seq_priority(td_abs.float().contiguous(), eta, alpha, out)
+        return out
+    return torch_ref.sequence_priority(td_abs, alpha, eta)
+
+
+# ---------------------------------------------------------------------------
+# K8 — V-trace
+# ---------------------------------------------------------------------------
+
+
+def vtrace(behavior_log_prob, target_log_prob, rewards, values, bootstrap_value,
+           not_done, gamma, rho_bar=1.0, c_bar=1.0, lam=1.0):
+    if _use_hip(rewards):
+        T, B = rewards.shape
+        dev = rewards.device
+        vs = torch.empty(T, B, dtype=torch.float32, device=dev)
+        pg_adv = torch.empty_like(vs)
+        rho_c = torch.empty_like(vs)
+        hip_ext().vtrace(
+            behavior_log_prob.float().contiguous(),
+            target_log_prob.float().contiguous(), rewards.float().contiguous(),
+            values.float().contiguous(), bootstrap_value.float().contiguous(),
+            not_done.float().contiguous(), gamma, rho_bar, c_bar, lam,
+            vs, pg_adv, rho_c,
+        )
+        return vs, pg_adv, rho_c
+    return torch_ref.vtrace(
+        behavior_log_prob, target_log_prob, rewards, values, bootstrap_value,
+        not_done, gamma, rho_bar, c_bar, lam
+    )
+
+
+# ---------------------------------------------------------------------------
+# K11 — fused grad clip over a flat buffer
+# ---------------------------------------------------------------------------
+
+
+def clip_flat_grad_(flat: torch.Tensor, max_norm: float,
+                    sqsum_buf: Optional[torch.Tensor] = None) -> None:
+    if _use_hip(flat):
+        if sqsum_buf is None:
+            sqsum_buf = torch.zeros(1, dtype=torch.float32, device=flat.device)
+        else:
+            sqsum_buf.zero_()
+        hip_ext().grad_clip(flat, max_norm, sqsum_buf)
+        return
+    norm = flat.norm(2)
+    scale = max_norm / (norm + 1e-6)
+    if norm > max_norm:
+        flat.mul_(scale)
