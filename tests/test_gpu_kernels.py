@@ -1,0 +1,174 @@
+"""HIP kernel numerics vs the plain-PyTorch fp32 references (torch_ref)."""
+
+import numpy as np
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+DEV = "cuda:0"
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from distributed_rl_amd.ops import hip_ext
+
+    return hip_ext(required=True)
+
+
+def test_dequant_bf16(ext):
+    from distributed_rl_amd import ops
+    from distributed_rl_amd.ops import torch_ref
+
+    x = torch.randint(0, 256, (33, 4, 84, 84), dtype=torch.uint8, device=DEV)
+    y = ops.dequant_frames(x, torch.bfloat16)
+    ref = torch_ref.dequant_frames(x.float().cpu()).to(torch.bfloat16)
+    assert y.dtype == torch.bfloat16
+    assert torch.allclose(y.float().cpu(), ref.float(), atol=1 / 255)
+    y32 = ops.dequant_frames(x, torch.float32)
+    assert torch.allclose(y32.cpu(), torch_ref.dequant_frames(x.cpu().float()) / 1.0,
+                          atol=1e-7)
+
+
+def test_sumtree_total_and_update(ext):
+    from distributed_rl_amd.replay.gpu_per import HipSumTreePER
+
+    schema = {"x": ((), torch.float32)}
+    per = HipSumTreePER(1000, schema, DEV)
+    prios = torch.rand(1000, device=DEV) + 0.01
+    per.push({"x": torch.arange(1000.0, device=DEV)}, prios)
+    assert abs(per.total_priority - float(prios.sum())) < 1e-2
+    # update a subset (with duplicate indices — atomicExch path)
+    idx = torch.tensor([3, 3, 500, 999], device=DEV)
+    newp = torch.tensor([5.0, 7.0, 1.0, 2.0], device=DEV)
+    per.update(idx, newp)
+    torch.cuda.synchronize()
+    expect = prios.clone()
+    expect[3] = 7.0  # last write wins
+    expect[500] = 1.0
+    expect[999] = 2.0
+    assert abs(per.total_priority - float(expect.sum())) < 1e-2
+    leaf = per.tree[per.P + 3].item()
+    assert leaf == 7.0
+
+
+def test_sumtree_sampling_distribution(ext):
+    from distributed_rl_amd.replay.gpu_per import HipSumTreePER
+
+    schema = {"x": ((), torch.float32)}
+    per = HipSumTreePER(4, schema, DEV)
+    prios = torch.tensor([1.0, 2.0, 3.0, 4.0], device=DEV)
+    per.push({"x": torch.arange(4.0, device=DEV)}, prios)
+    counts = torch.zeros(4, device=DEV)
+    for _ in range(40):
+        _, idx, _ = per.sample(1000, beta=0.4)
+        counts += torch.bincount(idx, minlength=4).float()
+    freq = (counts / counts.sum()).cpu()
+    expect = (prios / prios.sum()).cpu()
+    assert torch.allclose(freq, expect, atol=0.02), (freq, expect)
+
+
+def test_sumtree_is_weights_vs_oracle(ext):
+    from distributed_rl_amd.replay.gpu_per import HipSumTreePER
+    from distributed_rl_amd.ops import torch_ref
+
+    schema = {"x": ((), torch.float32)}
+    per = HipSumTreePER(64, schema, DEV)
+    prios = torch.rand(64, device=DEV) + 0.1
+    per.push({"x": torch.zeros(64, device=DEV)}, prios)
+    beta = 0.4
+    _, idx, w = per.sample(256, beta=beta)
+    torch.cuda.synchronize()
+    probs = prios[idx] / prios.sum()
+    expect_max_w = (1.0 / (64 * (prios.min() / prios.sum()))) ** beta
+    expect = (1.0 / (64 * probs)) ** beta / expect_max_w
+    assert torch.allclose(w, expect, rtol=1e-3), (w[:5], expect[:5])
+
+
+def test_dqn_loss_matches_ref(ext):
+    from distributed_rl_amd import ops
+    from distributed_rl_amd.ops import torch_ref
+
+    torch.manual_seed(0)
+    B, A, n = 64, 6, 3
+    gamma, alpha = 0.99, 0.6
+    q_s = torch.randn(B, A, device=DEV, requires_grad=True)
+    q_sp_on = torch.randn(B, A, device=DEV)
+    q_sp_tg = torch.randn(B, A, device=DEV)
+    act = torch.randint(0, A, (B,), device=DEV)
+    rew = torch.randn(B, device=DEV)
+    done = (torch.rand(B, device=DEV) < 0.2).float()
+    w = torch.rand(B, device=DEV)
+
+    loss, prio = ops.nstep_dqn_loss(q_s, q_sp_on, q_sp_tg, act, rew, done, w,
+                                    gamma, n, alpha)
+    loss.backward()
+    g_hip = q_s.grad.clone()
+
+    q_s2 = q_s.detach().cpu().requires_grad_(True)
+    loss_ref, prio_ref = torch_ref.nstep_dqn_loss(
+        q_s2, q_sp_on.cpu(), q_sp_tg.cpu(), act.cpu(), rew.cpu(), done.cpu(),
+        w.cpu(), gamma, n, alpha)
+    loss_ref.backward()
+    assert abs(loss.item() - loss_ref.item()) < 1e-5
+    assert torch.allclose(prio.cpu(), prio_ref, atol=1e-5)
+    assert torch.allclose(g_hip.cpu(), q_s2.grad, atol=1e-6)
+
+
+def test_vtrace_matches_ref(ext):
+    from distributed_rl_amd import ops
+    from distributed_rl_amd.ops import torch_ref
+
+    torch.manual_seed(1)
+    T, B = 20, 32
+    blogp = -torch.rand(T, B, device=DEV)
+    tlogp = -torch.rand(T, B, device=DEV)
+    rew = torch.randn(T, B, device=DEV)
+    val = torch.randn(T, B, device=DEV)
+    boot = torch.randn(B, device=DEV)
+    nd = (torch.rand(B, device=DEV) < 0.8).float()
+    vs, pg, rho = ops.vtrace(blogp, tlogp, rew, val, boot, nd, 0.99)
+    vs_r, pg_r, rho_r = torch_ref.vtrace(
+        blogp.cpu(), tlogp.cpu(), rew.cpu(), val.cpu(), boot.cpu(), nd.cpu(), 0.99)
+    assert torch.allclose(vs.cpu(), vs_r, atol=1e-4)
+    assert torch.allclose(pg.cpu(), pg_r, atol=1e-4)
+    assert torch.allclose(rho.cpu(), rho_r, atol=1e-5)
+
+
+def test_value_rescale_matches_ref(ext):
+    from distributed_rl_amd import ops
+    from distributed_rl_amd.ops import torch_ref
+
+    x = torch.linspace(-40, 40, 4096, device=DEV)
+    y = ops.value_rescale(x)
+    assert torch.allclose(y.cpu(), torch_ref.value_rescale(x.cpu()), atol=1e-5)
+    z = ops.inv_value_rescale(y)
+    assert torch.allclose(z.cpu(), x.cpu(), atol=1e-2)
+
+
+def test_seq_priority_matches_ref(ext):
+    from distributed_rl_amd import ops
+    from distributed_rl_amd.ops import torch_ref
+
+    td = torch.rand(80, 32, device=DEV)
+    p = ops.sequence_priority(td, alpha=0.9)
+    p_ref = torch_ref.sequence_priority(td.cpu(), alpha=0.9)
+    assert torch.allclose(p.cpu(), p_ref, atol=1e-5)
+
+
+def test_grad_clip_matches_torch(ext):
+    from distributed_rl_amd import ops
+
+    x = torch.randn(100_000, device=DEV) * 3
+    ref = x.clone()
+    norm = ref.norm(2)
+    max_norm = 40.0
+    if norm > max_norm:
+        ref.mul_(max_norm / (norm + 1e-6))
+    ops.clip_flat_grad_(x, max_norm)
+    assert torch.allclose(x, ref, rtol=1e-5)
+    # below the clip threshold: untouched
+    y = torch.randn(1000, device=DEV) * 0.001
+    y0 = y.clone()
+    ops.clip_flat_grad_(y, max_norm)
+    assert torch.equal(y, y0)
