@@ -118,10 +118,17 @@ class ApexLearner(LearnerBase):
         dones = data["done"].to(self.device)
         weights = weights.to(self.device)
 
-        q_s = self.model.forward([s])[0]
-        with torch.no_grad():
-            q_sp_on = self.model.forward([sp])[0]
-            q_sp_tg = self.target.forward([sp])[0]
+        if self.device.type == "cuda":
+            with torch.autocast("cuda", dtype=torch.bfloat16):
+                q_s = self.model.forward([s])[0]
+                with torch.no_grad():
+                    q_sp_on = self.model.forward([sp])[0]
+                    q_sp_tg = self.target.forward([sp])[0]
+        else:
+            q_s = self.model.forward([s])[0]
+            with torch.no_grad():
+                q_sp_on = self.model.forward([sp])[0]
+                q_sp_tg = self.target.forward([sp])[0]
 
         loss, prio = ops.nstep_dqn_loss(
             q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions, rewards,
