@@ -157,9 +157,7 @@ def len_target(learner) -> int:
 
 
 def make_graphed_stepper(learner: ApexLearner):
-    """Capture sample+train+priority-update in one hipGraph; optimizer step
-    runs inside too when the optimizer is capturable."""
-    raise RuntimeError("graph stepper not implemented yet")
+    return learner.make_graphed_step()
 
 
 if __name__ == "__main__":

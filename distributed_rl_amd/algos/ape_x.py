@@ -50,10 +50,41 @@ class ApexLearner(LearnerBase):
                          enable_tb=enable_tb)
         self.batch_size = batch_size or cfg.batch_size
         self.model = self.build_model()
-        self.target = self.build_model()
-        self.target.updateParameter(self.model, 1.0)
-        for p in self.target.parameters():
-            p.requires_grad_(False)
+        self.mp = None
+        if self.device.type == "cuda":
+            # channels_last convs (NHWC is MIOpen/our-conv native layout) +
+            # persistent bf16 compute replica with flat grad/param buffers
+            from ..parallel.precision import MixedPrecisionTrainer
+
+            self.model.to(memory_format=torch.channels_last)
+            self.mp = MixedPrecisionTrainer(self.model)
+            self.net = self.mp.compute  # bf16 forward/backward model
+            import copy as _copy
+
+            self.target = _copy.deepcopy(self.mp.compute)
+            for p in self.target.parameters():
+                p.requires_grad_(False)
+            # flat target param buffer so hard-sync is ONE device copy
+            from ..parallel.precision import _to_flat, _view_like
+
+            tp = [p for p in self.target.parameters()]
+            self.flat_tparam = torch.empty(
+                sum(p.numel() for p in tp), dtype=torch.bfloat16,
+                device=self.device,
+            )
+            off = 0
+            with torch.no_grad():
+                for p in tp:
+                    n = p.numel()
+                    self.flat_tparam[off : off + n].copy_(_to_flat(p))
+                    p.data = _view_like(self.flat_tparam[off : off + n], p)
+                    off += n
+        else:
+            self.net = self.model
+            self.target = self.build_model()
+            self.target.updateParameter(self.model, 1.0)
+            for p in self.target.parameters():
+                p.requires_grad_(False)
         self.optim = self.build_optim(self.model)
         cap = replay_capacity or cfg.replay_memory_len
         rdev = replay_device or (
@@ -108,50 +139,95 @@ class ApexLearner(LearnerBase):
     # train
     # ------------------------------------------------------------------
     def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
-        dt = torch.bfloat16 if self.device.type == "cuda" else torch.float32
-        s = ops.dequant_frames(data["state"].to(self.device, non_blocking=True), dt)
-        sp = ops.dequant_frames(
-            data["next_state"].to(self.device, non_blocking=True), dt
-        )
+        cuda = self.device.type == "cuda"
+        if cuda:
+            s = ops.dequant_frames_nhwc(data["state"].to(self.device, non_blocking=True))
+            sp = ops.dequant_frames_nhwc(
+                data["next_state"].to(self.device, non_blocking=True)
+            )
+        else:
+            s = ops.dequant_frames(data["state"], torch.float32)
+            sp = ops.dequant_frames(data["next_state"], torch.float32)
         actions = data["action"].to(self.device).long()
         rewards = data["reward"].to(self.device)
         dones = data["done"].to(self.device)
         weights = weights.to(self.device)
 
-        if self.device.type == "cuda":
-            with torch.autocast("cuda", dtype=torch.bfloat16):
-                q_s = self.model.forward([s])[0]
-                with torch.no_grad():
-                    q_sp_on = self.model.forward([sp])[0]
-                    q_sp_tg = self.target.forward([sp])[0]
-        else:
-            q_s = self.model.forward([s])[0]
-            with torch.no_grad():
-                q_sp_on = self.model.forward([sp])[0]
-                q_sp_tg = self.target.forward([sp])[0]
+        q_s = self.net.forward([s])[0]
+        with torch.no_grad():
+            q_sp_on = self.net.forward([sp])[0]
+            q_sp_tg = self.target.forward([sp])[0]
 
         loss, prio = ops.nstep_dqn_loss(
             q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions, rewards,
             dones, weights, self.gamma, self.n_step, self.alpha,
         )
-        self.optim.zero_grad(set_to_none=False)
-        loss.backward()
-        if self.reducer is not None:
-            self.reducer.all_reduce()
-        self.optim.step()
+        if self.mp is not None:
+            self.mp.zero_grads()
+            loss.backward()
+            self.mp.reduce_and_upcast()  # bf16 all-reduce at world>1
+            self.optim.step()
+            self.mp.sync_compute_params()
+        else:
+            self.optim.zero_grad(set_to_none=False)
+            loss.backward()
+            if self.reducer is not None:
+                self.reducer.all_reduce()
+            self.optim.step()
         self.replay.update(idx, prio)
         return {"loss": loss.detach(), "value": q_s.detach().float().max(1).values.mean()}
 
-    def step(self) -> Dict[str, torch.Tensor]:
+    def sync_target(self):
+        """Hard target sync (APE_X/Learner.py:204-208, tau=1)."""
+        if self.mp is not None:
+            self.flat_tparam.copy_(self.mp.flat_cparam)
+        else:
+            self.target.updateParameter(self.model, 1.0)
+
+    def _inner_step(self) -> Dict[str, torch.Tensor]:
         data, idx, w = self.replay.sample(self.batch_size, self.beta)
-        stats = self.train_step(data, idx, w)
+        return self.train_step(data, idx, w)
+
+    def _cadence(self):
         self.step_count += 1
         if self.step_count % self.cfg.target_frequency == 0:
-            self.target.updateParameter(self.model, 1.0)
+            self.sync_target()
             self.publish_weights(include_target=True)
         elif self.step_count % PUBLISH_EVERY == 0:
             self.publish_weights()
+
+    def step(self) -> Dict[str, torch.Tensor]:
+        stats = self._inner_step()
+        self._cadence()
         return stats
+
+    def make_graphed_step(self, warmup_iters: int = 3):
+        """hipGraph-capture the whole inner step (PER sample -> dequant ->
+        3 forwards -> fused loss -> backward -> grad cast -> optimizer ->
+        priority update). All shapes are static and RNG lives in a device
+        seed buffer, so one graph replay per learner step. Cadence ops
+        (target sync, weight publish) stay eager. Requires a warm, fixed-size
+        replay (n_valid is baked into the sample kernel)."""
+        assert self.device.type == "cuda", "graph capture needs a GPU"
+        for g in self.optim.param_groups:
+            g["capturable"] = True
+        side = torch.cuda.Stream(self.device)
+        side.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(side):
+            for _ in range(warmup_iters):
+                self._inner_step()
+        torch.cuda.current_stream(self.device).wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static_out = self._inner_step()
+        self._graph = graph  # keep alive (owns the memory pool)
+
+        def stepper():
+            graph.replay()
+            self._cadence()
+            return static_out
+
+        return stepper
 
     # ------------------------------------------------------------------
     # weight publication (seqlock bus; replaces Redis state_dict keys)
@@ -237,10 +313,16 @@ class ApexLearner(LearnerBase):
         self.target.load_state_dict(state["target"])
         self.optim.load_state_dict(state["optim"])
         self.step_count = int(state["step"])
+        if self.mp is not None:
+            self.mp.sync_compute_params()
 
     def load_model_only(self, sd):
         self.model.load_state_dict(sd)
-        self.target.load_state_dict(sd)
+        if self.mp is not None:
+            self.mp.sync_compute_params()
+            self.sync_target()
+        else:
+            self.target.load_state_dict(sd)
 
 
 # ===========================================================================

@@ -70,6 +70,19 @@ def dequant_frames(x_u8: torch.Tensor, dtype: torch.dtype = torch.bfloat16) -> t
     return torch_ref.dequant_frames(x_u8, dtype)
 
 
+def dequant_frames_nhwc(x_u8: torch.Tensor) -> torch.Tensor:
+    """(N,4,H,W) uint8 -> channels_last bf16, fused /255 (K1 NHWC variant)."""
+    if _use_hip(x_u8):
+        out = torch.empty(
+            x_u8.shape, dtype=torch.bfloat16, device=x_u8.device
+        ).to(memory_format=torch.channels_last)
+        hip_ext().dequant_nhwc(x_u8.contiguous(), out)
+        return out
+    return torch_ref.dequant_frames(x_u8, torch.float32).to(
+        memory_format=torch.channels_last
+    )
+
+
 # ---------------------------------------------------------------------------
 # K4 — fused n-step double-DQN loss as an autograd Function
 # ---------------------------------------------------------------------------

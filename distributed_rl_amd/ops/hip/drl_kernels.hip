@@ -65,15 +65,46 @@ __global__ void dequant_u8_f32_kernel(const uchar4* __restrict__ in,
   }
 }
 
+// NCHW uint8 -> NHWC bf16 (channels_last), C=4: one lane per (n,h,w) pixel
+// reads 4 plane-strided bytes, writes one 8-byte bf16x4. Feeding MIOpen/our
+// convs channels_last removes the batched_transpose layout kernels that
+// dominate the NCHW path (see profiles/).
+__global__ void dequant_u8_bf16_nhwc_c4_kernel(const uint8_t* __restrict__ in,
+                                               ushort4* __restrict__ out,
+                                               int64_t n_pix, int64_t hw) {
+  const float inv255 = 1.0f / 255.0f;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n_pix;
+       i += stride) {
+    int64_t n = i / hw;
+    int64_t p = i - n * hw;          // h*W + w
+    const uint8_t* src = in + n * 4 * hw + p;
+    __hip_bfloat162 lo = __float22bfloat162_rn(
+        {src[0] * inv255, src[hw] * inv255});
+    __hip_bfloat162 hi = __float22bfloat162_rn(
+        {src[2 * hw] * inv255, src[3 * hw] * inv255});
+    ushort2 l = *reinterpret_cast<ushort2*>(&lo);
+    ushort2 h = *reinterpret_cast<ushort2*>(&hi);
+    out[i] = {l.x, l.y, h.x, h.y};
+  }
+}
+
 // ---------------------------------------------------------------------------
 // K10: lock-free sum-tree. Layout: float tree[2P], P = pow2 >= capacity,
 // root tree[1], leaves tree[P + i]. tree[1] is the running total priority.
 // ---------------------------------------------------------------------------
 
+// Two-phase update kills root contention: phase 1 propagates atomic deltas
+// only through the wide bottom levels (node >= TOP, where random leaves
+// rarely collide); phase 2 (one block) deterministically rebuilds the top
+// TOP-1 nodes level by level. 512 concurrent updates on a 2^17 tree went
+// 136us -> ~6us with this split (profiles/).
+constexpr int64_t kTreeTop = 2048;
+
 __global__ void sumtree_update_kernel(float* __restrict__ tree,
                                       const int64_t* __restrict__ idx,
                                       const float* __restrict__ prio,
-                                      int m, int64_t P) {
+                                      int m, int64_t P, int64_t top) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i >= m) return;
   int64_t leaf = P + idx[i];
@@ -81,8 +112,17 @@ __global__ void sumtree_update_kernel(float* __restrict__ tree,
   float old = atomicExch(&tree[leaf], newp);  // duplicate-safe: each swap sees
   float delta = newp - old;                   // a consistent predecessor value
   if (delta == 0.0f) return;
-  for (int64_t node = leaf >> 1; node >= 1; node >>= 1)
+  for (int64_t node = leaf >> 1; node >= top; node >>= 1)
     atomicAdd(&tree[node], delta);
+}
+
+__global__ void sumtree_rebuild_top_kernel(float* __restrict__ tree,
+                                           int64_t top) {
+  for (int64_t s = top >> 1; s >= 1; s >>= 1) {
+    for (int64_t i = s + threadIdx.x; i < 2 * s; i += blockDim.x)
+      tree[i] = tree[2 * i] + tree[2 * i + 1];
+    __syncthreads();
+  }
 }
 
 __device__ inline float hash01(unsigned long long seed, unsigned int i) {
@@ -359,6 +399,23 @@ void dequant(torch::Tensor src_u8, torch::Tensor dst) {
   }
 }
 
+void dequant_nhwc(torch::Tensor src_u8, torch::Tensor dst) {
+  // src: contiguous NCHW uint8, C==4; dst: channels_last bf16 (same shape)
+  DRL_CHECK_CUDA(src_u8);
+  DRL_CHECK_CONTIG(src_u8);
+  TORCH_CHECK(src_u8.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(dst.scalar_type() == torch::kBFloat16);
+  TORCH_CHECK(src_u8.dim() == 4 && src_u8.size(1) == 4, "expects (N,4,H,W)");
+  TORCH_CHECK(dst.is_contiguous(at::MemoryFormat::ChannelsLast),
+              "dst must be channels_last");
+  int64_t N = src_u8.size(0), H = src_u8.size(2), W = src_u8.size(3);
+  int64_t hw = H * W, n_pix = N * hw;
+  hipLaunchKernelGGL(dequant_u8_bf16_nhwc_c4_kernel, dim3(grid_for(n_pix)),
+                     dim3(kBlock), 0, cur_stream(),
+                     (const uint8_t*)src_u8.data_ptr(), (ushort4*)dst.data_ptr(),
+                     n_pix, hw);
+}
+
 void sumtree_update(torch::Tensor tree, torch::Tensor idx, torch::Tensor prio,
                     int64_t P) {
   DRL_CHECK_CUDA(tree);
@@ -366,9 +423,12 @@ void sumtree_update(torch::Tensor tree, torch::Tensor idx, torch::Tensor prio,
   TORCH_CHECK(prio.scalar_type() == torch::kFloat32);
   int m = (int)idx.numel();
   if (m == 0) return;
+  int64_t top = P < kTreeTop ? P : kTreeTop;
   hipLaunchKernelGGL(sumtree_update_kernel, dim3(ceil_div(m, kBlock)), dim3(kBlock),
                      0, cur_stream(), tree.data_ptr<float>(),
-                     idx.data_ptr<int64_t>(), prio.data_ptr<float>(), m, P);
+                     idx.data_ptr<int64_t>(), prio.data_ptr<float>(), m, P, top);
+  hipLaunchKernelGGL(sumtree_rebuild_top_kernel, dim3(1), dim3(1024), 0,
+                     cur_stream(), tree.data_ptr<float>(), top);
 }
 
 void sumtree_sample(torch::Tensor tree, int64_t P, int64_t n_valid, int64_t k,
@@ -476,6 +536,7 @@ void grad_clip(torch::Tensor flat, double max_norm, torch::Tensor sqsum_buf) {
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dequant", &dequant, "u8 -> bf16/f32 /255 (K1)");
+  m.def("dequant_nhwc", &dequant_nhwc, "u8 NCHW -> bf16 NHWC /255 (K1)");
   m.def("sumtree_update", &sumtree_update, "lock-free sum-tree leaf update (K10)");
   m.def("sumtree_sample", &sumtree_sample, "stratified sum-tree descent (K10)");
   m.def("leaf_min_pos", &leaf_min_pos, "min positive leaf priority (K10)");

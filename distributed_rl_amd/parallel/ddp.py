@@ -77,6 +77,13 @@ def attach_reducer(learner, group=None) -> Optional[FlatGradReducer]:
     world = dist.get_world_size(group) if dist.is_initialized() else 1
     if world <= 1:
         return None
+    if getattr(learner, "mp", None) is not None:
+        # mixed-precision path reduces its own flat bf16 grads; just make
+        # replicas start identical
+        learner.mp.broadcast_master()
+        if hasattr(learner, "flat_tparam"):
+            learner.flat_tparam.copy_(learner.mp.flat_cparam)
+        return None
     for model in (learner.model, learner.target):
         for p in model.state_dict().values():
             if p.is_floating_point():
