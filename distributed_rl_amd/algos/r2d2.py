@@ -1,0 +1,450 @@
+"""R2D2 — recurrent replay distributed DQN, MI355X-native.
+
+Behavior parity targets (/root/reference/R2D2/):
+  * actor: LSTM policy, per-step hidden state captured BEFORE each step
+    (Player.py:30-32,300-301), fixed 80-step sequences with 40-step overlap
+    (FIXED_TRAJECTORY=80, emit drops the first 40 on non-terminal emission,
+    Player.py:37-62), actor-side sequence priority = eta-mix of n-step
+    double-DQN |TD| over the whole window replayed with the stored h_0
+    (Player.py:147-215), weight pull every 400 steps (Player.py:321-322),
+  * learner: 20-step no-grad burn-in then detachCellState, training on the
+    remaining 60 steps (Learner.py:83-107); value rescaling h/h^-1 around
+    the n-step bootstrap (Learner.py:22-35,143-167, USE_RESCALING); sequence
+    priority (0.9*max + 0.1*mean)^alpha (Learner.py:175-181); Adam(1e-4,
+    eps 1e-3), grad clip 40, target sync 2500, publish every 25 steps
+    (Learner.py:289-293).
+
+DEFECT FIXED: the reference slices actions with
+``action[FIXED_TRAJECTORY - MEM:-1]`` (R2D2/Learner.py:111) which crashes
+with the shipped cfg; the correct training window is steps [MEM, T-1) and
+that is what we implement: 59 n-step targets per 80-step sequence, truncated
+n at the tail (n_t = min(UNROLL_STEP, T-1-t)), bootstrap masked by the
+sequence-terminal flag at the final state.
+"""
+
+from __future__ import annotations
+
+import time
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..config import Config
+from ..models import BaseAgent
+from ..replay import make_per, make_r2d2_schema
+from .common import LearnerBase
+
+PUBLISH_EVERY = 25  # R2D2/Learner.py:289-293
+ACTOR_PULL_EVERY = 400  # R2D2/Player.py:321-322
+ETA = 0.9  # priority mix (R2D2 paper / Learner.py:178)
+
+
+def nstep_recurrent_targets(q_online: torch.Tensor, q_target: torch.Tensor,
+                            actions: torch.Tensor, rewards: torch.Tensor,
+                            done: torch.Tensor, burn_in: int, n_step: int,
+                            gamma: float, use_rescaling: bool):
+    """Shared target math for learner loss and actor priorities.
+
+    q_online, q_target: (T, B, A) — full-sequence Q values (q_online may be
+    detached or differentiable; only rows [burn_in, T-1) contribute).
+    actions/rewards: (T, B); done: (B,) terminal-at-sequence-end flag.
+    Returns (td, q_taken, targets) each (T-1-burn_in, B) where
+    td = h(G_t) - Q(s_t, a_t) in rescaled space.
+    """
+    T, B, A = q_online.shape
+    h = ops.value_rescale if use_rescaling else (lambda x: x)
+    h_inv = ops.inv_value_rescale if use_rescaling else (lambda x: x)
+
+    with torch.no_grad():
+        a_star = q_online.detach().argmax(dim=2)  # (T, B)
+        q_boot = q_target.gather(2, a_star.unsqueeze(2)).squeeze(2)  # (T, B)
+        q_boot = h_inv(q_boot.float())
+        # bootstrap at the final state is masked when the sequence terminated
+        q_boot = torch.cat(
+            [q_boot[:-1], (q_boot[-1] * (1.0 - done)).unsqueeze(0)], dim=0
+        )
+        ts = torch.arange(burn_in, T - 1, device=q_online.device)
+        n_t = torch.minimum(
+            torch.full_like(ts, n_step), (T - 1) - ts
+        )  # truncated n at the tail
+        # discounted n-step return via gamma-weighted prefix sums:
+        #   sum_{i<n_t} g^i r_{t+i} = (P[t+n_t] - P[t]) / g^t,
+        #   P[t] = sum_{s<t} g^s r_s
+        dev = q_online.device
+        g_pow = gamma ** torch.arange(T + 1, device=dev, dtype=torch.float64)
+        wr = rewards.double() * g_pow[:T].unsqueeze(1)  # (T, B)
+        P = torch.zeros(T + 1, B, dtype=torch.float64, device=dev)
+        P[1:] = torch.cumsum(wr, dim=0)
+        tn = ts + n_t  # (W,)
+        ret = (P[tn] - P[ts]) / g_pow[ts].unsqueeze(1)  # (W, B)
+        G = ret.float() + (gamma ** n_t.float()).unsqueeze(1) * q_boot[tn]
+        targets = h(G)
+    q_taken = q_online[burn_in : T - 1].gather(
+        2, actions[burn_in : T - 1].long().unsqueeze(2)
+    ).squeeze(2).float()
+    td = targets - q_taken
+    return td, q_taken, targets
+
+
+class R2D2Learner(LearnerBase):
+    ALG = "R2D2"
+
+    def __init__(self, cfg: Config, device: Optional[str] = None, rank: int = 0,
+                 world_size: int = 1, transport=None,
+                 batch_size: Optional[int] = None,
+                 replay_capacity: Optional[int] = None,
+                 enable_tb: bool = True, run_root: str = "."):
+        super().__init__(cfg, device, rank, world_size, run_root=run_root,
+                         enable_tb=enable_tb)
+        self.batch_size = batch_size or cfg.batch_size
+        self.T = cfg.fixed_trajectory
+        self.burn_in = cfg.burn_in
+        self.n_step = cfg.unroll_step
+        self.gamma = cfg.gamma
+        self.alpha = cfg.alpha
+        self.beta = cfg.beta
+        self.use_rescaling = cfg.use_rescaling
+        self.model = self.build_model()
+        self.net = self.model  # LSTM path runs fp32 (K5 HIP LSTM is future work)
+        self.target = self.build_model()
+        self.target.updateParameter(self.model, 1.0)
+        for p in self.target.parameters():
+            p.requires_grad_(False)
+        self.optim = self.build_optim(self.model)
+        self.mp = None
+        cap = replay_capacity or cfg.replay_memory_len
+        rdev = str(self.device) if self.device.type == "cuda" else "cpu"
+        self.replay = make_per(
+            cap, make_r2d2_schema(self.T, hidden=self._hidden_size()), device=rdev
+        )
+        self.transport = transport
+        self.reducer = None
+        self._sqsum_buf = (
+            torch.zeros(1, device=self.device) if self.device.type == "cuda" else None
+        )
+
+    def _hidden_size(self) -> int:
+        for node in self.cfg.model_info.values():
+            if str(node.get("netCat", "")).upper() == "LSTMNET":
+                return int(node["hiddenSize"])
+        raise ValueError("R2D2 cfg has no LSTMNET node")
+
+    # -- ingest -----------------------------------------------------------
+    def ingest(self) -> int:
+        if self.transport is None:
+            return 0
+        got = self.transport.drain()
+        if got is None:
+            return 0
+        cols_np, prio_np = got
+        cols = {k: torch.from_numpy(np.ascontiguousarray(v))
+                for k, v in cols_np.items()}
+        prio = torch.from_numpy(np.ascontiguousarray(prio_np))
+        if self.device.type == "cuda":
+            cols = {k: v.pin_memory().to(self.device, non_blocking=True)
+                    for k, v in cols.items()}
+            prio = prio.pin_memory().to(self.device, non_blocking=True)
+        self.replay.push(cols, prio)
+        return len(prio)
+
+    def push_sequences(self, cols, prio):
+        self.replay.push(cols, prio)
+
+    # -- forward helpers ---------------------------------------------------
+    def _seq_forward(self, net: BaseAgent, frames: torch.Tensor, h0, *,
+                     burn_in_split: bool):
+        """frames: (T, B, 4,84,84) float; returns (T or T-burn_in, B, A) Q.
+
+        burn_in_split=True runs [0, burn_in) under no_grad then detaches the
+        cell state (R2D2/Learner.py:99-104) and returns only the training
+        window's Q values; False runs the whole sequence (target net)."""
+        T, B = frames.shape[:2]
+        net.setCellState(h0)
+        if burn_in_split and self.burn_in > 0:
+            m = self.burn_in
+            with torch.no_grad():
+                net.forward([
+                    frames[:m].reshape(m * B, *frames.shape[2:]),
+                    torch.tensor([m, B, -1]),
+                ])
+            net.detachCellState()
+            q = net.forward([
+                frames[m:].reshape((T - m) * B, *frames.shape[2:]),
+                torch.tensor([T - m, B, -1]),
+            ])[0]
+            return q.view(T - m, B, -1)
+        with torch.no_grad():
+            q = net.forward([
+                frames.reshape(T * B, *frames.shape[2:]),
+                torch.tensor([T, B, -1]),
+            ])[0]
+        return q.view(T, B, -1)
+
+    def _h0_to_state(self, h0: torch.Tensor):
+        # h0: (B, 2, H) -> ((1,B,H), (1,B,H))
+        h = h0[:, 0].unsqueeze(0).contiguous()
+        c = h0[:, 1].unsqueeze(0).contiguous()
+        return (h.to(self.device), c.to(self.device))
+
+    # -- train -------------------------------------------------------------
+    def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
+        B = data["done"].shape[0]
+        T = self.T
+        states = data["states"].to(self.device, non_blocking=True)  # (B,T,4,84,84)
+        frames = ops.dequant_frames(
+            states.permute(1, 0, 2, 3, 4).contiguous(), torch.float32
+        )  # (T,B,4,84,84) seq-major (R2D2/Learner.py:93)
+        actions = data["actions"].to(self.device).t().contiguous()  # (T,B)
+        rewards = data["rewards"].to(self.device).t().contiguous()
+        done = data["done"].to(self.device)
+        weights = weights.to(self.device)
+        h0 = self._h0_to_state(data["h0"].to(self.device))
+
+        q_train = self._seq_forward(self.net, frames, h0, burn_in_split=True)
+        with torch.no_grad():
+            q_tgt_full = self._seq_forward(self.target, frames, h0,
+                                           burn_in_split=False)
+        # Double-DQN argmax is only consulted at steps t+n >= burn_in + 1,
+        # all inside the training window, so the burn-in rows of the online
+        # view are never read — pad them with (detached) target rows instead
+        # of paying a third sequence pass.
+        q_online_full = torch.cat([q_tgt_full[: self.burn_in], q_train], dim=0)
+
+        td, q_taken, targets = nstep_recurrent_targets(
+            q_online_full, q_tgt_full, actions, rewards, done, self.burn_in,
+            self.n_step, self.gamma, self.use_rescaling,
+        )
+        # loss: IS-weighted 0.5 * mean_t(td^2) per sequence
+        loss = 0.5 * (weights * td.pow(2).mean(dim=0)).mean()
+        prio = ops.sequence_priority(td.detach().abs(), self.alpha, ETA)
+
+        self.optim.zero_grad(set_to_none=False)
+        loss.backward()
+        if self.reducer is not None:
+            self.reducer.all_reduce()
+        self.model.clippingNorm(40.0)  # R2D2/Learner.py:208
+        self.optim.step()
+        self.replay.update(idx, prio)
+        return {
+            "loss": loss.detach(),
+            "value": q_taken.detach().mean(),
+            "td_abs": td.detach().abs().mean(),
+        }
+
+    def _inner_step(self):
+        data, idx, w = self.replay.sample(self.batch_size, self.beta)
+        return self.train_step(data, idx, w)
+
+    def _cadence(self):
+        self.step_count += 1
+        if self.step_count % self.cfg.target_frequency == 0:
+            self.target.updateParameter(self.model, 1.0)
+            self.publish_weights(include_target=True)
+        elif self.step_count % PUBLISH_EVERY == 0:
+            self.publish_weights()
+
+    def step(self):
+        stats = self._inner_step()
+        self._cadence()
+        return stats
+
+    # -- weights ------------------------------------------------------------
+    def publish_weights(self, include_target: bool = False):
+        if self.transport is None or self.rank != 0:
+            return
+        payload = {
+            "count": self.step_count,
+            "state_dict": {k: v.detach().to("cpu", torch.float32)
+                           for k, v in self.model.state_dict().items()},
+        }
+        if include_target:
+            payload["target_state_dict"] = {
+                k: v.detach().to("cpu", torch.float32)
+                for k, v in self.target.state_dict().items()
+            }
+        self.transport.publish(payload)
+
+    # -- run loop ------------------------------------------------------------
+    def run(self, max_steps: int = 1_000_000, warmup_items: Optional[int] = None):
+        need = warmup_items if warmup_items is not None else self.cfg.buffer_size
+        t0 = time.time()
+        while len(self.replay) <= need:
+            self.ingest()
+            if time.time() - t0 > 1200:
+                raise TimeoutError("R2D2 replay warmup stalled")
+            time.sleep(0.01)
+        self.publish_weights(include_target=True)
+        while self.step_count < max_steps:
+            self.ingest()
+            stats = self.step()
+            if self.step_count % self.LOG_EVERY == 0:
+                rewards = self.transport.drain_rewards() if self.transport else []
+                mean_r = float(np.mean(rewards)) if rewards else -21.0
+                self.log_scalar("Reward", mean_r)
+                self.log_scalar("value", float(stats["value"]))
+                self.log_scalar("norm", float(self.model.calculateNorm()))
+                if self.rank == 0:
+                    print(
+                        f"[R2D2] step={self.step_count} loss={float(stats['loss']):.5f} "
+                        f"value={float(stats['value']):.3f} reward={mean_r:.1f} "
+                        f"replay={len(self.replay)}", flush=True,
+                    )
+            if self.step_count % self.CKPT_EVERY == 0:
+                self.save_checkpoint()
+
+    # -- checkpoint ----------------------------------------------------------
+    def state_for_checkpoint(self):
+        return {
+            "alg": self.ALG,
+            "model": self.model.state_dict(),
+            "target": self.target.state_dict(),
+            "optim": self.optim.state_dict(),
+            "step": self.step_count,
+        }
+
+    def load_from_checkpoint(self, state):
+        self.model.load_state_dict(state["model"])
+        self.target.load_state_dict(state["target"])
+        self.optim.load_state_dict(state["optim"])
+        self.step_count = int(state["step"])
+
+    def load_model_only(self, sd):
+        self.model.load_state_dict(sd)
+        self.target.load_state_dict(sd)
+
+
+# ===========================================================================
+# Actor
+# ===========================================================================
+
+
+class R2D2Player:
+    """Recurrent actor: overlapping 80-step sequences with stored initial
+    hidden state and actor-side sequence priorities."""
+
+    def __init__(self, cfg: Config, idx: int, transport, env=None,
+                 env_kind: str = "auto", seed: Optional[int] = None):
+        from ..actors.env import make_env
+
+        self.cfg = cfg
+        self.idx = idx
+        self.transport = transport
+        self.env = env or make_env(env_kind, seed=seed if seed is not None else idx)
+        self.model = BaseAgent(cfg.model_info).to(cfg.actor_device).eval()
+        self.target = BaseAgent(cfg.model_info).to(cfg.actor_device).eval()
+        n_actors = max(cfg.num_actors, 2)
+        self.eps = 0.4 ** (1 + 7 * idx / (n_actors - 1))
+        self.T = cfg.fixed_trajectory
+        self.overlap = self.T // 2  # 40-step overlap (Player.py:60-61)
+        self.n_step = cfg.unroll_step
+        self.gamma = cfg.gamma
+        self.alpha = cfg.alpha
+        self.action_n = cfg.action_size
+        self.use_rescaling = cfg.use_rescaling
+        self.rng = np.random.default_rng(3000 + idx)
+        self.env_steps = 0
+        self.weight_version = -1
+        self._steps: List[tuple] = []  # (state, action, reward)
+        self._hiddens: List[tuple] = []  # cell state BEFORE each step
+
+    # -- inference ---------------------------------------------------------
+    @torch.no_grad()
+    def act(self, state_u8: np.ndarray) -> int:
+        x = torch.from_numpy(state_u8).unsqueeze(0).float() / 255.0
+        q = self.model.forward([x, torch.tensor([1, 1, -1])])[0]
+        if self.rng.random() < self.eps:
+            return int(self.rng.integers(0, self.action_n))
+        return int(q.argmax(1).item())
+
+    # -- sequence priority (replay window under no_grad) -------------------
+    @torch.no_grad()
+    def _sequence_priority(self, states_np, actions_np, rewards_np,
+                           h0_state, done: float) -> float:
+        T = self.T
+        frames = torch.from_numpy(states_np).float().unsqueeze(1) / 255.0  # (T,1,...)
+        live_m = self.model.getCellState()
+        live_t = self.target.getCellState()
+        self.model.setCellState((h0_state[0].clone(), h0_state[1].clone()))
+        self.target.setCellState((h0_state[0].clone(), h0_state[1].clone()))
+        hint = torch.tensor([T, 1, -1])
+        q_on = self.model.forward([frames.reshape(T, *frames.shape[2:]), hint]
+                                  )[0].view(T, 1, -1)
+        q_tg = self.target.forward([frames.reshape(T, *frames.shape[2:]), hint]
+                                   )[0].view(T, 1, -1)
+        self.model.setCellState(live_m)
+        self.target.setCellState(live_t)
+        td, _, _ = nstep_recurrent_targets(
+            q_on, q_tg, torch.from_numpy(actions_np).unsqueeze(1),
+            torch.from_numpy(rewards_np).unsqueeze(1),
+            torch.tensor([done]), 0, self.n_step, self.gamma,
+            self.use_rescaling,
+        )
+        mix = ETA * td.abs().max() + (1 - ETA) * td.abs().mean()
+        return float(mix ** self.alpha)
+
+    def _emit(self, done: bool):
+        T = self.T
+        steps = self._steps[:T]
+        h0_state = self._hiddens[0]
+        states_np = np.stack([s[0] for s in steps])
+        actions_np = np.array([s[1] for s in steps], np.int32)
+        rewards_np = np.array([s[2] for s in steps], np.float32)
+        prio = self._sequence_priority(states_np, actions_np, rewards_np,
+                                       h0_state, 1.0 if done else 0.0)
+        h0_np = torch.cat([h0_state[0][0], h0_state[1][0]], dim=0).numpy()  # (2,H)
+        cols = {
+            "h0": h0_np[None],
+            "states": states_np[None],
+            "actions": actions_np[None],
+            "rewards": rewards_np[None],
+            "done": np.array([1.0 if done else 0.0], np.float32),
+        }
+        self.transport.push(cols, np.array([prio], np.float32))
+
+    def pull_weights(self):
+        payload = self.transport.fetch()
+        if payload is None:
+            return
+        if payload.get("count", 0) == self.weight_version:
+            return
+        self.model.load_state_dict(payload["state_dict"])
+        if "target_state_dict" in payload:
+            self.target.load_state_dict(payload["target_state_dict"])
+        self.weight_version = payload.get("count", 0)
+
+    def _zero_hidden(self):
+        self.model.zeroCellState(1)
+        self.target.zeroCellState(1)
+
+    def run(self, max_env_steps: int = 1_000_000):
+        self.pull_weights()
+        self._zero_hidden()
+        state = self.env.reset()
+        episode_reward = 0.0
+        while self.env_steps < max_env_steps:
+            h = self.model.getCellState()
+            self._hiddens.append((h[0].clone(), h[1].clone()))
+            action = self.act(state)
+            next_state, reward, done, info = self.env.step(action)
+            episode_reward += reward
+            self._steps.append((state, action, reward))
+            if len(self._steps) == self.T:
+                self._emit(done=done)
+                if done:
+                    self._steps, self._hiddens = [], []
+                else:
+                    self._steps = self._steps[self.overlap:]
+                    self._hiddens = self._hiddens[self.overlap:]
+            elif done:
+                self._steps, self._hiddens = [], []
+            state = next_state
+            self.env_steps += 1
+            if self.env_steps % ACTOR_PULL_EVERY == 0:
+                self.pull_weights()
+            if done:
+                self.transport.push_reward(self.idx, episode_reward, self.eps)
+                episode_reward = 0.0
+                self._zero_hidden()
+                state = self.env.reset()
