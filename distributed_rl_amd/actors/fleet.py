@@ -1,0 +1,102 @@
+"""Actor fleet: one OS process per actor over a shared-memory transport
+session — the Ray + Redis replacement (run_actor.py:46-55 spawned Ray
+remote actors pinned to 1 CPU; here plain multiprocessing with the SPSC
+ring transport)."""
+
+from __future__ import annotations
+
+import multiprocessing as mp
+import os
+import signal
+import time
+from typing import List, Optional
+
+from ..algos import get_player_cls, get_wire_schema
+from ..config import load_config
+from .transport import ActorEndpoint, RecordCodec, TransportSession
+
+
+def _actor_main(cfg_spec: str, idx: int, transport_dir: str,
+                max_env_steps: int, env_kind: str):
+    # actors are CPU-only: keep torch single-threaded per actor
+    os.environ.setdefault("OMP_NUM_THREADS", "1")
+    import torch
+
+    torch.set_num_threads(1)
+    cfg = load_config(cfg_spec)
+    schema, with_prio = get_wire_schema(cfg)
+    codec = RecordCodec(schema, with_priority=with_prio)
+    session = TransportSession(transport_dir, codec, num_rings=0, create=False)
+    endpoint = ActorEndpoint(session, idx)
+    player = get_player_cls(cfg.alg)(cfg, idx=idx, transport=endpoint,
+                                     env_kind=env_kind)
+    try:
+        player.run(max_env_steps=max_env_steps)
+    except KeyboardInterrupt:
+        pass
+    finally:
+        session.close()
+
+
+class ActorFleet:
+    """Spawn/supervise N actor processes. respawn_on_exit keeps the fleet at
+    full strength (failure handling the reference lacks)."""
+
+    def __init__(self, cfg_spec: str, num_actors: int, transport_dir: str,
+                 start_idx: int = 0, env_kind: str = "auto",
+                 max_env_steps: int = 1 << 60, respawn_on_exit: bool = True):
+        self.cfg_spec = cfg_spec
+        self.num_actors = num_actors
+        self.start_idx = start_idx
+        self.transport_dir = transport_dir
+        self.env_kind = env_kind
+        self.max_env_steps = max_env_steps
+        self.respawn = respawn_on_exit
+        self.ctx = mp.get_context("spawn")
+        self.procs: List[Optional[mp.Process]] = [None] * num_actors
+
+    def _spawn(self, slot: int):
+        idx = self.start_idx + slot
+        p = self.ctx.Process(
+            target=_actor_main,
+            args=(self.cfg_spec, idx, self.transport_dir, self.max_env_steps,
+                  self.env_kind),
+            daemon=True,
+            name=f"drl-actor-{idx}",
+        )
+        p.start()
+        self.procs[slot] = p
+
+    def start(self):
+        for i in range(self.num_actors):
+            self._spawn(i)
+
+    def supervise(self, poll_s: float = 5.0):
+        """Blocking loop: respawn dead actors (heartbeat = process liveness)."""
+        while True:
+            time.sleep(poll_s)
+            alive = 0
+            for i, p in enumerate(self.procs):
+                if p is None or not p.is_alive():
+                    if self.respawn:
+                        self._spawn(i)
+                else:
+                    alive += 1
+            if alive == 0 and not self.respawn:
+                return
+
+    def alive_count(self) -> int:
+        return sum(1 for p in self.procs if p is not None and p.is_alive())
+
+    def join(self, timeout: Optional[float] = None):
+        for p in self.procs:
+            if p is not None:
+                p.join(timeout)
+
+    def stop(self):
+        for p in self.procs:
+            if p is not None and p.is_alive():
+                p.terminate()
+        for p in self.procs:
+            if p is not None:
+                p.join(5)

@@ -330,10 +330,10 @@ class TransportSession:
         return out
 
     # -- learner side -----------------------------------------------------
-    def drain(self, max_per_ring: int = 1 << 30
+    def drain(self, max_per_ring: int = 1 << 30, ring_ids=None
               ) -> Optional[Tuple[Dict[str, np.ndarray], Optional[np.ndarray]]]:
         chunks = []
-        for i in range(self.num_rings):
+        for i in (range(self.num_rings) if ring_ids is None else ring_ids):
             r = self.ring(i).pop_records(max_per_ring)
             if r is not None:
                 chunks.append(r)
@@ -377,13 +377,21 @@ class ActorEndpoint:
 
 
 class LearnerEndpoint:
-    """Learner-side view: drain all rings, publish weights."""
+    """Learner-side view: drain (a partition of) the rings, publish weights.
 
-    def __init__(self, session: TransportSession):
+    For learner data-parallelism, rank r of world w drains rings
+    i % w == r — each GPU replica owns a disjoint actor subset and its own
+    HBM replay shard (SURVEY §2.6 build implication c)."""
+
+    def __init__(self, session: TransportSession, rank: int = 0,
+                 world_size: int = 1):
         self.session = session
+        self.ring_ids = [
+            i for i in range(session.num_rings) if i % world_size == rank
+        ]
 
     def drain(self):
-        return self.session.drain()
+        return self.session.drain(ring_ids=self.ring_ids)
 
     def publish(self, obj):
         self.session.weight_bus.publish(obj)
