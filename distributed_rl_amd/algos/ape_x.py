@@ -90,7 +90,15 @@ class ApexLearner(LearnerBase):
         rdev = replay_device or (
             str(self.device) if self.device.type == "cuda" else "cpu"
         )
-        self.replay = make_per(cap, make_apex_schema(), device=rdev)
+        # On GPU, frames are stored NHWC so the first fused conv consumes the
+        # uint8 replay column directly (fused dequant in conv_mfma.hip) —
+        # no standalone dequant pass, no layout transposes.
+        self._nhwc = self.device.type == "cuda"
+        schema = (
+            make_apex_schema(frame_shape=(84, 84, 4)) if self._nhwc
+            else make_apex_schema()
+        )
+        self.replay = make_per(cap, schema, device=rdev)
         self.transport = transport
         self.gamma = cfg.gamma
         self.n_step = cfg.unroll_step
@@ -125,6 +133,9 @@ class ApexLearner(LearnerBase):
                 dev_cols = {
                     k: v.to(self.device, non_blocking=True) for k, v in cols.items()
                 }
+                if self._nhwc:
+                    for k in ("state", "next_state"):
+                        dev_cols[k] = dev_cols[k].permute(0, 2, 3, 1).contiguous()
                 self.replay.push(dev_cols, prio.to(self.device, non_blocking=True))
             torch.cuda.current_stream(self.device).wait_stream(self._ingest_stream)
         else:
@@ -132,7 +143,11 @@ class ApexLearner(LearnerBase):
         return n
 
     def push_experience(self, cols: Dict[str, torch.Tensor], prio: torch.Tensor):
-        """Direct (in-process) push, e.g. from the bench prefill."""
+        """Direct (in-process) push; frames arrive NCHW (wire format)."""
+        if self._nhwc:
+            cols = dict(cols)
+            for k in ("state", "next_state"):
+                cols[k] = cols[k].permute(0, 2, 3, 1).contiguous()
         self.replay.push(cols, prio)
 
     # ------------------------------------------------------------------
@@ -141,10 +156,9 @@ class ApexLearner(LearnerBase):
     def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
         cuda = self.device.type == "cuda"
         if cuda:
-            s = ops.dequant_frames_nhwc(data["state"].to(self.device, non_blocking=True))
-            sp = ops.dequant_frames_nhwc(
-                data["next_state"].to(self.device, non_blocking=True)
-            )
+            # NHWC uint8 straight into the fused conv stack (dequant fused)
+            s = data["state"].permute(0, 3, 1, 2)
+            sp = data["next_state"].permute(0, 3, 1, 2)
         else:
             s = ops.dequant_frames(data["state"], torch.float32)
             sp = ops.dequant_frames(data["next_state"], torch.float32)

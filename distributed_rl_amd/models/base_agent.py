@@ -54,18 +54,68 @@ class CNN2D(nn.Module):
         layers: List[nn.Module] = []
         n_conv = len(n_unit)
         ch = in_ch
+        self._convs: List[nn.Conv2d] = []
         for i in range(n_conv):
-            layers.append(
-                nn.Conv2d(ch, n_unit[i], kernel_size=f_size[i], stride=stride[i], padding=padding[i])
-            )
+            conv = nn.Conv2d(ch, n_unit[i], kernel_size=f_size[i],
+                             stride=stride[i], padding=padding[i])
+            self._convs.append(conv)
+            layers.append(conv)
             if i < len(bn) and bn[i]:
                 layers.append(nn.BatchNorm2d(n_unit[i]))
             layers.append(_act(acts[i]))
             ch = n_unit[i]
         self.body = nn.Sequential(*layers)
+        # fused-path eligibility (static part): plain strided ReLU convs
+        self._fusable_static = (
+            all(a.lower() == "relu" for a in acts[:n_conv])
+            and not any(bn[:n_conv])
+            and all(p == 0 for p in padding[:n_conv])
+        )
+        self._fused_checked: dict = {}
+
+    def _fused_ok(self, x: torch.Tensor) -> bool:
+        """Use the hand-written MFMA conv kernels (ops/hip/conv_mfma.hip)
+        when the whole stack's geometry is covered and weights are bf16
+        channels_last on a GPU."""
+        if not (self._fusable_static and x.is_cuda):
+            return False
+        if x.dtype not in (torch.uint8, torch.bfloat16):
+            return False
+        key = (tuple(x.shape[1:]), x.dtype)
+        hit = self._fused_checked.get(key)
+        if hit is not None:
+            return hit
+        from .. import ops as _ops
+
+        ok = self._convs[0].weight.dtype == torch.bfloat16 and x.is_contiguous(
+            memory_format=torch.channels_last
+        )
+        if ok:
+            C, H, W = x.shape[1:]
+            u8 = x.dtype == torch.uint8
+            for conv in self._convs:
+                COUT, _, KH, KW = conv.weight.shape
+                S = conv.stride[0]
+                if not _ops.conv_supported(H, W, C, KH, KW, S, COUT, u8):
+                    ok = False
+                    break
+                H = (H - KH) // S + 1
+                W = (W - KW) // S + 1
+                C = COUT
+                u8 = False
+        self._fused_checked[key] = ok
+        return ok
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        y = self.body(x)
+        if self._fused_ok(x):
+            from .. import ops as _ops
+
+            y = x
+            for conv in self._convs:
+                y = _ops.fused_conv_relu(y, conv.weight, conv.bias,
+                                         conv.stride[0])
+        else:
+            y = self.body(x)
         if self.flatten:
             y = torch.flatten(y, 1)
         return y

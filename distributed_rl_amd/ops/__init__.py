@@ -84,6 +84,65 @@ def dequant_frames_nhwc(x_u8: torch.Tensor) -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
+# K2 — fused MFMA conv (NHWC bf16, fused dequant/bias/ReLU) with aten backward
+# ---------------------------------------------------------------------------
+
+
+def conv_supported(H, W, C, KH, KW, S, COUT, u8: bool) -> bool:
+    ext = hip_ext(required=False)
+    if ext is None:
+        return False
+    return bool(ext.conv_fwd_supported(H, W, C, KH, KW, S, COUT, u8))
+
+
+class _FusedConvFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, stride):
+        ext = hip_ext()
+        N, C, H, W = x.shape
+        COUT, _, KH, KW = weight.shape
+        P = (H - KH) // stride + 1
+        Q = (W - KW) // stride + 1
+        out = torch.empty(
+            N, COUT, P, Q, dtype=torch.bfloat16, device=x.device
+        ).to(memory_format=torch.channels_last)
+        ext.conv_fwd(
+            x, weight,
+            bias if bias is not None else torch.empty(0, device=x.device),
+            out, stride,
+        )
+        ctx.save_for_backward(x, weight, out)
+        ctx.stride = stride
+        ctx.has_bias = bias is not None
+        return out
+
+    @staticmethod
+    def backward(ctx, gout):
+        x, weight, out = ctx.saved_tensors
+        stride = ctx.stride
+        gout = gout.contiguous(memory_format=torch.channels_last)
+        gout = gout * (out > 0)  # ReLU was folded into the forward
+        if x.dtype == torch.uint8:
+            xf = x.to(torch.bfloat16)
+            xf = xf / 255.0  # layout (channels_last) is preserved
+        else:
+            xf = x
+        need_x = ctx.needs_input_grad[0]
+        gi, gw, gb = torch.ops.aten.convolution_backward(
+            gout, xf, weight,
+            [weight.shape[0]] if ctx.has_bias else None,
+            [stride, stride], [0, 0], [1, 1], False, [0, 0], 1,
+            [need_x, ctx.needs_input_grad[1], ctx.has_bias and ctx.needs_input_grad[2]],
+        )
+        return (gi if need_x else None), gw, gb, None
+
+
+def fused_conv_relu(x, weight, bias, stride: int):
+    """One fused conv+bias+ReLU layer (u8 or bf16 channels_last input)."""
+    return _FusedConvFn.apply(x, weight, bias, stride)
+
+
+# ---------------------------------------------------------------------------
 # K4 — fused n-step double-DQN loss as an autograd Function
 # ---------------------------------------------------------------------------
 

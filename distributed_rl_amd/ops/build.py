@@ -13,7 +13,10 @@ import sys
 EXT_NAME = "_drl_hip"
 _THIS_DIR = os.path.dirname(os.path.abspath(__file__))
 BUILD_DIR = os.path.join(_THIS_DIR, "_build")
-SOURCES = [os.path.join(_THIS_DIR, "hip", "drl_kernels.hip")]
+SOURCES = [
+    os.path.join(_THIS_DIR, "hip", "drl_kernels.hip"),
+    os.path.join(_THIS_DIR, "hip", "conv_mfma.hip"),
+]
 
 
 def so_path() -> str:

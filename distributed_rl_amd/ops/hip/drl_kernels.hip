@@ -534,7 +534,10 @@ void grad_clip(torch::Tensor flat, double max_norm, torch::Tensor sqsum_buf) {
                      sqsum_buf.data_ptr<float>(), (float)max_norm);
 }
 
+void register_conv(pybind11::module_& m);  // conv_mfma.hip
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  register_conv(m);
   m.def("dequant", &dequant, "u8 -> bf16/f32 /255 (K1)");
   m.def("dequant_nhwc", &dequant_nhwc, "u8 NCHW -> bf16 NHWC /255 (K1)");
   m.def("sumtree_update", &sumtree_update, "lock-free sum-tree leaf update (K10)");
