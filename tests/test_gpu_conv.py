@@ -98,13 +98,17 @@ def test_conv_backward_matches_eager(ext):
     b2 = b.detach().clone().requires_grad_(True)
     ref = F.relu(F.conv2d(x2, w2, b2, stride=S))
     ref.backward(g)
-    # forward outputs match to bf16 noise, so the ReLU masks agree except on
-    # near-zero entries; compare with loose tolerance
+    # The ReLU mask can flip on entries where |out| ~ bf16 noise, which puts
+    # full |g| error on a handful of elements — compare statistically: small
+    # mean error, and <1% of entries off by more than 10% of scale.
     for mine, theirs, name in [(gx, x2.grad, "gx"), (gw, w2.grad, "gw"),
                                (gb, b2.grad, "gb")]:
-        scale = theirs.float().abs().max().item() + 1e-6
-        err = (mine.float() - theirs.float()).abs().max().item()
-        assert err / scale < 5e-2, f"{name}: {err} vs {scale}"
+        m, t = mine.float(), theirs.float()
+        scale = t.abs().max().item() + 1e-6
+        mae = (m - t).abs().mean().item()
+        frac_big = ((m - t).abs() > 0.1 * scale).float().mean().item()
+        assert mae / scale < 5e-3, f"{name}: mae {mae} scale {scale}"
+        assert frac_big < 0.01, f"{name}: {frac_big:.3%} entries off"
 
 
 def test_model_fused_path_active(ext):
