@@ -216,28 +216,76 @@ class ApexLearner(LearnerBase):
         return stats
 
     def make_graphed_step(self, warmup_iters: int = 3):
-        """hipGraph-capture the whole inner step (PER sample -> dequant ->
-        3 forwards -> fused loss -> backward -> grad cast -> optimizer ->
+        """hipGraph-capture the learner step (PER sample -> fused-conv
+        forwards -> fused loss -> backward -> grad cast -> optimizer ->
         priority update). All shapes are static and RNG lives in a device
         seed buffer, so one graph replay per learner step. Cadence ops
         (target sync, weight publish) stay eager. Requires a warm, fixed-size
-        replay (n_valid is baked into the sample kernel)."""
+        replay (n_valid is baked into the sample kernel).
+
+        At world_size > 1 the step is captured as TWO graphs with the RCCL
+        all-reduce running eagerly between them — collectives are never
+        captured, so the multi-GPU path cannot be broken by graph-capture
+        support gaps in the collective stack."""
         assert self.device.type == "cuda", "graph capture needs a GPU"
         for g in self.optim.param_groups:
             g["capturable"] = True
+        split = self.mp is not None and self.mp.world > 1
         side = torch.cuda.Stream(self.device)
         side.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(side):
             for _ in range(warmup_iters):
                 self._inner_step()
         torch.cuda.current_stream(self.device).wait_stream(side)
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            static_out = self._inner_step()
-        self._graph = graph  # keep alive (owns the memory pool)
+
+        if not split:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_out = self._inner_step()
+            self._graph = graph  # keep alive (owns the memory pool)
+
+            def stepper():
+                graph.replay()
+                self._cadence()
+                return static_out
+
+            return stepper
+
+        # ---- split capture: g1 = sample..backward, eager all-reduce,
+        # ---- g2 = upcast + optimizer + param sync + priority update
+        mp = self.mp
+        g1 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g1):
+            data, s_idx, s_w = self.replay.sample(self.batch_size, self.beta)
+            s = data["state"].permute(0, 3, 1, 2)
+            sp = data["next_state"].permute(0, 3, 1, 2)
+            actions = data["action"].long()
+            q_s = self.net.forward([s])[0]
+            with torch.no_grad():
+                q_sp_on = self.net.forward([sp])[0]
+                q_sp_tg = self.target.forward([sp])[0]
+            loss, prio = ops.nstep_dqn_loss(
+                q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions,
+                data["reward"], data["done"], s_w, self.gamma, self.n_step,
+                self.alpha,
+            )
+            mp.zero_grads()
+            loss.backward()
+            static_out = {"loss": loss.detach(),
+                          "value": q_s.detach().float().max(1).values.mean()}
+        g2 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g2, pool=g1.pool()):
+            mp.flat_mgrad.copy_(mp.flat_cgrad)
+            mp.flat_mgrad.mul_(1.0 / mp.world)
+            self.optim.step()
+            mp.sync_compute_params()
+            self.replay.update(s_idx, prio)
+        self._graph = (g1, g2)
 
         def stepper():
-            graph.replay()
+            g1.replay()
+            torch.distributed.all_reduce(mp.flat_cgrad, group=mp.group)
+            g2.replay()
             self._cadence()
             return static_out
 

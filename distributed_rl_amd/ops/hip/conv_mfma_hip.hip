@@ -47,11 +47,13 @@ struct ConvGeom {
   static constexpr int LDS_ELEMS = COUT * LDS_ROW;
 };
 
-// One block: 4 waves x 16 rows = 64 output pixels, full COUT width.
-template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8IN>
-__global__ __launch_bounds__(256) void conv_fwd_kernel(
+// A block = WAVES waves, each computing RPW rows x COUT cols
+// (RPW/16 row-fragments; B fragments shared across row-fragments).
+template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8IN,
+          int WAVES, int RPW>
+__global__ __launch_bounds__(WAVES * 64) void conv_fwd_kernel(
     const void* __restrict__ in_v,       // (N,H,W,C) u8 or bf16 (NHWC)
-    const __bf16* __restrict__ weight,   // (COUT, K) row-major (= torch OIHW->O,KH,KW,C NHWC flat)
+    const __bf16* __restrict__ weight,   // (COUT, K) row-major NHWC flat
     const __bf16* __restrict__ bias,     // (COUT), may be null
     __bf16* __restrict__ out,            // (N,P,Q,COUT)
     int batch) {
@@ -64,9 +66,8 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
   const int wave = tid >> 6;
 
   // ---- stage weights into LDS: [cout][K + pad] ----
-  // global W is [COUT][K] contiguous; 256 threads, 8 elems (16 B) each.
   constexpr int WELEMS = COUT * G::K;
-  for (int base = tid * 8; base < WELEMS; base += 256 * 8) {
+  for (int base = tid * 8; base < WELEMS; base += WAVES * 64 * 8) {
     int co = base / G::K;
     int k = base - co * G::K;
     bf16x8 v = *reinterpret_cast<const bf16x8*>(weight + co * G::K + k);
@@ -75,74 +76,85 @@ __global__ __launch_bounds__(256) void conv_fwd_kernel(
   __syncthreads();
 
   const int M = batch * G::P * G::Q;
-  // row tile for this wave
-  const int row0 = (blockIdx.x * 4 + wave) * 16;
+  const int row0 = (blockIdx.x * WAVES + wave) * RPW;
   if (row0 >= M) return;
-  // per-lane A row (row0 + lane&15), clamped for the ragged tail
-  int arow = row0 + (lane & 15);
-  if (arow >= M) arow = M - 1;
-  const int n = arow / (G::P * G::Q);
-  const int rem = arow - n * (G::P * G::Q);
-  const int p = rem / G::Q;
-  const int q = rem - p * G::Q;
-  // base element offset of this row's patch in the input
-  const int64_t in_row0 = ((int64_t)n * H + p * S) * (W * C) + q * S * C;
-  const int kpart = (lane >> 4) * 8;  // this lane's 8-element k offset in chunk
-
+  constexpr int RFRAG = RPW / 16;
   constexpr int NFRAG = COUT / 16;
-  f32x4 acc[NFRAG];
+  // per-lane A row bases, one per row-fragment (clamped for ragged tail)
+  int64_t in_row0[RFRAG];
 #pragma unroll
-  for (int f = 0; f < NFRAG; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
+  for (int rf = 0; rf < RFRAG; ++rf) {
+    int arow = row0 + rf * 16 + (lane & 15);
+    if (arow >= M) arow = M - 1;
+    const int n = arow / (G::P * G::Q);
+    const int rem = arow - n * (G::P * G::Q);
+    const int p = rem / G::Q;
+    const int q = rem - p * G::Q;
+    in_row0[rf] = ((int64_t)n * H + p * S) * (W * C) + q * S * C;
+  }
+  const int kpart = (lane >> 4) * 8;
+
+  f32x4 acc[RFRAG][NFRAG];
+#pragma unroll
+  for (int rf = 0; rf < RFRAG; ++rf)
+#pragma unroll
+    for (int f = 0; f < NFRAG; ++f) acc[rf][f] = {0.f, 0.f, 0.f, 0.f};
 
 #pragma unroll
   for (int kc = 0; kc < G::KCHUNKS; ++kc) {
-    const int kelem = kc * 32 + kpart;           // element within K
-    const int dy = kelem / G::ROWC;              // kh
-    const int dx = kelem - dy * G::ROWC;         // offset within patch row
-    // ---- A fragment: 8 input elements (16 B for bf16, 8 B for u8) ----
-    bf16x8 a;
-    const int64_t goff = in_row0 + (int64_t)dy * (W * C) + dx;
-    if constexpr (U8IN) {
-      const uint8_t* src = reinterpret_cast<const uint8_t*>(in_v) + goff;
-      // 8 bytes -> 8 bf16 * (1/255)
-      uint2 raw = *reinterpret_cast<const uint2*>(src);
-      const float inv255 = 1.0f / 255.0f;
+    const int kelem = kc * 32 + kpart;
+    const int dy = kelem / G::ROWC;
+    const int dx = kelem - dy * G::ROWC;
+    const int64_t koff = (int64_t)dy * (W * C) + dx;
+    bf16x8 a[RFRAG];
 #pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        unsigned byte = (j < 4 ? raw.x >> (8 * j) : raw.y >> (8 * (j - 4))) & 0xFF;
-        a[j] = (__bf16)(byte * inv255);
+    for (int rf = 0; rf < RFRAG; ++rf) {
+      const int64_t goff = in_row0[rf] + koff;
+      if constexpr (U8IN) {
+        const uint8_t* src = reinterpret_cast<const uint8_t*>(in_v) + goff;
+        uint2 raw = *reinterpret_cast<const uint2*>(src);
+        const float inv255 = 1.0f / 255.0f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned byte =
+              (j < 4 ? raw.x >> (8 * j) : raw.y >> (8 * (j - 4))) & 0xFF;
+          a[rf][j] = (__bf16)(byte * inv255);
+        }
+      } else {
+        a[rf] = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const __bf16*>(in_v) + goff);
       }
-    } else {
-      a = *reinterpret_cast<const bf16x8*>(
-          reinterpret_cast<const __bf16*>(in_v) + goff);
     }
-    // ---- B fragments from LDS + MFMA ----
 #pragma unroll
     for (int f = 0; f < NFRAG; ++f) {
       const int col = f * 16 + (lane & 15);
       bf16x8 b = *reinterpret_cast<const bf16x8*>(
           wlds + col * G::LDS_ROW + kc * 32 + kpart);
-      acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc[f], 0, 0, 0);
+#pragma unroll
+      for (int rf = 0; rf < RFRAG; ++rf)
+        acc[rf][f] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[rf], b, acc[rf][f], 0, 0, 0);
     }
   }
 
   // ---- epilogue: bias + ReLU + bf16 store ----
-  // lane l holds rows (l>>4)*4 + r (r=0..3), col l&15 of each fragment
   const int crow_base = (lane >> 4) * 4;
 #pragma unroll
-  for (int f = 0; f < NFRAG; ++f) {
-    const int col = f * 16 + (lane & 15);
-    const float bv = bias ? (float)bias[col] : 0.0f;
+  for (int rf = 0; rf < RFRAG; ++rf)
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      const int orow = row0 + crow_base + r;
-      if (orow < M) {
-        float v = acc[f][r] + bv;
-        v = v > 0.0f ? v : 0.0f;  // ReLU (all reference convs use relu)
-        out[(int64_t)orow * COUT + col] = (__bf16)v;
+    for (int f = 0; f < NFRAG; ++f) {
+      const int col = f * 16 + (lane & 15);
+      const float bv = bias ? (float)bias[col] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int orow = row0 + rf * 16 + crow_base + r;
+        if (orow < M) {
+          float v = acc[rf][f][r] + bv;
+          v = v > 0.0f ? v : 0.0f;
+          out[(int64_t)orow * COUT + col] = (__bf16)v;
+        }
       }
     }
-  }
 }
 
 struct ConvLaunch {
@@ -150,26 +162,45 @@ struct ConvLaunch {
   bool u8;
   void (*fn)(const void*, const __bf16*, const __bf16*, __bf16*, int);
   int lds_bytes;
+  int waves, rpw;
 };
 
-template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8>
+template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8,
+          int WAVES = 4, int RPW = 16>
 ConvLaunch make_launch() {
   using G = ConvGeom<H, W, C, KH, KW, S, COUT, U8>;
   return ConvLaunch{H, W, C, KH, KW, S, COUT, U8,
-                    conv_fwd_kernel<H, W, C, KH, KW, S, COUT, U8>,
-                    (int)(G::LDS_ELEMS * sizeof(__bf16))};
+                    conv_fwd_kernel<H, W, C, KH, KW, S, COUT, U8, WAVES, RPW>,
+                    (int)(G::LDS_ELEMS * sizeof(__bf16)), WAVES, RPW};
 }
 
 static const ConvLaunch kLaunches[] = {
-    // Ape-X / R2D2 stack (cfg/ape_x.json:38-51)
-    make_launch<84, 84, 4, 8, 8, 4, 32, true>(),
-    make_launch<84, 84, 4, 8, 8, 4, 32, false>(),
-    make_launch<20, 20, 32, 4, 4, 2, 64, false>(),
-    make_launch<9, 9, 64, 3, 3, 1, 64, false>(),
+    // Ape-X / R2D2 stack (cfg/ape_x.json:38-51); WAVES/RPW picked by the
+    // variant sweep in tools/gpu_conv_tune.py (profiles/)
+    make_launch<84, 84, 4, 8, 8, 4, 32, true, 8, 16>(),
+    make_launch<84, 84, 4, 8, 8, 4, 32, false, 8, 16>(),
+    make_launch<20, 20, 32, 4, 4, 2, 64, false, 8, 16>(),
+    make_launch<9, 9, 64, 3, 3, 1, 64, false, 8, 16>(),
     // IMPALA stack (cfg/impala.json:26-39)
-    make_launch<84, 84, 4, 8, 8, 4, 16, true>(),
-    make_launch<84, 84, 4, 8, 8, 4, 16, false>(),
-    make_launch<20, 20, 16, 4, 4, 2, 32, false>(),
+    make_launch<84, 84, 4, 8, 8, 4, 16, true, 8, 16>(),
+    make_launch<84, 84, 4, 8, 8, 4, 16, false, 8, 16>(),
+    make_launch<20, 20, 16, 4, 4, 2, 32, false, 8, 16>(),
+};
+
+// variant table for on-GPU tuning (tools/gpu_conv_tune.py)
+static const ConvLaunch kVariants[] = {
+    make_launch<84, 84, 4, 8, 8, 4, 32, true, 4, 16>(),
+    make_launch<84, 84, 4, 8, 8, 4, 32, true, 8, 16>(),
+    make_launch<84, 84, 4, 8, 8, 4, 32, true, 4, 32>(),
+    make_launch<84, 84, 4, 8, 8, 4, 32, true, 8, 32>(),
+    make_launch<20, 20, 32, 4, 4, 2, 64, false, 4, 16>(),
+    make_launch<20, 20, 32, 4, 4, 2, 64, false, 8, 16>(),
+    make_launch<20, 20, 32, 4, 4, 2, 64, false, 4, 32>(),
+    make_launch<20, 20, 32, 4, 4, 2, 64, false, 8, 32>(),
+    make_launch<9, 9, 64, 3, 3, 1, 64, false, 4, 16>(),
+    make_launch<9, 9, 64, 3, 3, 1, 64, false, 8, 16>(),
+    make_launch<9, 9, 64, 3, 3, 1, 64, false, 4, 32>(),
+    make_launch<9, 9, 64, 3, 3, 1, 64, false, 8, 32>(),
 };
 
 // Layout probe: C(16,16) = A(16,32) x B(32,16) with exactly the fragment
@@ -183,8 +214,8 @@ __global__ void mfma_probe_kernel(const __bf16* __restrict__ A,
   int kpart = (lane >> 4) * 8;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    a[j] = A[(lane & 15) * 32 + kpart + j];  // A[row][k]
-    b[j] = B[(kpart + j) * 16 + (lane & 15)];  // B[k][col]
+    a[j] = A[(lane & 15) * 32 + kpart + j];   // A[row][k]
+    b[j] = B[(kpart + j) * 16 + (lane & 15)]; // B[k][col]
   }
   f32x4 acc = {0.f, 0.f, 0.f, 0.f};
   acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, b, acc, 0, 0, 0);
@@ -242,17 +273,49 @@ void conv_fwd(torch::Tensor in, torch::Tensor weight, torch::Tensor bias,
               " k", KH, "x", KW, " s", stride, " -> ", COUT, " u8=", u8);
   const int P = (H - KH) / (int)stride + 1, Q = (W - KW) / (int)stride + 1;
   const int M = N * P * Q;
-  const int blocks = (M + 63) / 64;
+  const int rows_per_block = L->waves * L->rpw;
+  const int blocks = (M + rows_per_block - 1) / rows_per_block;
   const __bf16* bias_ptr = nullptr;
   if (bias.defined() && bias.numel() > 0) {
     TORCH_CHECK(bias.scalar_type() == torch::kBFloat16);
     bias_ptr = (const __bf16*)bias.data_ptr();
   }
-  hipLaunchKernelGGL(L->fn, dim3(blocks), dim3(256), L->lds_bytes,
+  hipLaunchKernelGGL(L->fn, dim3(blocks), dim3(L->waves * 64), L->lds_bytes,
                      (hipStream_t)at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream(),
                      (const void*)in.data_ptr(),
                      (const __bf16*)weight.data_ptr(), bias_ptr,
                      (__bf16*)out.data_ptr(), N);
+}
+
+// Run a specific tuning variant (tools/gpu_conv_tune.py); returns false if
+// the variant index does not match the geometry.
+bool conv_fwd_variant(torch::Tensor in, torch::Tensor weight,
+                      torch::Tensor bias, torch::Tensor out, int64_t stride,
+                      int64_t variant) {
+  const int N = (int)in.size(0), C = (int)in.size(1), H = (int)in.size(2),
+            W = (int)in.size(3);
+  const int COUT = (int)weight.size(0), KH = (int)weight.size(2),
+            KW = (int)weight.size(3);
+  const bool u8 = in.scalar_type() == torch::kUInt8;
+  if (variant < 0 || variant >= (int64_t)(sizeof(kVariants) / sizeof(ConvLaunch)))
+    return false;
+  const ConvLaunch* L = &kVariants[variant];
+  if (!(L->H == H && L->W == W && L->C == C && L->KH == KH && L->KW == KW &&
+        L->S == (int)stride && L->COUT == COUT && L->u8 == u8))
+    return false;
+  const int P = (H - KH) / (int)stride + 1, Q = (W - KW) / (int)stride + 1;
+  const int M = N * P * Q;
+  const int rows_per_block = L->waves * L->rpw;
+  const int blocks = (M + rows_per_block - 1) / rows_per_block;
+  const __bf16* bias_ptr = nullptr;
+  if (bias.defined() && bias.numel() > 0)
+    bias_ptr = (const __bf16*)bias.data_ptr();
+  hipLaunchKernelGGL(L->fn, dim3(blocks), dim3(L->waves * 64), L->lds_bytes,
+                     (hipStream_t)at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream(),
+                     (const void*)in.data_ptr(),
+                     (const __bf16*)weight.data_ptr(), bias_ptr,
+                     (__bf16*)out.data_ptr(), N);
+  return true;
 }
 
 void register_conv(pybind11::module_& m) {
@@ -260,4 +323,5 @@ void register_conv(pybind11::module_& m) {
         "fused NHWC bf16 MFMA conv fwd (+dequant on u8 input, bias, ReLU)");
   m.def("conv_fwd_supported", &conv_fwd_supported);
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA fragment-map probe");
+  m.def("conv_fwd_variant", &conv_fwd_variant, "tuning-variant conv fwd");
 }
