@@ -120,11 +120,16 @@ class _FusedConvFn(torch.autograd.Function):
     def backward(ctx, gout):
         x, weight, out = ctx.saved_tensors
         stride = ctx.stride
-        gout = gout.contiguous(memory_format=torch.channels_last)
-        gout = gout * (out > 0)  # ReLU was folded into the forward
+        if not gout.is_contiguous(memory_format=torch.channels_last):
+            gout = gout.contiguous(memory_format=torch.channels_last)
+        # fused ReLU mask (one kernel instead of compare+mul)
+        masked = torch.empty_like(gout)
+        hip_ext().relu_mask_bwd(gout, out, masked)
+        gout = masked
         if x.dtype == torch.uint8:
-            xf = x.to(torch.bfloat16)
-            xf = xf / 255.0  # layout (channels_last) is preserved
+            # NHWC u8 -> NHWC bf16: flat dequant on the permuted contiguous view
+            xf = dequant_frames(x.permute(0, 2, 3, 1), torch.bfloat16)
+            xf = xf.permute(0, 3, 1, 2)  # logical NCHW, channels_last memory
         else:
             xf = x
         need_x = ctx.needs_input_grad[0]

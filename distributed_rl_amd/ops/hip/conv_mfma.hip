@@ -177,8 +177,8 @@ ConvLaunch make_launch() {
 static const ConvLaunch kLaunches[] = {
     // Ape-X / R2D2 stack (cfg/ape_x.json:38-51); WAVES/RPW picked by the
     // variant sweep in tools/gpu_conv_tune.py (profiles/)
-    make_launch<84, 84, 4, 8, 8, 4, 32, true, 8, 16>(),
-    make_launch<84, 84, 4, 8, 8, 4, 32, false, 8, 16>(),
+    make_launch<84, 84, 4, 8, 8, 4, 32, true, 8, 32>(),
+    make_launch<84, 84, 4, 8, 8, 4, 32, false, 8, 32>(),
     make_launch<20, 20, 32, 4, 4, 2, 64, false, 8, 16>(),
     make_launch<9, 9, 64, 3, 3, 1, 64, false, 8, 16>(),
     // IMPALA stack (cfg/impala.json:26-39)
