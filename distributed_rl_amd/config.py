@@ -180,9 +180,11 @@ class Config:
 
 
 def cfg_path_for(alg: str, cfg_dir: str = DEFAULT_CFG_DIR) -> str:
-    alg = alg.upper()
-    name = {"APE_X": "ape_x.json", "R2D2": "r2d2.json", "IMPALA": "impala.json"}[alg]
-    return os.path.join(cfg_dir, name)
+    name = f"{alg.lower()}.json"
+    path = os.path.join(cfg_dir, name)
+    if not os.path.exists(path):
+        raise FileNotFoundError(f"no such cfg: {path}")
+    return path
 
 
 def load_config(spec: Optional[str] = None) -> Config:

@@ -208,6 +208,46 @@ class LSTMNET(nn.Module):
         return out
 
 
+class ResidualBlock(nn.Module):
+    def __init__(self, ch: int):
+        super().__init__()
+        self.c1 = nn.Conv2d(ch, ch, 3, padding=1)
+        self.c2 = nn.Conv2d(ch, ch, 3, padding=1)
+
+    def forward(self, x):
+        y = self.c1(torch.relu(x))
+        y = self.c2(torch.relu(y))
+        return x + y
+
+
+class ImpalaResNet(nn.Module):
+    """IMPALA deep residual torso (Espeholt et al. 2018, fig. 3): per section
+    conv3x3 -> maxpool3x3 s2 -> nBlocks residual blocks; final ReLU+flatten.
+    Used by cfg/impala_resnet.json (BASELINE.json config 3: 'IMPALA deep
+    ResNet V-trace, 8x MI355X learner-DP'). Beyond the reference repo's
+    capability (it only ships the shallow torso) — netCat: IMPALA_RESNET."""
+
+    def __init__(self, cfg: Dict[str, Any]):
+        super().__init__()
+        in_ch = int(cfg.get("iSize", 4))
+        channels = list(cfg.get("channels", [16, 32, 32]))
+        n_blocks = int(cfg.get("nBlocks", 2))
+        sections: List[nn.Module] = []
+        ch = in_ch
+        for out_ch in channels:
+            sections.append(nn.Conv2d(ch, out_ch, 3, padding=1))
+            sections.append(nn.MaxPool2d(3, stride=2, padding=1))
+            for _ in range(n_blocks):
+                sections.append(ResidualBlock(out_ch))
+            ch = out_ch
+        self.body = nn.Sequential(*sections)
+
+    def forward(self, x):
+        y = self.body(x)
+        y = torch.relu(y)
+        return torch.flatten(y, 1)
+
+
 class _Arith(nn.Module):
     KIND = "add"
 
@@ -238,6 +278,7 @@ _NETCAT = {
     "ADD": Add,
     "MEAN": Mean,
     "SUBSTRACT": Substract,
+    "IMPALA_RESNET": ImpalaResNet,
 }
 
 
@@ -264,7 +305,8 @@ class BaseAgent(nn.Module):
             kind = str(ncfg["netCat"]).upper()
             if kind not in _NETCAT:
                 raise ValueError(f"unsupported netCat {ncfg['netCat']!r} for node {name}")
-            self.nodes[name] = _NETCAT[kind](ncfg) if kind in ("CNN2D", "MLP", "LSTMNET") else _NETCAT[kind]()
+            needs_cfg = kind in ("CNN2D", "MLP", "LSTMNET", "IMPALA_RESNET")
+            self.nodes[name] = _NETCAT[kind](ncfg) if needs_cfg else _NETCAT[kind]()
             if ncfg.get("output", False):
                 self.output_nodes.append(name)
         if not self.output_nodes:

@@ -82,3 +82,22 @@ def test_impala_learning_signal():
     )
     assert changed
     assert np.isfinite(float(stats["loss"]))
+
+
+def test_impala_resnet_learner_cpu():
+    raw = copy.deepcopy(load_config("impala_resnet").raw)
+    raw["BATCHSIZE"] = 2
+    cfg = Config(raw=raw)
+    learner = ImpalaLearner(cfg, device="cpu", enable_tb=False,
+                            replay_capacity=16)
+    B, T = 4, cfg.unroll_step
+    cols = {
+        "states": torch.randint(0, 255, (B, T + 1, 4, 84, 84), dtype=torch.uint8),
+        "actions": torch.randint(0, 6, (B, T), dtype=torch.int32),
+        "mu": torch.full((B, T), 1 / 6, dtype=torch.float32),
+        "rewards": torch.randn(B, T),
+        "not_done": torch.ones(B),
+    }
+    learner.push_trajectories(cols)
+    stats = learner.step()
+    assert np.isfinite(float(stats["loss"]))

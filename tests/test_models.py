@@ -99,3 +99,15 @@ def test_state_dict_roundtrip():
     b.load_state_dict(a.state_dict())
     for p, q in zip(a.parameters(), b.parameters()):
         assert torch.equal(p, q)
+
+
+def test_impala_resnet_model():
+    cfg = load_config("impala_resnet")
+    net = BaseAgent(cfg.model_info)
+    out = net.forward([torch.rand(3, 4, 84, 84)])[0]
+    assert out.shape == (3, 7)
+    # residual path actually contributes
+    from distributed_rl_amd.models.base_agent import ResidualBlock
+
+    blocks = [m for m in net.modules() if isinstance(m, ResidualBlock)]
+    assert len(blocks) == 6  # 3 sections x 2 blocks
