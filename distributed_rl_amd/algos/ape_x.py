@@ -45,6 +45,7 @@ class ApexLearner(LearnerBase):
                  batch_size: Optional[int] = None,
                  replay_capacity: Optional[int] = None,
                  replay_device: Optional[str] = None,
+                 replay_state_dtype: Optional[torch.dtype] = None,
                  enable_tb: bool = True, run_root: str = "."):
         super().__init__(cfg, device, rank, world_size, run_root=run_root,
                          enable_tb=enable_tb)
@@ -94,9 +95,13 @@ class ApexLearner(LearnerBase):
         # uint8 replay column directly (fused dequant in conv_mfma.hip) —
         # no standalone dequant pass, no layout transposes.
         self._nhwc = self.device.type == "cuda"
+        # Replay frame storage dtype: uint8 (default; 4x denser than fp16)
+        # or float16 (BASELINE config 5's "fp16 replay compression" option —
+        # frames stored pre-scaled to [0,1]).
+        self.state_dtype = replay_state_dtype or torch.uint8
         schema = (
-            make_apex_schema(frame_shape=(84, 84, 4)) if self._nhwc
-            else make_apex_schema()
+            make_apex_schema(frame_shape=(84, 84, 4), state_dtype=self.state_dtype)
+            if self._nhwc else make_apex_schema(state_dtype=self.state_dtype)
         )
         self.replay = make_per(cap, schema, device=rdev)
         self.transport = transport
@@ -143,11 +148,15 @@ class ApexLearner(LearnerBase):
         return n
 
     def push_experience(self, cols: Dict[str, torch.Tensor], prio: torch.Tensor):
-        """Direct (in-process) push; frames arrive NCHW (wire format)."""
-        if self._nhwc:
-            cols = dict(cols)
-            for k in ("state", "next_state"):
-                cols[k] = cols[k].permute(0, 2, 3, 1).contiguous()
+        """Direct (in-process) push; frames arrive NCHW uint8 (wire format)."""
+        cols = dict(cols)
+        for k in ("state", "next_state"):
+            v = cols[k]
+            if self._nhwc:
+                v = v.permute(0, 2, 3, 1)
+            if self.state_dtype != torch.uint8 and v.dtype == torch.uint8:
+                v = v.to(self.state_dtype) / 255.0
+            cols[k] = v.contiguous()
         self.replay.push(cols, prio)
 
     # ------------------------------------------------------------------
@@ -156,12 +165,19 @@ class ApexLearner(LearnerBase):
     def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
         cuda = self.device.type == "cuda"
         if cuda:
-            # NHWC uint8 straight into the fused conv stack (dequant fused)
+            # NHWC frames straight into the fused conv stack (u8: dequant
+            # fused into conv1; fp16-compressed: one cast to bf16)
             s = data["state"].permute(0, 3, 1, 2)
             sp = data["next_state"].permute(0, 3, 1, 2)
-        else:
+            if s.dtype == torch.float16:
+                s = s.to(torch.bfloat16)
+                sp = sp.to(torch.bfloat16)
+        elif data["state"].dtype == torch.uint8:
             s = ops.dequant_frames(data["state"], torch.float32)
             sp = ops.dequant_frames(data["next_state"], torch.float32)
+        else:
+            s = data["state"].float()
+            sp = data["next_state"].float()
         actions = data["action"].to(self.device).long()
         rewards = data["reward"].to(self.device)
         dones = data["done"].to(self.device)
@@ -438,7 +454,8 @@ class ApexPlayer:
     PUSH_BATCH = 16
 
     def __init__(self, cfg: Config, idx: int, transport, env=None,
-                 env_kind: str = "auto", seed: Optional[int] = None):
+                 env_kind: str = "auto", seed: Optional[int] = None,
+                 max_staleness: Optional[int] = None):
         from ..actors.env import make_env
 
         self.cfg = cfg
@@ -446,6 +463,15 @@ class ApexPlayer:
         self.transport = transport
         self.env = env or make_env(env_kind, seed=seed if seed is not None else idx)
         self.device = torch.device(cfg.actor_device)
+        # bounded staleness (SURVEY §5.3 gap): if the learner's published
+        # `count` hasn't advanced for this many env steps, the actor blocks
+        # instead of generating arbitrarily off-policy data. None = the
+        # reference's behavior (proceed forever on stale weights).
+        self.max_staleness = (
+            max_staleness if max_staleness is not None
+            else cfg.get("MAX_STALENESS_STEPS")
+        )
+        self._steps_since_fresh = 0
         self.model = BaseAgent(cfg.model_info).to(self.device).eval()
         self.target = BaseAgent(cfg.model_info).to(self.device).eval()
         n_actors = max(cfg.num_actors, 2)
@@ -513,6 +539,18 @@ class ApexPlayer:
         if "target_state_dict" in payload:
             self.target.load_state_dict(payload["target_state_dict"])
         self.weight_version = count
+        self._steps_since_fresh = 0
+
+    def _staleness_gate(self):
+        if self.max_staleness is None:
+            return
+        self._steps_since_fresh += 1
+        while self._steps_since_fresh > int(self.max_staleness):
+            time.sleep(0.05)
+            before = self.weight_version
+            self.pull_weights()
+            if self.weight_version != before:
+                break
 
     # push via transport: ShmTransport actor-side adapter provides .push
     def run(self, max_env_steps: int = 1_000_000):
@@ -534,6 +572,7 @@ class ApexPlayer:
             self.env_steps += 1
             if self.env_steps % ACTOR_PULL_EVERY == 0:
                 self.pull_weights()
+            self._staleness_gate()
             if done:
                 # eval telemetry gate per Player.py:272-277
                 if self.eps < 0.05 or True:
