@@ -17,7 +17,7 @@ from .transport import ActorEndpoint, RecordCodec, TransportSession
 
 
 def _actor_main(cfg_spec: str, idx: int, transport_dir: str,
-                max_env_steps: int, env_kind: str):
+                max_env_steps: int, env_kind: str, tcp: str = ""):
     # actors are CPU-only: keep torch single-threaded per actor
     os.environ.setdefault("OMP_NUM_THREADS", "1")
     import torch
@@ -26,8 +26,16 @@ def _actor_main(cfg_spec: str, idx: int, transport_dir: str,
     cfg = load_config(cfg_spec)
     schema, with_prio = get_wire_schema(cfg)
     codec = RecordCodec(schema, with_priority=with_prio)
-    session = TransportSession(transport_dir, codec, num_rings=0, create=False)
-    endpoint = ActorEndpoint(session, idx)
+    session = None
+    if tcp:
+        from .tcp_transport import TcpActorEndpoint
+
+        host, port = tcp.rsplit(":", 1)
+        endpoint = TcpActorEndpoint(host, int(port), codec, idx=idx)
+    else:
+        session = TransportSession(transport_dir, codec, num_rings=0,
+                                   create=False)
+        endpoint = ActorEndpoint(session, idx)
     player = get_player_cls(cfg.alg)(cfg, idx=idx, transport=endpoint,
                                      env_kind=env_kind)
     try:
@@ -35,7 +43,8 @@ def _actor_main(cfg_spec: str, idx: int, transport_dir: str,
     except KeyboardInterrupt:
         pass
     finally:
-        session.close()
+        if session is not None:
+            session.close()
 
 
 class ActorFleet:
@@ -44,11 +53,13 @@ class ActorFleet:
 
     def __init__(self, cfg_spec: str, num_actors: int, transport_dir: str,
                  start_idx: int = 0, env_kind: str = "auto",
-                 max_env_steps: int = 1 << 60, respawn_on_exit: bool = True):
+                 max_env_steps: int = 1 << 60, respawn_on_exit: bool = True,
+                 tcp: str = ""):
         self.cfg_spec = cfg_spec
         self.num_actors = num_actors
         self.start_idx = start_idx
         self.transport_dir = transport_dir
+        self.tcp = tcp
         self.env_kind = env_kind
         self.max_env_steps = max_env_steps
         self.respawn = respawn_on_exit
@@ -60,7 +71,7 @@ class ActorFleet:
         p = self.ctx.Process(
             target=_actor_main,
             args=(self.cfg_spec, idx, self.transport_dir, self.max_env_steps,
-                  self.env_kind),
+                  self.env_kind, self.tcp),
             daemon=True,
             name=f"drl-actor-{idx}",
         )

@@ -32,23 +32,26 @@ def main():
                     help="auto | atari | synthetic")
     ap.add_argument("--max-env-steps", type=int, default=1 << 60)
     ap.add_argument("--no-respawn", action="store_true")
+    ap.add_argument("--tcp", default="",
+                    help="learner host:port for TCP mode (multi-host)")
     args = ap.parse_args()
 
     cfg = load_config(args.cfg)
     tdir = args.transport_dir or cfg.transport_dir
     n = args.num_worker or cfg.num_actors
-    # wait for the learner's session manifest
-    manifest = os.path.join(tdir, "session.json")
-    t0 = time.time()
-    while not os.path.exists(manifest):
-        if time.time() - t0 > 300:
-            raise TimeoutError(f"no learner session at {manifest}")
-        time.sleep(0.5)
+    if not args.tcp:
+        # wait for the learner's session manifest
+        manifest = os.path.join(tdir, "session.json")
+        t0 = time.time()
+        while not os.path.exists(manifest):
+            if time.time() - t0 > 300:
+                raise TimeoutError(f"no learner session at {manifest}")
+            time.sleep(0.5)
 
     fleet = ActorFleet(args.cfg or cfg.alg.lower(), n, tdir,
                        start_idx=args.start_idx, env_kind=args.env,
                        max_env_steps=args.max_env_steps,
-                       respawn_on_exit=not args.no_respawn)
+                       respawn_on_exit=not args.no_respawn, tcp=args.tcp)
     fleet.start()
     print(f"[run_actor] {n} actors running (start_idx={args.start_idx})",
           flush=True)

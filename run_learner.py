@@ -41,6 +41,9 @@ def main():
     ap.add_argument("--resume", default=None, help="checkpoint to resume from")
     ap.add_argument("--no-transport", action="store_true",
                     help="run without an actor session (debug)")
+    ap.add_argument("--tcp-port", type=int, default=None,
+                    help="serve actors over TCP on this port (multi-host "
+                         "mode) instead of shared-memory rings")
     args = ap.parse_args()
 
     rank, local_rank, world = init_distributed()
@@ -53,20 +56,28 @@ def main():
 
     transport = None
     session = None
+    tcp_srv = None
     if not args.no_transport:
         schema, with_prio = get_wire_schema(cfg)
         codec = RecordCodec(schema, with_priority=with_prio)
-        tdir = args.transport_dir or cfg.transport_dir
-        n_act = args.num_actors or cfg.num_actors
-        if rank == 0:
-            session = TransportSession(tdir, codec, num_rings=n_act,
-                                       ring_slots=args.ring_slots, create=True)
-        if world > 1:
-            torch.distributed.barrier()
-        if rank != 0:
-            session = TransportSession(tdir, codec, num_rings=n_act,
-                                       create=False)
-        transport = LearnerEndpoint(session, rank=rank, world_size=world)
+        if args.tcp_port is not None:
+            from distributed_rl_amd.actors.tcp_transport import TcpTransportServer
+
+            tcp_srv = TcpTransportServer(codec, port=args.tcp_port + rank).start()
+            transport = tcp_srv.endpoint()
+        else:
+            tdir = args.transport_dir or cfg.transport_dir
+            n_act = args.num_actors or cfg.num_actors
+            if rank == 0:
+                session = TransportSession(tdir, codec, num_rings=n_act,
+                                           ring_slots=args.ring_slots,
+                                           create=True)
+            if world > 1:
+                torch.distributed.barrier()
+            if rank != 0:
+                session = TransportSession(tdir, codec, num_rings=n_act,
+                                           create=False)
+            transport = LearnerEndpoint(session, rank=rank, world_size=world)
 
     learner = get_learner_cls(cfg.alg)(
         cfg, device=device, rank=rank, world_size=world, transport=transport,
@@ -79,6 +90,8 @@ def main():
     finally:
         if session is not None:
             session.close()
+        if tcp_srv is not None:
+            tcp_srv.stop()
         if world > 1:
             torch.distributed.destroy_process_group()
 
