@@ -126,19 +126,43 @@ class _FusedConvFn(torch.autograd.Function):
         masked = torch.empty_like(gout)
         hip_ext().relu_mask_bwd(gout, out, masked)
         gout = masked
-        if x.dtype == torch.uint8:
-            # NHWC u8 -> NHWC bf16: flat dequant on the permuted contiguous view
-            xf = dequant_frames(x.permute(0, 2, 3, 1), torch.bfloat16)
-            xf = xf.permute(0, 3, 1, 2)  # logical NCHW, channels_last memory
-        else:
-            xf = x
         need_x = ctx.needs_input_grad[0]
-        gi, gw, gb = torch.ops.aten.convolution_backward(
-            gout, xf, weight,
-            [weight.shape[0]] if ctx.has_bias else None,
-            [stride, stride], [0, 0], [1, 1], False, [0, 0], 1,
-            [need_x, ctx.needs_input_grad[1], ctx.has_bias and ctx.needs_input_grad[2]],
+        need_w = ctx.needs_input_grad[1]
+        need_b = ctx.has_bias and ctx.needs_input_grad[2]
+        COUT, C, KH, KW = weight.shape
+        ext = hip_ext()
+        use_own_wrw = need_w and bool(
+            ext.conv_wrw_supported(x.shape[2], x.shape[3], C, KH, KW, stride,
+                                   COUT, x.dtype == torch.uint8)
         )
+        gw = gb = None
+        if use_own_wrw:
+            # hand-written MFMA weight-grad (conv_mfma.hip): fp32 workspace
+            # accumulated by atomics, cast to the bf16 channels_last grad.
+            ws = torch.zeros(COUT, KH * KW * C, dtype=torch.float32,
+                             device=x.device)
+            ext.conv_wrw(x, gout, ws, stride)
+            gw = ws.view(COUT, KH, KW, C).permute(0, 3, 1, 2).to(torch.bfloat16)
+            if need_b:
+                gb = gout.sum(dim=(0, 2, 3)).to(torch.bfloat16)
+        if need_x or (need_w and not use_own_wrw):
+            if x.dtype == torch.uint8:
+                # NHWC u8 -> bf16 for the aten path (own wrw reads u8 directly)
+                xf = dequant_frames(x.permute(0, 2, 3, 1), torch.bfloat16)
+                xf = xf.permute(0, 3, 1, 2)
+            else:
+                xf = x
+            gi, gw2, gb2 = torch.ops.aten.convolution_backward(
+                gout, xf, weight,
+                [COUT] if ctx.has_bias else None,
+                [stride, stride], [0, 0], [1, 1], False, [0, 0], 1,
+                [need_x, need_w and not use_own_wrw,
+                 need_b and not use_own_wrw],
+            )
+            if gw is None:
+                gw, gb = gw2, gb2
+        else:
+            gi = None
         return (gi if need_x else None), gw, gb, None
 
 

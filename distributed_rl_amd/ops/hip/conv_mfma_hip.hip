@@ -318,10 +318,222 @@ bool conv_fwd_variant(torch::Tensor in, torch::Tensor weight,
   return true;
 }
 
+void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
+              int64_t stride);  // defined below
+bool conv_wrw_supported(int64_t H, int64_t W, int64_t C, int64_t KH,
+                        int64_t KW, int64_t S, int64_t COUT, bool u8);
+
 void register_conv(pybind11::module_& m) {
   m.def("conv_fwd", &conv_fwd,
         "fused NHWC bf16 MFMA conv fwd (+dequant on u8 input, bias, ReLU)");
   m.def("conv_fwd_supported", &conv_fwd_supported);
   m.def("mfma_probe", &mfma_probe, "16x16x32 bf16 MFMA fragment-map probe");
   m.def("conv_fwd_variant", &conv_fwd_variant, "tuning-variant conv fwd");
+  m.def("conv_wrw", &conv_wrw, "MFMA conv weight-grad (fp32 workspace)");
+  m.def("conv_wrw_supported", &conv_wrw_supported);
+}
+
+// ===========================================================================
+// Weight-grad (wrw) kernel: gw[COUT][K] = sum_m gout^T[COUT][m] * im2col[m][K]
+// — the tall-skinny reduction GEMM of conv backward (MIOpen's igemm_wrw was
+// the single biggest post-fusion cost, profiles/). The reduction dim is the
+// sample dim m, so both operands need m-major fragments: we stage 32-sample
+// chunks into LDS TRANSPOSED (scalar b16 writes, padded rows) so fragment
+// reads are clean b128s, MFMA accumulates COUT x 64-kelem tiles, and each
+// block atomicAdds its partial into an fp32 workspace.
+//   grid = (K/64 ktiles, MB m-slices); block = 4 waves; wave w owns the
+//   16-kelem column block w.
+// ===========================================================================
+
+namespace {
+
+constexpr int kPadM = 8;  // pad the 32-wide m dim to break b128 bank groups
+
+template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8IN>
+__global__ __launch_bounds__(256) void conv_wrw_kernel(
+    const void* __restrict__ in_v,      // (N,H,W,C) NHWC u8/bf16
+    const __bf16* __restrict__ gout,    // (M, COUT) = NHWC grad (relu-masked)
+    float* __restrict__ gw_ws,          // (COUT, K) fp32, pre-zeroed
+    int batch, int mblocks) {
+  using G = ConvGeom<H, W, C, KH, KW, S, COUT, U8IN>;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  constexpr int MROW = 32 + kPadM;
+  __bf16* a_t = reinterpret_cast<__bf16*>(smem);          // [64][MROW]
+  __bf16* g_t = a_t + 64 * MROW;                          // [COUT][MROW]
+
+  const int ktile = blockIdx.x;
+  const int mb = blockIdx.y;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int M = batch * G::P * G::Q;
+  const int n_chunks = (M + 31) / 32;
+
+  constexpr int CBLK = COUT / 16;
+  f32x4 acc[CBLK];
+#pragma unroll
+  for (int cb = 0; cb < CBLK; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
+
+  // thread's staging assignment (one 8-element piece each)
+  const int sm = tid >> 3;            // sample row 0..31
+  const int sk8 = (tid & 7) * 8;      // kelem piece base 0..56
+
+  for (int mc = mb; mc < n_chunks; mc += mblocks) {
+    const int m0 = mc * 32;
+    // ---- stage im2col chunk -> a_t[kelem][m] (transposed scalar writes)
+    {
+      const int m = m0 + sm;
+      const int mcl = m < M ? m : M - 1;
+      const int n = mcl / (G::P * G::Q);
+      const int rem = mcl - n * (G::P * G::Q);
+      const int p = rem / G::Q;
+      const int q = rem - p * G::Q;
+      const int kelem = ktile * 64 + sk8;
+      const int dy = kelem / G::ROWC;
+      const int dx = kelem - dy * G::ROWC;
+      const int64_t goff =
+          ((int64_t)n * H + p * S + dy) * (W * C) + q * S * C + dx;
+      __bf16 vals[8];
+      if constexpr (U8IN) {
+        const uint8_t* src = reinterpret_cast<const uint8_t*>(in_v) + goff;
+        uint2 raw = *reinterpret_cast<const uint2*>(src);
+        const float inv255 = 1.0f / 255.0f;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          unsigned byte =
+              (j < 4 ? raw.x >> (8 * j) : raw.y >> (8 * (j - 4))) & 0xFF;
+          vals[j] = (__bf16)(byte * inv255);
+        }
+      } else {
+        bf16x8 v = *reinterpret_cast<const bf16x8*>(
+            reinterpret_cast<const __bf16*>(in_v) + goff);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) vals[j] = v[j];
+      }
+      const bool valid = m < M;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        a_t[(sk8 + j) * MROW + sm] = valid ? vals[j] : (__bf16)0.0f;
+    }
+    // ---- stage gout chunk -> g_t[cout][m] (transposed, zero-padded)
+    {
+      constexpr int PIECES = COUT / 8;       // 8-cout pieces per sample row
+      // 256 threads cover 32 * PIECES pieces; PIECES in {2,4,8}
+      for (int piece = tid; piece < 32 * PIECES; piece += 256) {
+        const int m = m0 + piece / PIECES;
+        const int c0 = (piece % PIECES) * 8;
+        __bf16 vals[8];
+        if (m < M) {
+          bf16x8 v = *reinterpret_cast<const bf16x8*>(
+              gout + (int64_t)m * COUT + c0);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vals[j] = v[j];
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vals[j] = (__bf16)0.0f;
+        }
+        const int sm2 = piece / PIECES;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          g_t[(c0 + j) * MROW + sm2] = vals[j];
+      }
+    }
+    __syncthreads();
+    // ---- fragments + MFMA: wave w owns kelem block [w*16, w*16+16)
+    const int kcol = wave * 16 + (lane & 15);
+    const int mpart = (lane >> 4) * 8;
+    bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(a_t + kcol * MROW + mpart);
+#pragma unroll
+    for (int cb = 0; cb < CBLK; ++cb) {
+      bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
+          g_t + (cb * 16 + (lane & 15)) * MROW + mpart);
+      acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[cb],
+                                                        0, 0, 0);
+    }
+    __syncthreads();
+  }
+
+  // ---- epilogue: atomic-accumulate partials into the fp32 workspace
+  const int kout = ktile * 64 + wave * 16 + (lane & 15);
+#pragma unroll
+  for (int cb = 0; cb < CBLK; ++cb)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      const int cout = cb * 16 + (lane >> 4) * 4 + r;
+      atomicAdd(&gw_ws[(int64_t)cout * G::K + kout], acc[cb][r]);
+    }
+}
+
+struct WrwLaunch {
+  int H, W, C, KH, KW, S, COUT;
+  bool u8;
+  void (*fn)(const void*, const __bf16*, float*, int, int);
+  int lds_bytes;
+  int ktiles;
+};
+
+template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8>
+WrwLaunch make_wrw() {
+  using G = ConvGeom<H, W, C, KH, KW, S, COUT, U8>;
+  static_assert(G::K % 64 == 0, "wrw needs K % 64 == 0");
+  return WrwLaunch{H, W, C, KH, KW, S, COUT, U8,
+                   conv_wrw_kernel<H, W, C, KH, KW, S, COUT, U8>,
+                   (int)((64 + COUT) * (32 + kPadM) * sizeof(__bf16)),
+                   G::K / 64};
+}
+
+static const WrwLaunch kWrwLaunches[] = {
+    make_wrw<84, 84, 4, 8, 8, 4, 32, true>(),
+    make_wrw<84, 84, 4, 8, 8, 4, 32, false>(),
+    make_wrw<20, 20, 32, 4, 4, 2, 64, false>(),
+    make_wrw<9, 9, 64, 3, 3, 1, 64, false>(),
+    make_wrw<84, 84, 4, 8, 8, 4, 16, true>(),
+    make_wrw<84, 84, 4, 8, 8, 4, 16, false>(),
+    make_wrw<20, 20, 16, 4, 4, 2, 32, false>(),
+};
+
+}  // namespace
+
+bool conv_wrw_supported(int64_t H, int64_t W, int64_t C, int64_t KH, int64_t KW,
+                        int64_t S, int64_t COUT, bool u8) {
+  for (const auto& l : kWrwLaunches)
+    if (l.H == H && l.W == W && l.C == C && l.KH == KH && l.KW == KW &&
+        l.S == S && l.COUT == COUT && l.u8 == u8)
+      return true;
+  return false;
+}
+
+// in: (N,C,H,W) channels_last u8/bf16; gout: (N,COUT,P,Q) channels_last bf16
+// (already relu-masked); gw_ws: (COUT, KH*KW*C) fp32 pre-zeroed.
+void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
+              int64_t stride) {
+  TORCH_CHECK(in.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(gout.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(gw_ws.scalar_type() == torch::kFloat32 && gw_ws.is_contiguous());
+  const int N = (int)in.size(0), C = (int)in.size(1), H = (int)in.size(2),
+            W = (int)in.size(3);
+  const int COUT = (int)gout.size(1);
+  const bool u8 = in.scalar_type() == torch::kUInt8;
+  const WrwLaunch* L = nullptr;
+  for (const auto& l : kWrwLaunches) {
+    int KHl = l.KH, KWl = l.KW;
+    if (l.H == H && l.W == W && l.C == C && l.S == (int)stride &&
+        l.COUT == COUT && l.u8 == u8 &&
+        (H - KHl) / (int)stride + 1 == (int)gout.size(2) &&
+        (W - KWl) / (int)stride + 1 == (int)gout.size(3)) {
+      L = &l;
+      break;
+    }
+  }
+  TORCH_CHECK(L, "no wrw kernel for this geometry");
+  const int P = (int)gout.size(2), Q = (int)gout.size(3);
+  const int M = N * P * Q;
+  int mb = 512 / L->ktiles;
+  const int n_chunks = (M + 31) / 32;
+  if (mb > n_chunks) mb = n_chunks;
+  if (mb < 1) mb = 1;
+  hipLaunchKernelGGL(L->fn, dim3(L->ktiles, mb), dim3(256), L->lds_bytes,
+                     (hipStream_t)at::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream(),
+                     (const void*)in.data_ptr(), (const __bf16*)gout.data_ptr(),
+                     gw_ws.data_ptr<float>(), N, mb);
 }

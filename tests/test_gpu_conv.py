@@ -130,3 +130,48 @@ def test_model_fused_path_active(ext):
     err = (y_fused.float() - y_eager.float()).abs().max().item()
     scale = y_eager.float().abs().max().item() + 1e-6
     assert err / scale < 3e-2
+
+
+@pytest.mark.parametrize("shape", SHAPES)
+def test_conv_wrw_vs_aten(ext, shape):
+    H, W, C, KH, KW, S, COUT = shape
+    torch.manual_seed(5)
+    N = 37
+    P, Q = (H - KH) // S + 1, (W - KW) // S + 1
+    x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).to(
+        memory_format=torch.channels_last)
+    gout = torch.randn(N, COUT, P, Q, device=DEV).to(torch.bfloat16).to(
+        memory_format=torch.channels_last)
+    ws = torch.zeros(COUT, KH * KW * C, dtype=torch.float32, device=DEV)
+    ext.conv_wrw(x, gout, ws, S)
+    torch.cuda.synchronize()
+    gw = ws.view(COUT, KH, KW, C).permute(0, 3, 1, 2)
+    _, gw_ref, _ = torch.ops.aten.convolution_backward(
+        gout, x, torch.empty(COUT, C, KH, KW, device=DEV, dtype=torch.bfloat16
+                             ).to(memory_format=torch.channels_last),
+        None, [S, S], [0, 0], [1, 1], False, [0, 0], 1, [False, True, False])
+    scale = gw_ref.float().abs().max().item() + 1e-6
+    err = (gw - gw_ref.float()).abs().max().item()
+    assert err / scale < 2e-2, f"{shape}: {err} vs {scale}"
+
+
+def test_conv_wrw_u8_input(ext):
+    torch.manual_seed(6)
+    N, C, H, W, COUT, S, KH = 21, 4, 84, 84, 32, 4, 8
+    P = Q = 20
+    x = torch.randint(0, 256, (N, C, H, W), dtype=torch.uint8, device=DEV).to(
+        memory_format=torch.channels_last)
+    gout = torch.randn(N, COUT, P, Q, device=DEV).to(torch.bfloat16).to(
+        memory_format=torch.channels_last)
+    ws = torch.zeros(COUT, KH * KH * C, dtype=torch.float32, device=DEV)
+    ext.conv_wrw(x, gout, ws, S)
+    torch.cuda.synchronize()
+    gw = ws.view(COUT, KH, KH, C).permute(0, 3, 1, 2)
+    xf = (x.float() / 255.0).to(torch.bfloat16)
+    _, gw_ref, _ = torch.ops.aten.convolution_backward(
+        gout, xf, torch.empty(COUT, C, KH, KH, device=DEV, dtype=torch.bfloat16
+                              ).to(memory_format=torch.channels_last),
+        None, [S, S], [0, 0], [1, 1], False, [0, 0], 1, [False, True, False])
+    scale = gw_ref.float().abs().max().item() + 1e-6
+    err = (gw - gw_ref.float()).abs().max().item()
+    assert err / scale < 2e-2, f"{err} vs {scale}"

@@ -64,3 +64,38 @@ def main():
 
 if __name__ == "__main__":
     main()
+    bench_wrw()
+
+
+def bench_wrw():
+    import torch.nn.functional as F
+    from distributed_rl_amd.ops import hip_ext
+
+    ext = hip_ext(required=True)
+    N = 512
+    for name, H, W, C, KH, KW, S, COUT, u8, _ in SHAPES:
+        P, Q = (H - KH) // S + 1, (W - KW) // S + 1
+        flops = 2.0 * N * P * Q * COUT * KH * KW * C
+        if u8:
+            x = torch.randint(0, 256, (N, C, H, W), dtype=torch.uint8,
+                              device=DEV).to(memory_format=torch.channels_last)
+        else:
+            x = torch.randn(N, C, H, W, device=DEV).to(torch.bfloat16).to(
+                memory_format=torch.channels_last)
+        gout = torch.randn(N, COUT, P, Q, device=DEV).to(torch.bfloat16).to(
+            memory_format=torch.channels_last)
+        ws = torch.zeros(COUT, KH * KW * C, dtype=torch.float32, device=DEV)
+        us = timeit(lambda: (ws.zero_(), ext.conv_wrw(x, gout, ws, S)))
+        line = f"wrw {name}: {us:7.1f} us  {flops/us/1e6:7.1f} TF"
+        if not u8:
+            w = torch.empty(COUT, C, KH, KW, device=DEV, dtype=torch.bfloat16
+                            ).to(memory_format=torch.channels_last)
+            ref = timeit(lambda: torch.ops.aten.convolution_backward(
+                gout, x, w, None, [S, S], [0, 0], [1, 1], False, [0, 0], 1,
+                [False, True, False]))
+            line += f"   (miopen {ref:7.1f} us {flops/ref/1e6:6.1f} TF)"
+        print(line)
+
+
+if __name__ == "__main__" or True:
+    pass
