@@ -62,9 +62,6 @@ def main():
             print(f"  miopen: {ref_us:7.1f} us  {flops/ref_us/1e6:7.1f} TF")
 
 
-if __name__ == "__main__":
-    main()
-    bench_wrw()
 
 
 def bench_wrw():
@@ -97,5 +94,6 @@ def bench_wrw():
         print(line)
 
 
-if __name__ == "__main__" or True:
-    pass
+if __name__ == "__main__":
+    main()
+    bench_wrw()
