@@ -322,6 +322,7 @@ void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
               int64_t stride);  // defined below
 bool conv_wrw_supported(int64_t H, int64_t W, int64_t C, int64_t KH,
                         int64_t KW, int64_t S, int64_t COUT, bool u8);
+void tr16_probe(torch::Tensor out, int64_t mode);
 
 void register_conv(pybind11::module_& m) {
   m.def("conv_fwd", &conv_fwd,
@@ -331,6 +332,7 @@ void register_conv(pybind11::module_& m) {
   m.def("conv_fwd_variant", &conv_fwd_variant, "tuning-variant conv fwd");
   m.def("conv_wrw", &conv_wrw, "MFMA conv weight-grad (fp32 workspace)");
   m.def("conv_wrw_supported", &conv_wrw_supported);
+  m.def("tr16_probe", &tr16_probe);
 }
 
 // ===========================================================================
@@ -347,7 +349,31 @@ void register_conv(pybind11::module_& m) {
 
 namespace {
 
-constexpr int kPadM = 8;  // pad the 32-wide m dim to break b128 bank groups
+constexpr int kPadM = 8;  // row pad (elements) for the staged tiles
+
+typedef __bf16 v4bf_w __attribute__((ext_vector_type(4)));
+
+// ds_read_tr16_b64 fragment read from a PLAIN row-major [32][NCOL+pad] tile:
+// the 16-lane group transposes the 64 elements its lanes address, so with
+// per-lane address base + (g*8 + (r>>2))*stride + 4*(r&3)  (g = lane>>4,
+// r = lane&15) the result is exactly the MFMA operand fragment
+// lane l -> elems j: tile[m = g*8 + j][col16 = r]  (j 0..3; second read at
+// +4 rows gives j 4..7). Mapping verified on-GPU by tr16_probe (mode 1).
+template <int STRIDE>
+__device__ __forceinline__ bf16x8 tr_frag(const __bf16* tile, int lane,
+                                          int colblk) {
+  const int g = lane >> 4, r = lane & 15;
+  const __bf16* base =
+      tile + (g * 8 + (r >> 2)) * STRIDE + colblk * 16 + 4 * (r & 3);
+  auto* p0 = (__attribute__((address_space(3))) v4bf_w*)base;
+  auto* p1 = (__attribute__((address_space(3))) v4bf_w*)(base + 4 * STRIDE);
+  v4bf_w lo = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p0);
+  v4bf_w hi = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+  bf16x8 out;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) { out[j] = lo[j]; out[j + 4] = hi[j]; }
+  return out;
+}
 
 template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8IN>
 __global__ __launch_bounds__(256) void conv_wrw_kernel(
@@ -357,9 +383,10 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
     int batch, int mblocks) {
   using G = ConvGeom<H, W, C, KH, KW, S, COUT, U8IN>;
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  constexpr int MROW = 32 + kPadM;
-  __bf16* a_t = reinterpret_cast<__bf16*>(smem);          // [64][MROW]
-  __bf16* g_t = a_t + 64 * MROW;                          // [COUT][MROW]
+  constexpr int ASTRIDE = 64 + kPadM;
+  constexpr int GSTRIDE = COUT + kPadM;
+  __bf16* a_t = reinterpret_cast<__bf16*>(smem);          // [32][ASTRIDE]
+  __bf16* g_t = a_t + 32 * ASTRIDE;                       // [32][GSTRIDE]
 
   const int ktile = blockIdx.x;
   const int mb = blockIdx.y;
@@ -374,13 +401,13 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
 #pragma unroll
   for (int cb = 0; cb < CBLK; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
 
-  // thread's staging assignment (one 8-element piece each)
+  // staging assignment: 8-elem pieces, plain row-major destination
   const int sm = tid >> 3;            // sample row 0..31
   const int sk8 = (tid & 7) * 8;      // kelem piece base 0..56
 
   for (int mc = mb; mc < n_chunks; mc += mblocks) {
     const int m0 = mc * 32;
-    // ---- stage im2col chunk -> a_t[kelem][m] (transposed scalar writes)
+    // ---- stage im2col chunk -> a_t[m][kelem] (coalesced b128 writes)
     {
       const int m = m0 + sm;
       const int mcl = m < M ? m : M - 1;
@@ -393,7 +420,7 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
       const int dx = kelem - dy * G::ROWC;
       const int64_t goff =
           ((int64_t)n * H + p * S + dy) * (W * C) + q * S * C + dx;
-      __bf16 vals[8];
+      bf16x8 v;
       if constexpr (U8IN) {
         const uint8_t* src = reinterpret_cast<const uint8_t*>(in_v) + goff;
         uint2 raw = *reinterpret_cast<const uint2*>(src);
@@ -402,51 +429,40 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
         for (int j = 0; j < 8; ++j) {
           unsigned byte =
               (j < 4 ? raw.x >> (8 * j) : raw.y >> (8 * (j - 4))) & 0xFF;
-          vals[j] = (__bf16)(byte * inv255);
+          v[j] = (__bf16)(byte * inv255);
         }
       } else {
-        bf16x8 v = *reinterpret_cast<const bf16x8*>(
+        v = *reinterpret_cast<const bf16x8*>(
             reinterpret_cast<const __bf16*>(in_v) + goff);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) vals[j] = v[j];
       }
-      const bool valid = m < M;
+      if (m >= M) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j)
-        a_t[(sk8 + j) * MROW + sm] = valid ? vals[j] : (__bf16)0.0f;
+        for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
+      }
+      *reinterpret_cast<bf16x8*>(a_t + sm * ASTRIDE + sk8) = v;
     }
-    // ---- stage gout chunk -> g_t[cout][m] (transposed, zero-padded)
+    // ---- stage gout chunk -> g_t[m][cout] (coalesced, zero-padded)
     {
-      constexpr int PIECES = COUT / 8;       // 8-cout pieces per sample row
-      // 256 threads cover 32 * PIECES pieces; PIECES in {2,4,8}
+      constexpr int PIECES = COUT / 8;
       for (int piece = tid; piece < 32 * PIECES; piece += 256) {
         const int m = m0 + piece / PIECES;
         const int c0 = (piece % PIECES) * 8;
-        __bf16 vals[8];
+        bf16x8 v;
         if (m < M) {
-          bf16x8 v = *reinterpret_cast<const bf16x8*>(
-              gout + (int64_t)m * COUT + c0);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) vals[j] = v[j];
+          v = *reinterpret_cast<const bf16x8*>(gout + (int64_t)m * COUT + c0);
         } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) vals[j] = (__bf16)0.0f;
+          for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
         }
-        const int sm2 = piece / PIECES;
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          g_t[(c0 + j) * MROW + sm2] = vals[j];
+        *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
       }
     }
     __syncthreads();
-    // ---- fragments + MFMA: wave w owns kelem block [w*16, w*16+16)
-    const int kcol = wave * 16 + (lane & 15);
-    const int mpart = (lane >> 4) * 8;
-    bf16x8 bfrag = *reinterpret_cast<const bf16x8*>(a_t + kcol * MROW + mpart);
+    // ---- fragments via tr16 reads + MFMA: wave w owns kelem block w
+    bf16x8 bfrag = tr_frag<ASTRIDE>(a_t, lane, wave);
 #pragma unroll
     for (int cb = 0; cb < CBLK; ++cb) {
-      bf16x8 afrag = *reinterpret_cast<const bf16x8*>(
-          g_t + (cb * 16 + (lane & 15)) * MROW + mpart);
+      bf16x8 afrag = tr_frag<GSTRIDE>(g_t, lane, cb);
       acc[cb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(afrag, bfrag, acc[cb],
                                                         0, 0, 0);
     }
@@ -478,7 +494,7 @@ WrwLaunch make_wrw() {
   static_assert(G::K % 64 == 0, "wrw needs K % 64 == 0");
   return WrwLaunch{H, W, C, KH, KW, S, COUT, U8,
                    conv_wrw_kernel<H, W, C, KH, KW, S, COUT, U8>,
-                   (int)((64 + COUT) * (32 + kPadM) * sizeof(__bf16)),
+                   (int)(32 * (64 + kPadM + COUT + kPadM) * sizeof(__bf16)),
                    G::K / 64};
 }
 
@@ -536,4 +552,33 @@ void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
                      (hipStream_t)at::cuda::getCurrentCUDAStream().stream(),
                      (const void*)in.data_ptr(), (const __bf16*)gout.data_ptr(),
                      gw_ws.data_ptr<float>(), N, mb);
+}
+
+// tr16 semantics probe: fill LDS with 0..511, each lane passes base +
+// per-lane offset `mode`, dump what each lane's 4 elements are.
+namespace {
+typedef __bf16 v4bf __attribute__((ext_vector_type(4)));
+__global__ void tr16_probe_kernel(float* out, int mode) {
+  __shared__ __bf16 lds[512];
+  int lane = threadIdx.x & 63;
+  for (int i = threadIdx.x; i < 512; i += 64) lds[i] = (__bf16)(float)i;
+  __syncthreads();
+  int off;
+  switch (mode) {
+    case 0: off = 0; break;                 // uniform base
+    case 1: off = lane * 4; break;          // natural v4 stride
+    case 2: off = (lane >> 4) * 64; break;  // group stride 64
+    default: off = (lane & 15) * 4; break;
+  }
+  auto* p = (__attribute__((address_space(3))) v4bf*)(lds + off);
+  v4bf v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[lane * 4 + j] = (float)v[j];
+}
+}  // namespace
+
+void tr16_probe(torch::Tensor out, int64_t mode) {
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)at::cuda::getCurrentCUDAStream().stream(),
+                     out.data_ptr<float>(), (int)mode);
 }
