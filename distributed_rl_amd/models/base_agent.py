@@ -315,13 +315,52 @@ class BaseAgent(nn.Module):
         self._lstm_nodes = [
             n for n in self.node_order if isinstance(self.nodes[n], LSTMNET)
         ]
+        # detect the dueling pattern Substract(Add(A,V), Mean(A)) so it can
+        # run as ONE fused kernel on GPU (ops K3) instead of three graph nodes
+        consumers: Dict[str, int] = {}
+        for name in self.node_order:
+            for p in self.cfg[name].get("prevNodeNames", []):
+                consumers[p] = consumers.get(p, 0) + 1
+        self._dueling: Dict[str, Tuple[str, str]] = {}
+        self._dueling_skip: set = set()
+        for name in self.node_order:
+            ncfg = self.cfg[name]
+            if str(ncfg.get("netCat", "")).upper() != "SUBSTRACT":
+                continue
+            prevs = ncfg.get("prevNodeNames", [])
+            if len(prevs) != 2:
+                continue
+            addn, meann = prevs
+            acfg = self.cfg.get(addn, {})
+            mcfg = self.cfg.get(meann, {})
+            if (str(acfg.get("netCat", "")).upper() == "ADD"
+                    and str(mcfg.get("netCat", "")).upper() == "MEAN"):
+                aprev = acfg.get("prevNodeNames", [])
+                mprev = mcfg.get("prevNodeNames", [])
+                if (len(aprev) == 2 and mprev == [aprev[0]]
+                        and consumers.get(addn) == 1
+                        and consumers.get(meann) == 1):
+                    self._dueling[name] = (aprev[0], aprev[1])
+                    self._dueling_skip.update((addn, meann))
 
     # ---- forward -------------------------------------------------------
     def forward(self, inputs: Sequence[torch.Tensor]) -> List[torch.Tensor]:
         if isinstance(inputs, torch.Tensor):
             inputs = [inputs]
         produced: Dict[str, torch.Tensor] = {}
+        use_fused_dueling = bool(self._dueling) and inputs[0].is_cuda
         for name in self.node_order:
+            if use_fused_dueling:
+                if name in self._dueling_skip:
+                    continue
+                if name in self._dueling:
+                    from .. import ops as _ops
+
+                    a_name, v_name = self._dueling[name]
+                    produced[name] = _ops.dueling_head(
+                        produced[a_name].float(), produced[v_name].float()
+                    )
+                    continue
             ncfg = self.cfg[name]
             args: List[torch.Tensor] = [
                 produced[p] for p in ncfg.get("prevNodeNames", [])

@@ -172,6 +172,40 @@ def fused_conv_relu(x, weight, bias, stride: int):
 
 
 # ---------------------------------------------------------------------------
+# K3 — fused dueling-head epilogue (A + V) - mean(A)
+# ---------------------------------------------------------------------------
+
+
+class _DuelingFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, adv, val):
+        ext = hip_ext()
+        B, A = adv.shape
+        out = torch.empty(B, A, dtype=torch.float32, device=adv.device)
+        ext.dueling_fwd(adv.float().contiguous(),
+                        val.float().reshape(B).contiguous(), out)
+        ctx.A = A
+        return out
+
+    @staticmethod
+    def backward(ctx, g):
+        ext = hip_ext()
+        B, A = g.shape
+        g = g.contiguous()
+        gadv = torch.empty_like(g)
+        gval = torch.empty(B, dtype=torch.float32, device=g.device)
+        ext.dueling_bwd(g, gadv, gval)
+        return gadv, gval.unsqueeze(1)
+
+
+def dueling_head(adv: torch.Tensor, val: torch.Tensor) -> torch.Tensor:
+    """out = (adv + val) - mean(adv) (reference dueling graph nodes)."""
+    if _use_hip(adv):
+        return _DuelingFn.apply(adv, val)
+    return (adv + val) - adv.mean(dim=-1, keepdim=True)
+
+
+# ---------------------------------------------------------------------------
 # K4 — fused n-step double-DQN loss as an autograd Function
 # ---------------------------------------------------------------------------
 

@@ -536,6 +536,8 @@ void grad_clip(torch::Tensor flat, double max_norm, torch::Tensor sqsum_buf) {
 
 void register_conv(pybind11::module_& m);  // conv_mfma.hip
 void relu_mask_bwd(torch::Tensor gout, torch::Tensor out, torch::Tensor dst);
+void dueling_fwd(torch::Tensor adv, torch::Tensor val, torch::Tensor out);
+void dueling_bwd(torch::Tensor g, torch::Tensor gadv, torch::Tensor gval);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -554,6 +556,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vtrace", &vtrace, "IMPALA V-trace reversed scan (K8)");
   m.def("grad_clip", &grad_clip, "fused global grad-norm clip (K11)");
   m.def("relu_mask_bwd", &relu_mask_bwd, "dst = gout * (out > 0), bf16");
+  m.def("dueling_fwd", &dueling_fwd, "fused (A+V)-mean(A) (K3)");
+  m.def("dueling_bwd", &dueling_bwd, "dueling epilogue backward (K3)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -586,4 +590,51 @@ void relu_mask_bwd(torch::Tensor gout, torch::Tensor out, torch::Tensor dst) {
                      0, cur_stream(), (const ushort4*)out.data_ptr(),
                      (const ushort4*)gout.data_ptr(), (ushort4*)dst.data_ptr(),
                      n / 4);
+}
+
+// K3: fused dueling-head epilogue  out = (A + V) - mean(A)
+// (reference graph nodes Add/Mean/Substract, cfg/ape_x.json:72-88 — three
+// eager kernels per forward collapse into one; backward likewise).
+namespace {
+__global__ void dueling_fwd_kernel(const float* __restrict__ adv,
+                                   const float* __restrict__ val, int B, int A,
+                                   float* __restrict__ out) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B) return;
+  const float* a = adv + (int64_t)i * A;
+  float mean = 0.0f;
+  for (int j = 0; j < A; ++j) mean += a[j];
+  mean /= A;
+  float v = val[i];
+  float* o = out + (int64_t)i * A;
+  for (int j = 0; j < A; ++j) o[j] = a[j] + v - mean;
+}
+
+__global__ void dueling_bwd_kernel(const float* __restrict__ g, int B, int A,
+                                   float* __restrict__ gadv,
+                                   float* __restrict__ gval) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B) return;
+  const float* gi = g + (int64_t)i * A;
+  float s = 0.0f;
+  for (int j = 0; j < A; ++j) s += gi[j];
+  gval[i] = s;
+  float m = s / A;
+  float* ga = gadv + (int64_t)i * A;
+  for (int j = 0; j < A; ++j) ga[j] = gi[j] - m;
+}
+}  // namespace
+
+void dueling_fwd(torch::Tensor adv, torch::Tensor val, torch::Tensor out) {
+  int B = (int)adv.size(0), A = (int)adv.size(1);
+  hipLaunchKernelGGL(dueling_fwd_kernel, dim3(ceil_div(B, kBlock)), dim3(kBlock),
+                     0, cur_stream(), adv.data_ptr<float>(),
+                     val.data_ptr<float>(), B, A, out.data_ptr<float>());
+}
+
+void dueling_bwd(torch::Tensor g, torch::Tensor gadv, torch::Tensor gval) {
+  int B = (int)g.size(0), A = (int)g.size(1);
+  hipLaunchKernelGGL(dueling_bwd_kernel, dim3(ceil_div(B, kBlock)), dim3(kBlock),
+                     0, cur_stream(), g.data_ptr<float>(), B, A,
+                     gadv.data_ptr<float>(), gval.data_ptr<float>());
 }

@@ -172,3 +172,36 @@ def test_grad_clip_matches_torch(ext):
     y0 = y.clone()
     ops.clip_flat_grad_(y, max_norm)
     assert torch.equal(y, y0)
+
+
+def test_dueling_fused_vs_eager(ext):
+    from distributed_rl_amd import ops
+
+    torch.manual_seed(9)
+    B, A = 128, 6
+    adv = torch.randn(B, A, device=DEV, requires_grad=True)
+    val = torch.randn(B, 1, device=DEV, requires_grad=True)
+    out = ops.dueling_head(adv, val)
+    ref = (adv + val) - adv.mean(dim=-1, keepdim=True)
+    assert torch.allclose(out, ref, atol=1e-6)
+    g = torch.randn_like(out)
+    out.backward(g, retain_graph=False)
+    ga, gv = adv.grad.clone(), val.grad.clone()
+    adv.grad = None
+    val.grad = None
+    ref2 = (adv + val) - adv.mean(dim=-1, keepdim=True)
+    ref2.backward(g)
+    assert torch.allclose(ga, adv.grad, atol=1e-5)
+    assert torch.allclose(gv, val.grad, atol=1e-5)
+
+
+def test_model_dueling_fused_matches(ext):
+    from distributed_rl_amd.config import load_config
+    from distributed_rl_amd.models import BaseAgent
+
+    net = BaseAgent(load_config("ape_x").model_info).to(DEV)
+    assert net._dueling  # pattern detected
+    x = torch.rand(4, 4, 84, 84, device=DEV)
+    y_gpu = net.forward([x])[0]
+    y_cpu = net.cpu().forward([x.cpu()])[0]
+    assert torch.allclose(y_gpu.cpu(), y_cpu, atol=1e-4)
