@@ -227,7 +227,7 @@ class ApexLearner(LearnerBase):
             self.publish_weights()
 
     def step(self) -> Dict[str, torch.Tensor]:
-        stats = self._inner_step()
+        stats = self.maybe_profile_first_step(self._inner_step)
         self._cadence()
         return stats
 

@@ -65,6 +65,20 @@ class LearnerBase:
             self._writer.add_text("configuration", writeTrainInfo(self.cfg.raw).info, 0)
         return self._writer
 
+    def maybe_profile_first_step(self, fn, *args, **kw):
+        """cProfile the first train call (reference parity:
+        APE_X/Learner.py:177-180 profiles iteration 0). Enabled by
+        DRL_PROFILE_FIRST_STEP=1; prints cumulative-time top-20."""
+        if self.step_count == 0 and os.environ.get("DRL_PROFILE_FIRST_STEP") == "1":
+            import cProfile
+            import pstats
+
+            prof = cProfile.Profile()
+            out = prof.runcall(fn, *args, **kw)
+            pstats.Stats(prof).sort_stats("cumulative").print_stats(20)
+            return out
+        return fn(*args, **kw)
+
     def log_scalar(self, tag: str, value: float, step: Optional[int] = None):
         w = self.writer
         if w is not None:

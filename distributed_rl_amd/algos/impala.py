@@ -182,7 +182,7 @@ class ImpalaLearner(LearnerBase):
             self.publish_weights()
 
     def step(self):
-        stats = self._inner_step()
+        stats = self.maybe_profile_first_step(self._inner_step)
         self._cadence()
         return stats
 
