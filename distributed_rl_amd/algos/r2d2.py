@@ -107,13 +107,36 @@ class R2D2Learner(LearnerBase):
         self.beta = cfg.beta
         self.use_rescaling = cfg.use_rescaling
         self.model = self.build_model()
-        self.net = self.model  # LSTM path runs fp32 (K5 HIP LSTM is future work)
-        self.target = self.build_model()
-        self.target.updateParameter(self.model, 1.0)
+        self.mp = None
+        # Mixed trunk on GPU: the conv stack runs bf16 channels_last (our
+        # MFMA kernels) with fp32 master weights; the LSTM + heads stay fp32
+        # end-to-end — measured on MI355X the MIOpen fp32 RNN path beats
+        # bf16 nn.LSTM 2.4x, so until the K5 persistent-LSTM HIP kernel
+        # lands this split is the fast shape.
+        self._bf16_trunk = self.device.type == "cuda"
+        if self._bf16_trunk:
+            from ..parallel.precision import MixedPrecisionTrainer
+
+            self.model.to(memory_format=torch.channels_last)
+            conv_prefixes = tuple(
+                f"nodes.{name}." for name, node in self.cfg.model_info.items()
+                if str(node.get("netCat", "")).upper() == "CNN2D"
+            )
+            self.mp = MixedPrecisionTrainer(
+                self.model,
+                keep_fp32=lambda n, p: not n.startswith(conv_prefixes),
+            )
+            self.net = self.mp.compute
+            import copy as _copy
+
+            self.target = _copy.deepcopy(self.net)
+        else:
+            self.net = self.model
+            self.target = self.build_model()
+            self.target.updateParameter(self.model, 1.0)
         for p in self.target.parameters():
             p.requires_grad_(False)
         self.optim = self.build_optim(self.model)
-        self.mp = None
         cap = replay_capacity or cfg.replay_memory_len
         rdev = str(self.device) if self.device.type == "cuda" else "cpu"
         self.replay = make_per(
@@ -162,24 +185,24 @@ class R2D2Learner(LearnerBase):
         window's Q values; False runs the whole sequence (target net)."""
         T, B = frames.shape[:2]
         net.setCellState(h0)
+
+        def prep(chunk, steps):
+            x = chunk.reshape(steps * B, *frames.shape[2:])
+            if self._bf16_trunk:
+                x = x.contiguous(memory_format=torch.channels_last)
+            return x
+
         if burn_in_split and self.burn_in > 0:
             m = self.burn_in
             with torch.no_grad():
-                net.forward([
-                    frames[:m].reshape(m * B, *frames.shape[2:]),
-                    torch.tensor([m, B, -1]),
-                ])
+                net.forward([prep(frames[:m], m), torch.tensor([m, B, -1])])
             net.detachCellState()
-            q = net.forward([
-                frames[m:].reshape((T - m) * B, *frames.shape[2:]),
-                torch.tensor([T - m, B, -1]),
-            ])[0]
+            q = net.forward(
+                [prep(frames[m:], T - m), torch.tensor([T - m, B, -1])]
+            )[0]
             return q.view(T - m, B, -1)
         with torch.no_grad():
-            q = net.forward([
-                frames.reshape(T * B, *frames.shape[2:]),
-                torch.tensor([T, B, -1]),
-            ])[0]
+            q = net.forward([prep(frames, T), torch.tensor([T, B, -1])])[0]
         return q.view(T, B, -1)
 
     def _h0_to_state(self, h0: torch.Tensor):
@@ -193,9 +216,13 @@ class R2D2Learner(LearnerBase):
         B = data["done"].shape[0]
         T = self.T
         states = data["states"].to(self.device, non_blocking=True)  # (B,T,4,84,84)
-        frames = ops.dequant_frames(
-            states.permute(1, 0, 2, 3, 4).contiguous(), torch.float32
-        )  # (T,B,4,84,84) seq-major (R2D2/Learner.py:93)
+        if self._bf16_trunk:
+            # keep frames uint8, seq-major; the fused conv1 dequants in-kernel
+            frames = states.permute(1, 0, 2, 3, 4).contiguous()
+        else:
+            frames = ops.dequant_frames(
+                states.permute(1, 0, 2, 3, 4).contiguous(), torch.float32
+            )  # (T,B,4,84,84) seq-major (R2D2/Learner.py:93)
         actions = data["actions"].to(self.device).t().contiguous()  # (T,B)
         rewards = data["rewards"].to(self.device).t().contiguous()
         done = data["done"].to(self.device)
@@ -220,12 +247,20 @@ class R2D2Learner(LearnerBase):
         loss = 0.5 * (weights * td.pow(2).mean(dim=0)).mean()
         prio = ops.sequence_priority(td.detach().abs(), self.alpha, ETA)
 
-        self.optim.zero_grad(set_to_none=False)
-        loss.backward()
-        if self.reducer is not None:
-            self.reducer.all_reduce()
-        self.model.clippingNorm(40.0)  # R2D2/Learner.py:208
-        self.optim.step()
+        if self.mp is not None:
+            self.mp.zero_grads()
+            loss.backward()
+            self.mp.reduce_and_upcast()
+            self.model.clippingNorm(40.0)  # joint norm over both dtype groups
+            self.optim.step()
+            self.mp.sync_compute_params()
+        else:
+            self.optim.zero_grad(set_to_none=False)
+            loss.backward()
+            if self.reducer is not None:
+                self.reducer.all_reduce()
+            self.model.clippingNorm(40.0)  # R2D2/Learner.py:208
+            self.optim.step()
         self.replay.update(idx, prio)
         return {
             "loss": loss.detach(),
@@ -240,7 +275,7 @@ class R2D2Learner(LearnerBase):
     def _cadence(self):
         self.step_count += 1
         if self.step_count % self.cfg.target_frequency == 0:
-            self.target.updateParameter(self.model, 1.0)
+            self.target.load_state_dict(self.model.state_dict())  # casts per tensor
             self.publish_weights(include_target=True)
         elif self.step_count % PUBLISH_EVERY == 0:
             self.publish_weights()
@@ -249,6 +284,32 @@ class R2D2Learner(LearnerBase):
         stats = self.maybe_profile_first_step(self._inner_step)
         self._cadence()
         return stats
+
+    def make_graphed_step(self, warmup_iters: int = 3):
+        """hipGraph-capture the whole R2D2 step (PER sample -> burn-in +
+        train + target sequence passes -> n-step rescaled targets -> loss ->
+        backward -> Adam -> sequence-priority update)."""
+        assert self.device.type == "cuda"
+        assert self.reducer is None, "graphed R2D2 step is single-replica"
+        for g in self.optim.param_groups:
+            g["capturable"] = True
+        side = torch.cuda.Stream(self.device)
+        side.wait_stream(torch.cuda.current_stream(self.device))
+        with torch.cuda.stream(side):
+            for _ in range(warmup_iters):
+                self._inner_step()
+        torch.cuda.current_stream(self.device).wait_stream(side)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static_out = self._inner_step()
+        self._graph = graph
+
+        def stepper():
+            graph.replay()
+            self._cadence()
+            return static_out
+
+        return stepper
 
     # -- weights ------------------------------------------------------------
     def publish_weights(self, include_target: bool = False):
@@ -309,10 +370,14 @@ class R2D2Learner(LearnerBase):
         self.target.load_state_dict(state["target"])
         self.optim.load_state_dict(state["optim"])
         self.step_count = int(state["step"])
+        if self.mp is not None:
+            self.mp.sync_compute_params()
 
     def load_model_only(self, sd):
         self.model.load_state_dict(sd)
         self.target.load_state_dict(sd)
+        if self.mp is not None:
+            self.mp.sync_compute_params()
 
 
 # ===========================================================================

@@ -192,12 +192,15 @@ class LSTMNET(nn.Module):
             self._state = (self._state[0].detach(), self._state[1].detach())
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        p = next(self.lstm.parameters())
+        if x.dtype != p.dtype:
+            # mixed-trunk path: bf16 conv features into the fp32 LSTM
+            x = x.to(p.dtype)
         if x.dim() == 2:
             x = x.unsqueeze(0)  # (1, batch, feat)
         seq, batch, _ = x.shape
         if self._state is None or self._state[0].shape[1] != batch:
             self.zero_cell_state(batch)
-        # match input dtype under autocast
         st = self._state
         if st[0].dtype != x.dtype or st[0].device != x.device:
             st = (st[0].to(x.device, x.dtype), st[1].to(x.device, x.dtype))

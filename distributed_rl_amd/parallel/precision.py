@@ -2,30 +2,28 @@
 
 Instead of autocast (which re-casts every weight on every forward — ~29
 cast kernels/step measured on the Ape-X model, profiles/), we keep a
-persistent bf16 *compute* replica of the fp32 *master* model:
+persistent *compute* replica of the fp32 *master* model:
 
-  forward/backward run on the bf16 replica -> grads land in ONE flat bf16
-  buffer (param.grad are views) -> [optional RCCL all-reduce on the bf16
-  flat buffer — half the xGMI bytes of fp32] -> one fused cast into the
-  master's flat fp32 grad buffer -> optimizer.step() on fp32 -> one fused
-  cast back into the bf16 replica's flat param buffer.
+  forward/backward run on the compute replica -> grads land in flat
+  per-dtype buffers (param.grad are views) -> [optional RCCL all-reduce on
+  the flat buffers — bf16 halves the xGMI bytes] -> one fused cast per
+  dtype group into the master's flat fp32 grad buffer -> optimizer.step()
+  on fp32 -> one fused cast per group back into the compute replica.
 
-Every arrow is a single kernel (or one collective), all fixed-shape ->
-hipGraph-capturable.
+A ``keep_fp32`` predicate pins chosen parameters to fp32 in the compute
+replica (e.g. R2D2's LSTM, where the MIOpen fp32 RNN path is 2.4x faster
+than bf16 on MI355X) — the master layout is grouped [bf16-group |
+fp32-group] so each sync is still one contiguous cast/copy per group.
+Every step op is a fixed-shape kernel -> hipGraph-capturable.
 """
 
 from __future__ import annotations
 
 import copy
-from typing import Optional
+from typing import Callable, Optional
 
 import torch
 import torch.distributed as dist
-
-
-def _flatten_like(params, dtype, device) -> torch.Tensor:
-    total = sum(p.numel() for p in params)
-    return torch.zeros(total, dtype=dtype, device=device)
 
 
 def _is_cl(p: torch.Tensor) -> bool:
@@ -48,68 +46,108 @@ def _view_like(flat_slice: torch.Tensor, p: torch.Tensor) -> torch.Tensor:
     return flat_slice.view_as(p)
 
 
+class _Group:
+    """One dtype group: flat compute param/grad buffers + the master's
+    corresponding flat fp32 region."""
+
+    def __init__(self, dtype, c_params, m_params, device):
+        self.dtype = dtype
+        self.c_params = c_params
+        self.m_params = m_params
+        total = sum(p.numel() for p in c_params)
+        self.flat_cparam = torch.empty(total, dtype=dtype, device=device)
+        self.flat_cgrad = torch.zeros(total, dtype=dtype, device=device)
+        self.flat_mparam = torch.empty(total, dtype=torch.float32, device=device)
+        self.flat_mgrad = torch.zeros(total, dtype=torch.float32, device=device)
+        off = 0
+        with torch.no_grad():
+            for cp, mp in zip(c_params, m_params):
+                n = cp.numel()
+                self.flat_cparam[off : off + n].copy_(_to_flat(cp))
+                self.flat_mparam[off : off + n].copy_(_to_flat(mp).float())
+                cp.data = _view_like(self.flat_cparam[off : off + n], cp)
+                mp.data = _view_like(self.flat_mparam[off : off + n], mp)
+                cp.grad = _view_like(self.flat_cgrad[off : off + n], cp)
+                mp.grad = _view_like(self.flat_mgrad[off : off + n], mp)
+                off += n
+
+
 class MixedPrecisionTrainer:
     def __init__(self, master: torch.nn.Module,
                  group: Optional["dist.ProcessGroup"] = None,
-                 compute_dtype: torch.dtype = torch.bfloat16):
+                 compute_dtype: torch.dtype = torch.bfloat16,
+                 keep_fp32: Optional[Callable[[str, torch.Tensor], bool]] = None):
         self.master = master
         dev = next(master.parameters()).device
-        self.compute = copy.deepcopy(master).to(compute_dtype)
+        self.compute = copy.deepcopy(master)
+        # cast only the non-pinned params of the compute replica
+        named_master = [(n, p) for n, p in master.named_parameters()
+                        if p.requires_grad]
+        named_compute = dict(self.compute.named_parameters())
+        lo_c, lo_m, hi_c, hi_m = [], [], [], []
+        for name, mp_ in named_master:
+            cp = named_compute[name]
+            if keep_fp32 is not None and keep_fp32(name, mp_):
+                hi_c.append(cp)
+                hi_m.append(mp_)
+            else:
+                cp.data = cp.data.to(compute_dtype)
+                lo_c.append(cp)
+                lo_m.append(mp_)
         self.compute_dtype = compute_dtype
-        self.m_params = [p for p in master.parameters() if p.requires_grad]
-        self.c_params = [p for p in self.compute.parameters() if p.requires_grad]
-        assert len(self.m_params) == len(self.c_params)
-
-        # flat param buffer for the compute replica (params become views,
-        # preserving each param's memory format, e.g. channels_last convs)
-        total = sum(p.numel() for p in self.c_params)
-        self.flat_cparam = torch.empty(total, dtype=compute_dtype, device=dev)
-        off = 0
-        with torch.no_grad():
-            for p in self.c_params:
-                n = p.numel()
-                self.flat_cparam[off : off + n].copy_(_to_flat(p))
-                p.data = _view_like(self.flat_cparam[off : off + n], p)
-                off += n
-        # flat master params (views) so the downcast is one kernel
-        self.flat_mparam = _flatten_like(self.m_params, torch.float32, dev)
-        off = 0
-        with torch.no_grad():
-            for p in self.m_params:
-                n = p.numel()
-                self.flat_mparam[off : off + n].copy_(_to_flat(p).float())
-                p.data = _view_like(self.flat_mparam[off : off + n], p)
-                off += n
-        # flat grad buffers; compute grads accumulate in-place
-        self.flat_cgrad = _flatten_like(self.c_params, compute_dtype, dev)
-        self.flat_mgrad = _flatten_like(self.m_params, torch.float32, dev)
-        off = 0
-        for p, mp in zip(self.c_params, self.m_params):
-            n = p.numel()
-            p.grad = _view_like(self.flat_cgrad[off : off + n], p)
-            mp.grad = _view_like(self.flat_mgrad[off : off + n], mp)
-            off += n
-        self.group = group
+        self.groups = []
+        if lo_c:
+            self.groups.append(_Group(compute_dtype, lo_c, lo_m, dev))
+        if hi_c:
+            self.groups.append(_Group(torch.float32, hi_c, hi_m, dev))
+        self.pg = group
         self.world = dist.get_world_size(group) if dist.is_initialized() else 1
+
+    # -- convenience views (single-group fast paths used by callers) -------
+    @property
+    def flat_cparam(self):
+        assert len(self.groups) == 1
+        return self.groups[0].flat_cparam
+
+    @property
+    def flat_cgrad(self):
+        assert len(self.groups) == 1
+        return self.groups[0].flat_cgrad
+
+    @property
+    def flat_mgrad(self):
+        assert len(self.groups) == 1
+        return self.groups[0].flat_mgrad
+
+    @property
+    def flat_mparam(self):
+        assert len(self.groups) == 1
+        return self.groups[0].flat_mparam
+
+    def all_mgrads(self):
+        return [g.flat_mgrad for g in self.groups]
 
     # -- per-step plumbing -------------------------------------------------
     def zero_grads(self):
-        self.flat_cgrad.zero_()
+        for g in self.groups:
+            g.flat_cgrad.zero_()
 
     def reduce_and_upcast(self):
-        """bf16 all-reduce (if distributed) then one cast to fp32 grads."""
-        if self.world > 1:
-            dist.all_reduce(self.flat_cgrad, group=self.group)
-            self.flat_mgrad.copy_(self.flat_cgrad)
-            self.flat_mgrad.mul_(1.0 / self.world)
-        else:
-            self.flat_mgrad.copy_(self.flat_cgrad)
+        """all-reduce compute grads (if distributed) then cast to fp32."""
+        for g in self.groups:
+            if self.world > 1:
+                dist.all_reduce(g.flat_cgrad, group=self.pg)
+            g.flat_mgrad.copy_(g.flat_cgrad)
+            if self.world > 1:
+                g.flat_mgrad.mul_(1.0 / self.world)
 
     def sync_compute_params(self):
-        """fp32 master -> bf16 compute replica (one cast kernel)."""
-        self.flat_cparam.copy_(self.flat_mparam)
+        """fp32 master -> compute replica (one cast kernel per group)."""
+        for g in self.groups:
+            g.flat_cparam.copy_(g.flat_mparam)
 
     def broadcast_master(self):
         if self.world > 1:
-            dist.broadcast(self.flat_mparam, src=0, group=self.group)
+            for g in self.groups:
+                dist.broadcast(g.flat_mparam, src=0, group=self.pg)
             self.sync_compute_params()
