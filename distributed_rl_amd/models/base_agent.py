@@ -204,7 +204,14 @@ class LSTMNET(nn.Module):
         st = self._state
         if st[0].dtype != x.dtype or st[0].device != x.device:
             st = (st[0].to(x.device, x.dtype), st[1].to(x.device, x.dtype))
-        out, new_state = self.lstm(x, st)
+        if x.is_cuda and self.num_layers == 1:
+            # K5-lite: fused-cell manual recurrence (hipGraph-capturable;
+            # MIOpen's RNN path is not) — models/manual_lstm.py
+            from .manual_lstm import manual_lstm_seq
+
+            out, new_state = manual_lstm_seq(x, st, self.lstm)
+        else:
+            out, new_state = self.lstm(x, st)
         self._state = new_state
         if self.flatten_mode:
             out = out.reshape(seq * batch, self.hidden_size)

@@ -1,0 +1,87 @@
+"""Manual LSTM sequence (K5-lite): host-driven recurrence over fused HIP
+cell kernels.
+
+Why not MIOpen's RNN (what nn.LSTM dispatches to on ROCm): (a) it rejects
+hipGraph capture (hipErrorStreamCaptureUnsupported), so the R2D2 step can't
+be graphed around it; (b) its per-step kernels leave the input projection
+inside the loop. Here the input projection x @ W_ih^T + b is hoisted into
+ONE GEMM over all T timesteps, the loop body is one hh-addmm + one fused
+cell kernel (ops/hip/drl_kernels.hip lstm_cell_*), and backward is one
+fused cell-bwd kernel + one GEMM per step with the weight grads batched
+into two big GEMMs at the end. Everything is fixed-shape -> capturable.
+
+Uses nn.LSTM's own parameters (weight_ih_l0 / weight_hh_l0 / bias_*_l0) so
+state_dict parity with the eager/CPU path is exact.
+"""
+
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+
+class _ManualLSTMSeq(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, h0, c0, w_ih, w_hh, b_ih, b_hh):
+        # x: (T, B, IN); h0/c0: (B, H)
+        from ..ops import hip_ext
+
+        ext = hip_ext()
+        T, B, IN = x.shape
+        H = w_hh.shape[1]
+        dev = x.device
+        bias = b_ih + b_hh
+        xp = torch.addmm(bias, x.reshape(T * B, IN), w_ih.t()).view(T, B, 4 * H)
+        hs = torch.empty(T + 1, B, H, device=dev)
+        cs = torch.empty(T + 1, B, H, device=dev)
+        acts = torch.empty(T, B, 4 * H, device=dev)
+        tanhc = torch.empty(T, B, H, device=dev)
+        hs[0] = h0
+        cs[0] = c0
+        for t in range(T):
+            gates = torch.addmm(xp[t].reshape(B, 4 * H), hs[t], w_hh.t())
+            ext.lstm_cell_fwd(gates, cs[t], hs[t + 1], cs[t + 1], acts[t],
+                              tanhc[t])
+        ctx.save_for_backward(x, hs, cs, acts, tanhc, w_ih, w_hh)
+        ctx.dims = (T, B, IN, H)
+        return hs[1:].clone(), hs[T].clone(), cs[T].clone()
+
+    @staticmethod
+    def backward(ctx, gout, gh_T, gc_T):
+        from ..ops import hip_ext
+
+        ext = hip_ext()
+        x, hs, cs, acts, tanhc, w_ih, w_hh = ctx.saved_tensors
+        T, B, IN, H = ctx.dims
+        dev = x.device
+        dgates_all = torch.empty(T, B, 4 * H, device=dev)
+        dh = gh_T.contiguous().clone()
+        dc = gc_T.contiguous().clone()
+        dc_next = torch.empty(B, H, device=dev)
+        for t in range(T - 1, -1, -1):
+            dh = dh + gout[t]
+            ext.lstm_cell_bwd(dh.contiguous(), dc, acts[t], tanhc[t], cs[t],
+                              dgates_all[t], dc_next)
+            dc, dc_next = dc_next, dc
+            dh = dgates_all[t].mm(w_hh)
+        dg_flat = dgates_all.reshape(T * B, 4 * H)
+        dw_hh = dg_flat.t().mm(hs[:-1].reshape(T * B, H))
+        dw_ih = dg_flat.t().mm(x.reshape(T * B, IN))
+        db = dg_flat.sum(0)
+        dx = dg_flat.mm(w_ih).view(T, B, IN)
+        return dx, dh, dc, dw_ih, dw_hh, db, db
+
+
+def manual_lstm_seq(x: torch.Tensor, state: Tuple[torch.Tensor, torch.Tensor],
+                    lstm: torch.nn.LSTM):
+    """Run a single-layer LSTM over (T, B, IN) with fused cell kernels.
+
+    Returns (out (T,B,H), (h_T (1,B,H), c_T (1,B,H))) like nn.LSTM."""
+    assert lstm.num_layers == 1 and not lstm.bidirectional
+    h0, c0 = state
+    out, hT, cT = _ManualLSTMSeq.apply(
+        x.float(), h0[0].float(), c0[0].float(), lstm.weight_ih_l0,
+        lstm.weight_hh_l0, lstm.bias_ih_l0, lstm.bias_hh_l0,
+    )
+    return out, (hT.unsqueeze(0), cT.unsqueeze(0))

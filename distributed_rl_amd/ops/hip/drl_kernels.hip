@@ -538,6 +538,12 @@ void register_conv(pybind11::module_& m);  // conv_mfma.hip
 void relu_mask_bwd(torch::Tensor gout, torch::Tensor out, torch::Tensor dst);
 void dueling_fwd(torch::Tensor adv, torch::Tensor val, torch::Tensor out);
 void dueling_bwd(torch::Tensor g, torch::Tensor gadv, torch::Tensor gval);
+void lstm_cell_fwd(torch::Tensor gates, torch::Tensor c_prev,
+                   torch::Tensor h_out, torch::Tensor c_out,
+                   torch::Tensor acts, torch::Tensor tanhc);
+void lstm_cell_bwd(torch::Tensor dh, torch::Tensor dc_in, torch::Tensor acts,
+                   torch::Tensor tanhc, torch::Tensor c_prev,
+                   torch::Tensor dgates, torch::Tensor dc_prev);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -558,6 +564,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("relu_mask_bwd", &relu_mask_bwd, "dst = gout * (out > 0), bf16");
   m.def("dueling_fwd", &dueling_fwd, "fused (A+V)-mean(A) (K3)");
   m.def("dueling_bwd", &dueling_bwd, "dueling epilogue backward (K3)");
+  m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LSTM cell forward (K5)");
+  m.def("lstm_cell_bwd", &lstm_cell_bwd, "fused LSTM cell backward (K5)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -637,4 +645,96 @@ void dueling_bwd(torch::Tensor g, torch::Tensor gadv, torch::Tensor gval) {
   hipLaunchKernelGGL(dueling_bwd_kernel, dim3(ceil_div(B, kBlock)), dim3(kBlock),
                      0, cur_stream(), g.data_ptr<float>(), B, A,
                      gadv.data_ptr<float>(), gval.data_ptr<float>());
+}
+
+// K5-lite: fused LSTM cell pointwise (forward + backward). The sequence
+// loop lives on the host (one hh-GEMM + one cell kernel per step, all
+// fixed-shape -> hipGraph-capturable, unlike MIOpen's RNN path); the
+// input projection is hoisted into one big GEMM over all timesteps.
+// Gate order matches nn.LSTM: [i | f | g | o].
+namespace {
+__global__ void lstm_cell_fwd_kernel(
+    const float* __restrict__ gates,   // (B, 4H) pre-activation
+    const float* __restrict__ c_prev,  // (B, H)
+    float* __restrict__ h_out, float* __restrict__ c_out,
+    float* __restrict__ acts,          // (B, 4H) post-activation save
+    float* __restrict__ tanhc,         // (B, H) save
+    int64_t BH, int H) {
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; idx < BH; idx += stride) {
+    int64_t b = idx / H;
+    int64_t j = idx - b * H;
+    const float* g4 = gates + b * 4 * H;
+    float i = 1.0f / (1.0f + __expf(-g4[j]));
+    float f = 1.0f / (1.0f + __expf(-g4[H + j]));
+    float g = tanhf(g4[2 * H + j]);
+    float o = 1.0f / (1.0f + __expf(-g4[3 * H + j]));
+    float c = f * c_prev[idx] + i * g;
+    float tc = tanhf(c);
+    h_out[idx] = o * tc;
+    c_out[idx] = c;
+    float* a4 = acts + b * 4 * H;
+    a4[j] = i;
+    a4[H + j] = f;
+    a4[2 * H + j] = g;
+    a4[3 * H + j] = o;
+    tanhc[idx] = tc;
+  }
+}
+
+__global__ void lstm_cell_bwd_kernel(
+    const float* __restrict__ dh,      // (B, H) total upstream dL/dh_t
+    const float* __restrict__ dc_in,   // (B, H) dL/dc_t from t+1
+    const float* __restrict__ acts,    // (B, 4H) saved post-activations
+    const float* __restrict__ tanhc,   // (B, H)
+    const float* __restrict__ c_prev,  // (B, H)
+    float* __restrict__ dgates,        // (B, 4H) pre-activation grads
+    float* __restrict__ dc_prev,       // (B, H)
+    int64_t BH, int H) {
+  int64_t idx = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (; idx < BH; idx += stride) {
+    int64_t b = idx / H;
+    int64_t j = idx - b * H;
+    const float* a4 = acts + b * 4 * H;
+    float i = a4[j], f = a4[H + j], g = a4[2 * H + j], o = a4[3 * H + j];
+    float tc = tanhc[idx];
+    float dhv = dh[idx];
+    float do_ = dhv * tc;
+    float dct = dc_in[idx] + dhv * o * (1.0f - tc * tc);
+    float di = dct * g;
+    float df = dct * c_prev[idx];
+    float dg = dct * i;
+    dc_prev[idx] = dct * f;
+    float* d4 = dgates + b * 4 * H;
+    d4[j] = di * i * (1.0f - i);
+    d4[H + j] = df * f * (1.0f - f);
+    d4[2 * H + j] = dg * (1.0f - g * g);
+    d4[3 * H + j] = do_ * o * (1.0f - o);
+  }
+}
+}  // namespace
+
+void lstm_cell_fwd(torch::Tensor gates, torch::Tensor c_prev,
+                   torch::Tensor h_out, torch::Tensor c_out,
+                   torch::Tensor acts, torch::Tensor tanhc) {
+  int64_t B = c_prev.size(0), H = c_prev.size(1);
+  hipLaunchKernelGGL(lstm_cell_fwd_kernel, dim3(grid_for(B * H)), dim3(kBlock),
+                     0, cur_stream(), gates.data_ptr<float>(),
+                     c_prev.data_ptr<float>(), h_out.data_ptr<float>(),
+                     c_out.data_ptr<float>(), acts.data_ptr<float>(),
+                     tanhc.data_ptr<float>(), B * H, (int)H);
+}
+
+void lstm_cell_bwd(torch::Tensor dh, torch::Tensor dc_in, torch::Tensor acts,
+                   torch::Tensor tanhc, torch::Tensor c_prev,
+                   torch::Tensor dgates, torch::Tensor dc_prev) {
+  int64_t B = dh.size(0), H = dh.size(1);
+  hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(grid_for(B * H)), dim3(kBlock),
+                     0, cur_stream(), dh.data_ptr<float>(),
+                     dc_in.data_ptr<float>(), acts.data_ptr<float>(),
+                     tanhc.data_ptr<float>(), c_prev.data_ptr<float>(),
+                     dgates.data_ptr<float>(), dc_prev.data_ptr<float>(),
+                     B * H, (int)H);
 }
