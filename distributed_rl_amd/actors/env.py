@@ -40,8 +40,11 @@ class SyntheticEnv:
         self._stack = np.zeros((STACK, *FRAME_SHAPE), dtype=np.uint8)
 
     def _frame(self) -> np.ndarray:
-        base = (self._hidden * 37) % 200
-        f = self.rng.integers(0, 56, size=FRAME_SHAPE, dtype=np.uint8) + base
+        # frame mean encodes (hidden mod action_n) exactly, so the optimal
+        # policy is decodable from pixels (learning sanity checks depend on
+        # this: tools/learning_sanity.py)
+        base = 20 + (self._hidden % self.action_n) * 32
+        f = self.rng.integers(0, 48, size=FRAME_SHAPE, dtype=np.uint8) + base
         return f.astype(np.uint8)
 
     def reset(self) -> np.ndarray:

@@ -300,7 +300,7 @@ class ApexLearner(LearnerBase):
 
         def stepper():
             g1.replay()
-            torch.distributed.all_reduce(mp.flat_cgrad, group=mp.group)
+            torch.distributed.all_reduce(mp.flat_cgrad, group=mp.pg)
             g2.replay()
             self._cadence()
             return static_out
