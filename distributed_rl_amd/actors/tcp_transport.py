@@ -191,3 +191,28 @@ class TcpActorEndpoint:
 
     def close(self):
         self.sock.close()
+
+
+class SplitActorEndpoint:
+    """3-tier mode actor endpoint: experience goes to the replay-server
+    node, weights/rewards to the learner's transport server (the reference
+    splits these across its two Redis servers the same way)."""
+
+    def __init__(self, exp_endpoint: TcpActorEndpoint,
+                 weight_endpoint: TcpActorEndpoint):
+        self.exp = exp_endpoint
+        self.weights = weight_endpoint
+        self.idx = weight_endpoint.idx
+
+    def push(self, columns, priorities=None):
+        self.exp.push(columns, priorities)
+
+    def fetch(self):
+        return self.weights.fetch()
+
+    def push_reward(self, idx, reward, eps=0.0):
+        self.weights.push_reward(idx, reward, eps)
+
+    def close(self):
+        self.exp.close()
+        self.weights.close()

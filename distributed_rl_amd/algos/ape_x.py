@@ -46,7 +46,7 @@ class ApexLearner(LearnerBase):
                  replay_capacity: Optional[int] = None,
                  replay_device: Optional[str] = None,
                  replay_state_dtype: Optional[torch.dtype] = None,
-                 enable_tb: bool = True, run_root: str = "."):
+                 replay=None, enable_tb: bool = True, run_root: str = "."):
         super().__init__(cfg, device, rank, world_size, run_root=run_root,
                          enable_tb=enable_tb)
         self.batch_size = batch_size or cfg.batch_size
@@ -99,11 +99,18 @@ class ApexLearner(LearnerBase):
         # or float16 (BASELINE config 5's "fp16 replay compression" option —
         # frames stored pre-scaled to [0,1]).
         self.state_dtype = replay_state_dtype or torch.uint8
-        schema = (
-            make_apex_schema(frame_shape=(84, 84, 4), state_dtype=self.state_dtype)
-            if self._nhwc else make_apex_schema(state_dtype=self.state_dtype)
-        )
-        self.replay = make_per(cap, schema, device=rdev)
+        if replay is not None:
+            # injected replay (e.g. replay.server.RemoteReplay, 3-tier mode);
+            # batches arrive NCHW on the replay's own device
+            self.replay = replay
+            self._nhwc = False
+        else:
+            schema = (
+                make_apex_schema(frame_shape=(84, 84, 4),
+                                 state_dtype=self.state_dtype)
+                if self._nhwc else make_apex_schema(state_dtype=self.state_dtype)
+            )
+            self.replay = make_per(cap, schema, device=rdev)
         self.transport = transport
         self.gamma = cfg.gamma
         self.n_step = cfg.unroll_step
@@ -166,9 +173,17 @@ class ApexLearner(LearnerBase):
         cuda = self.device.type == "cuda"
         if cuda:
             # NHWC frames straight into the fused conv stack (u8: dequant
-            # fused into conv1; fp16-compressed: one cast to bf16)
-            s = data["state"].permute(0, 3, 1, 2)
-            sp = data["next_state"].permute(0, 3, 1, 2)
+            # fused into conv1; fp16-compressed: one cast to bf16). Batches
+            # from a remote replay arrive NCHW on CPU -> move + relayout.
+            def to_cl(t):
+                if t.shape[-1] == 4:  # NHWC storage
+                    return t.to(self.device, non_blocking=True).permute(0, 3, 1, 2)
+                return t.to(self.device, non_blocking=True).contiguous(
+                    memory_format=torch.channels_last
+                )
+
+            s = to_cl(data["state"])
+            sp = to_cl(data["next_state"])
             if s.dtype == torch.float16:
                 s = s.to(torch.bfloat16)
                 sp = sp.to(torch.bfloat16)

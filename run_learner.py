@@ -44,6 +44,9 @@ def main():
     ap.add_argument("--tcp-port", type=int, default=None,
                     help="serve actors over TCP on this port (multi-host "
                          "mode) instead of shared-memory rings")
+    ap.add_argument("--replay-server", default=None,
+                    help="host:port of a dedicated replay-server node "
+                         "(3-tier mode, run_replay_server.py)")
     args = ap.parse_args()
 
     rank, local_rank, world = init_distributed()
@@ -79,8 +82,15 @@ def main():
                                            create=False)
             transport = LearnerEndpoint(session, rank=rank, world_size=world)
 
+    learner_kw = {}
+    if args.replay_server:
+        from distributed_rl_amd.replay.server import RemoteReplay
+
+        host, port = args.replay_server.rsplit(":", 1)
+        learner_kw["replay"] = RemoteReplay(host, int(port))
     learner = get_learner_cls(cfg.alg)(
         cfg, device=device, rank=rank, world_size=world, transport=transport,
+        **learner_kw,
     )
     attach_reducer(learner)
     if args.resume:
