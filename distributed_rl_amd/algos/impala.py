@@ -55,7 +55,7 @@ class ImpalaLearner(LearnerBase):
     def __init__(self, cfg: Config, device: Optional[str] = None, rank: int = 0,
                  world_size: int = 1, transport=None,
                  batch_size: Optional[int] = None,
-                 replay_capacity: Optional[int] = None,
+                 replay_capacity: Optional[int] = None, replay=None,
                  publish_every: int = 1, enable_tb: bool = True,
                  run_root: str = "."):
         super().__init__(cfg, device, rank, world_size, run_root=run_root,
@@ -75,7 +75,8 @@ class ImpalaLearner(LearnerBase):
         self.optim = self.build_optim(self.model)
         cap = replay_capacity or cfg.replay_memory_len
         rdev = str(self.device) if self.device.type == "cuda" else "cpu"
-        self.replay = FifoReplay(cap, make_impala_schema(self.unroll), device=rdev)
+        self.replay = replay if replay is not None else FifoReplay(
+            cap, make_impala_schema(self.unroll), device=rdev)
         self.transport = transport
         self.publish_every = publish_every
         self.gamma = cfg.gamma

@@ -94,7 +94,7 @@ class R2D2Learner(LearnerBase):
     def __init__(self, cfg: Config, device: Optional[str] = None, rank: int = 0,
                  world_size: int = 1, transport=None,
                  batch_size: Optional[int] = None,
-                 replay_capacity: Optional[int] = None,
+                 replay_capacity: Optional[int] = None, replay=None,
                  enable_tb: bool = True, run_root: str = "."):
         super().__init__(cfg, device, rank, world_size, run_root=run_root,
                          enable_tb=enable_tb)
@@ -139,7 +139,7 @@ class R2D2Learner(LearnerBase):
         self.optim = self.build_optim(self.model)
         cap = replay_capacity or cfg.replay_memory_len
         rdev = str(self.device) if self.device.type == "cuda" else "cpu"
-        self.replay = make_per(
+        self.replay = replay if replay is not None else make_per(
             cap, make_r2d2_schema(self.T, hidden=self._hidden_size()), device=rdev
         )
         self.transport = transport
