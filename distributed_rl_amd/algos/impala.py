@@ -127,14 +127,23 @@ class ImpalaLearner(LearnerBase):
         rewards = data["rewards"].to(self.device)
         not_done = data["not_done"].to(self.device)
 
-        log_pi = F.log_softmax(logits[:, :T], dim=-1)  # (B, T, A)
-        log_pi_a = log_pi.gather(2, actions.unsqueeze(2)).squeeze(2)  # (B, T)
         v_t = values[:, :T]
         bootstrap = values[:, T]
+        if cuda:
+            # K9 fused path: one softmax-stats kernel feeds V-trace, then the
+            # objective with a closed-form backward kernel
+            logits_flat = logits[:, :T].reshape(B * T, A).contiguous()
+            actions_flat = actions.reshape(-1)
+            stats9 = ops.policy_softmax_stats(logits_flat, actions_flat)
+            log_pi_a_det = stats9[0].view(B, T)
+        else:
+            log_pi = F.log_softmax(logits[:, :T], dim=-1)  # (B, T, A)
+            log_pi_a = log_pi.gather(2, actions.unsqueeze(2)).squeeze(2)
+            log_pi_a_det = log_pi_a.detach()
 
         with torch.no_grad():
             vs_T, pg_adv_T, _ = ops.vtrace(
-                mu.log().t().contiguous(), log_pi_a.detach().t().contiguous(),
+                mu.log().t().contiguous(), log_pi_a_det.t().contiguous(),
                 rewards.t().contiguous(), v_t.detach().t().contiguous(),
                 bootstrap.detach(), not_done, self.gamma,
                 rho_bar=self.cfg.p_value, c_bar=self.cfg.c_value,
@@ -143,9 +152,15 @@ class ImpalaLearner(LearnerBase):
         vs = vs_T.t()
         pg_adv = pg_adv_T.t()
 
-        pi = log_pi.exp()
-        entropy = -(pi * log_pi).sum(-1).mean()
-        obj_actor = (log_pi_a * pg_adv).mean() + self.cfg.entropy_r * entropy
+        if cuda:
+            obj_actor, entropy = ops.impala_policy_objective(
+                logits_flat, actions_flat, pg_adv.reshape(-1),
+                self.cfg.entropy_r, stats=stats9,
+            )
+        else:
+            pi = log_pi.exp()
+            entropy = -(pi * log_pi).sum(-1).mean()
+            obj_actor = (log_pi_a * pg_adv).mean() + self.cfg.entropy_r * entropy
         critic_loss = 0.5 * F.mse_loss(v_t, vs)
         loss = -obj_actor + critic_loss
 

@@ -313,6 +313,68 @@ def vtrace(behavior_log_prob, target_log_prob, rewards, values, bootstrap_value,
 
 
 # ---------------------------------------------------------------------------
+# K9 — fused IMPALA policy objective
+# ---------------------------------------------------------------------------
+
+
+def policy_softmax_stats(logits, actions):
+    """One fused pass over (N, A) logits -> (log_pi_a, pi, H, mean_entropy),
+    all detached (K9 part 1; feeds V-trace before the objective exists)."""
+    ext = hip_ext()
+    N, A = logits.shape
+    dev = logits.device
+    dummy = torch.zeros(N, device=dev)
+    obj = torch.zeros(1, device=dev)
+    ent = torch.zeros(1, device=dev)
+    logpa = torch.empty(N, device=dev)
+    pi = torch.empty(N, A, device=dev)
+    H = torch.empty(N, device=dev)
+    ext.policy_loss_fwd(logits.detach().contiguous(), actions.contiguous(),
+                        dummy, 0.0, obj, ent, logpa, pi, H)
+    return logpa, pi, H, ent.squeeze(0)
+
+
+class _PolicyObjFn(torch.autograd.Function):
+    """obj = mean(log_pi_a * adv) + er * mean(H), differentiable in logits
+    via the closed-form softmax gradient (K9 part 2)."""
+
+    @staticmethod
+    def forward(ctx, logits, logpa, pi, H, actions, adv, er):
+        ctx.save_for_backward(pi, H, actions, adv)
+        ctx.er = er
+        return (logpa * adv).mean() + er * H.mean()
+
+    @staticmethod
+    def backward(ctx, gobj):
+        ext = hip_ext()
+        pi, H, actions, adv = ctx.saved_tensors
+        dlogits = torch.empty_like(pi)
+        ext.policy_loss_bwd(pi, H, actions, adv,
+                            gobj.reshape(1).contiguous(), ctx.er, dlogits)
+        return dlogits, None, None, None, None, None, None
+
+
+def impala_policy_objective(logits, actions, adv, entropy_coef,
+                            stats=None):
+    """obj_actor = mean(log pi(a) * adv) + er * mean(H); returns
+    (obj, entropy). ``stats`` is the policy_softmax_stats tuple when it was
+    already computed for V-trace."""
+    if _use_hip(logits):
+        if stats is None:
+            stats = policy_softmax_stats(logits, actions.long())
+        logpa, pi, H, ent = stats
+        obj = _PolicyObjFn.apply(logits, logpa, pi, H, actions.long(),
+                                 adv.float(), entropy_coef)
+        return obj, ent
+    log_pi = torch.log_softmax(logits.float(), dim=-1)
+    pi = log_pi.exp()
+    entropy = -(pi * log_pi).sum(-1).mean()
+    log_pi_a = log_pi.gather(1, actions.long().unsqueeze(1)).squeeze(1)
+    obj = (log_pi_a * adv).mean() + entropy_coef * entropy
+    return obj, entropy.detach()
+
+
+# ---------------------------------------------------------------------------
 # K11 — fused grad clip over a flat buffer
 # ---------------------------------------------------------------------------
 

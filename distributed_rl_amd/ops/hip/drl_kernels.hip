@@ -544,6 +544,13 @@ void lstm_cell_fwd(torch::Tensor gates, torch::Tensor c_prev,
 void lstm_cell_bwd(torch::Tensor dh, torch::Tensor dc_in, torch::Tensor acts,
                    torch::Tensor tanhc, torch::Tensor c_prev,
                    torch::Tensor dgates, torch::Tensor dc_prev);
+void policy_loss_fwd(torch::Tensor logits, torch::Tensor act, torch::Tensor adv,
+                     double er, torch::Tensor obj_out, torch::Tensor ent_out,
+                     torch::Tensor logpa_out, torch::Tensor pi_save,
+                     torch::Tensor H_save);
+void policy_loss_bwd(torch::Tensor pi_save, torch::Tensor H_save,
+                     torch::Tensor act, torch::Tensor adv, torch::Tensor gout,
+                     double er, torch::Tensor dlogits);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -566,6 +573,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("dueling_bwd", &dueling_bwd, "dueling epilogue backward (K3)");
   m.def("lstm_cell_fwd", &lstm_cell_fwd, "fused LSTM cell forward (K5)");
   m.def("lstm_cell_bwd", &lstm_cell_bwd, "fused LSTM cell backward (K5)");
+  m.def("policy_loss_fwd", &policy_loss_fwd, "fused IMPALA policy obj fwd (K9)");
+  m.def("policy_loss_bwd", &policy_loss_bwd, "fused IMPALA policy obj bwd (K9)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -737,4 +746,97 @@ void lstm_cell_bwd(torch::Tensor dh, torch::Tensor dc_in, torch::Tensor acts,
                      tanhc.data_ptr<float>(), c_prev.data_ptr<float>(),
                      dgates.data_ptr<float>(), dc_prev.data_ptr<float>(),
                      B * H, (int)H);
+}
+
+// K9: fused IMPALA policy objective (IMPALA/Learner.py:95-114):
+//   obj = mean_i( log pi(a_i) * adv_i ) + er * mean_i( H_i )
+// forward computes softmax stats in one pass (A is small); backward emits
+// d(-obj)/dlogits in closed form, so the log_softmax/exp/gather/entropy
+// chain (and its autograd tape) collapses into 2 kernels.
+namespace {
+__global__ void policy_loss_fwd_kernel(
+    const float* __restrict__ logits,  // (N, A)
+    const int64_t* __restrict__ act, const float* __restrict__ adv,
+    int N, int A, float er,
+    float* __restrict__ obj_out,      // scalar, pre-zeroed (atomicAdd)
+    float* __restrict__ ent_out,      // scalar, pre-zeroed
+    float* __restrict__ logpa_out,    // (N,)
+    float* __restrict__ pi_save,      // (N, A)
+    float* __restrict__ H_save) {     // (N,)
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  float obj = 0.0f, entv = 0.0f;
+  if (i < N) {
+    const float* row = logits + (int64_t)i * A;
+    float mx = row[0];
+    for (int j = 1; j < A; ++j) mx = fmaxf(mx, row[j]);
+    float Z = 0.0f;
+    for (int j = 0; j < A; ++j) Z += __expf(row[j] - mx);
+    float logZ = __logf(Z) + mx;
+    float H = 0.0f;
+    float* pr = pi_save + (int64_t)i * A;
+    for (int j = 0; j < A; ++j) {
+      float lp = row[j] - logZ;
+      float p = __expf(lp);
+      pr[j] = p;
+      H -= p * lp;
+    }
+    H_save[i] = H;
+    float lpa = row[act[i]] - logZ;
+    logpa_out[i] = lpa;
+    float invN = 1.0f / N;
+    obj = lpa * adv[i] * invN;
+    entv = H * invN;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    obj += __shfl_down(obj, off, 64);
+    entv += __shfl_down(entv, off, 64);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    if (obj != 0.0f) atomicAdd(obj_out, obj);
+    if (entv != 0.0f) atomicAdd(ent_out, entv);
+  }
+}
+
+__global__ void policy_loss_bwd_kernel(
+    const float* __restrict__ pi_save, const float* __restrict__ H_save,
+    const int64_t* __restrict__ act, const float* __restrict__ adv,
+    const float* __restrict__ gout,  // d(total)/d(obj): scalar (usually -1)
+    int N, int A, float er, float* __restrict__ dlogits) {
+  int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (k >= (int64_t)N * A) return;
+  int i = (int)(k / A);
+  int j = (int)(k - (int64_t)i * A);
+  float p = pi_save[k];
+  float lp = __logf(fmaxf(p, 1e-30f));
+  float invN = 1.0f / N;
+  // d obj / d logit_ij = [ adv_i*(1{j=a} - p) - er * p*(lp + H_i) ] / N
+  float d = adv[i] * (((int)act[i] == j ? 1.0f : 0.0f) - p)
+            - er * p * (lp + H_save[i]);
+  dlogits[k] = gout[0] * d * invN;
+}
+}  // namespace
+
+void policy_loss_fwd(torch::Tensor logits, torch::Tensor act, torch::Tensor adv,
+                     double er, torch::Tensor obj_out, torch::Tensor ent_out,
+                     torch::Tensor logpa_out, torch::Tensor pi_save,
+                     torch::Tensor H_save) {
+  int N = (int)logits.size(0), A = (int)logits.size(1);
+  hipLaunchKernelGGL(policy_loss_fwd_kernel, dim3(ceil_div(N, kBlock)),
+                     dim3(kBlock), 0, cur_stream(), logits.data_ptr<float>(),
+                     act.data_ptr<int64_t>(), adv.data_ptr<float>(), N, A,
+                     (float)er, obj_out.data_ptr<float>(),
+                     ent_out.data_ptr<float>(), logpa_out.data_ptr<float>(),
+                     pi_save.data_ptr<float>(), H_save.data_ptr<float>());
+}
+
+void policy_loss_bwd(torch::Tensor pi_save, torch::Tensor H_save,
+                     torch::Tensor act, torch::Tensor adv, torch::Tensor gout,
+                     double er, torch::Tensor dlogits) {
+  int N = (int)pi_save.size(0), A = (int)pi_save.size(1);
+  hipLaunchKernelGGL(policy_loss_bwd_kernel,
+                     dim3(ceil_div((int64_t)N * A, kBlock)), dim3(kBlock), 0,
+                     cur_stream(), pi_save.data_ptr<float>(),
+                     H_save.data_ptr<float>(), act.data_ptr<int64_t>(),
+                     adv.data_ptr<float>(), gout.data_ptr<float>(), N, A,
+                     (float)er, dlogits.data_ptr<float>());
 }

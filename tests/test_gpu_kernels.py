@@ -205,3 +205,40 @@ def test_model_dueling_fused_matches(ext):
     y_gpu = net.forward([x])[0]
     y_cpu = net.cpu().forward([x.cpu()])[0]
     assert torch.allclose(y_gpu.cpu(), y_cpu, atol=1e-4)
+
+
+def test_policy_objective_fused_vs_torch(ext):
+    from distributed_rl_amd import ops
+
+    torch.manual_seed(13)
+    N, A = 256, 6
+    er = 0.01
+    logits = torch.randn(N, A, device=DEV, requires_grad=True)
+    actions = torch.randint(0, A, (N,), device=DEV)
+    adv = torch.randn(N, device=DEV)
+
+    stats = ops.policy_softmax_stats(logits, actions)
+    logpa, pi, H, ent = stats
+    ref_logpi = torch.log_softmax(logits.detach(), -1)
+    assert torch.allclose(logpa, ref_logpi.gather(1, actions.unsqueeze(1)
+                                                  ).squeeze(1), atol=1e-5)
+    assert torch.allclose(pi, ref_logpi.exp(), atol=1e-5)
+    ref_H = -(ref_logpi.exp() * ref_logpi).sum(-1)
+    assert torch.allclose(H, ref_H, atol=1e-5)
+    assert abs(ent.item() - ref_H.mean().item()) < 1e-5
+
+    obj, ent2 = ops.impala_policy_objective(logits, actions, adv, er,
+                                            stats=stats)
+    (-obj).backward()
+    g_fused = logits.grad.clone()
+
+    logits2 = logits.detach().clone().requires_grad_(True)
+    lp = torch.log_softmax(logits2, -1)
+    p2 = lp.exp()
+    entropy = -(p2 * lp).sum(-1).mean()
+    lpa = lp.gather(1, actions.unsqueeze(1)).squeeze(1)
+    obj_ref = (lpa * adv).mean() + er * entropy
+    assert abs(obj.item() - obj_ref.item()) < 1e-5
+    (-obj_ref).backward()
+    assert torch.allclose(g_fused, logits2.grad, atol=1e-5), (
+        (g_fused - logits2.grad).abs().max())
