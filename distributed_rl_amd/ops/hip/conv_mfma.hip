@@ -32,7 +32,8 @@ namespace {
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-constexpr int kPad = 8;  // LDS row pad (elements) to break bank conflicts
+constexpr int kPad = 32;  // LDS row pad: (K+pad)*2 % 256 in {64,192} spreads the
+// b128 16-lane service quads across all 16 bank-row slots (PMC-verified)
 
 template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8IN>
 struct ConvGeom {
