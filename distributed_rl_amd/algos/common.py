@@ -48,7 +48,14 @@ class LearnerBase:
     def build_model(self) -> BaseAgent:
         return BaseAgent(self.cfg.model_info).to(self.device)
 
-    def build_optim(self, model) -> torch.optim.Optimizer:
+    def build_optim(self, model):
+        if getattr(self, "mp", None) is not None:
+            # fused flat-buffer optimizer over the mp master (K12)
+            from ..parallel.flat_optim import make_flat_optimizer
+
+            fo = make_flat_optimizer(self.cfg.optim_info, self.mp)
+            if fo is not None:
+                return fo
         return get_optim(self.cfg.optim_info, model)
 
     # -- telemetry --------------------------------------------------------

@@ -551,6 +551,12 @@ void policy_loss_fwd(torch::Tensor logits, torch::Tensor act, torch::Tensor adv,
 void policy_loss_bwd(torch::Tensor pi_save, torch::Tensor H_save,
                      torch::Tensor act, torch::Tensor adv, torch::Tensor gout,
                      double er, torch::Tensor dlogits);
+void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor sq,
+                  torch::Tensor ga, torch::Tensor mom, double lr, double alpha,
+                  double eps, double wd, double mu, bool centered, bool has_mom);
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor step, double lr, double b1,
+               double b2, double eps, double wd);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -575,6 +581,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_cell_bwd", &lstm_cell_bwd, "fused LSTM cell backward (K5)");
   m.def("policy_loss_fwd", &policy_loss_fwd, "fused IMPALA policy obj fwd (K9)");
   m.def("policy_loss_bwd", &policy_loss_bwd, "fused IMPALA policy obj bwd (K9)");
+  m.def("rmsprop_step", &rmsprop_step, "fused flat centered RMSprop (K12)");
+  m.def("adam_step", &adam_step, "fused flat Adam (K12)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -839,4 +847,90 @@ void policy_loss_bwd(torch::Tensor pi_save, torch::Tensor H_save,
                      H_save.data_ptr<float>(), act.data_ptr<int64_t>(),
                      adv.data_ptr<float>(), gout.data_ptr<float>(), N, A,
                      (float)er, dlogits.data_ptr<float>());
+}
+
+// K12: fused optimizers over the flat master buffers. torch's capturable
+// foreach RMSprop costs ~170 us/step for a 1.7M-param model (6 x
+// multi_tensor_apply passes, profiles/); the whole centered-RMSprop update
+// is one memory-bound pass here. Math matches torch.optim exactly.
+namespace {
+__global__ void rmsprop_step_kernel(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ sq, float* __restrict__ ga, float* __restrict__ mom,
+    int64_t n, float lr, float alpha, float eps, float wd, float mu,
+    bool centered, bool has_mom) {
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float grad = g[i] + wd * p[i];
+    float s = alpha * sq[i] + (1.0f - alpha) * grad * grad;
+    sq[i] = s;
+    float avg;
+    if (centered) {
+      float a = alpha * ga[i] + (1.0f - alpha) * grad;
+      ga[i] = a;
+      avg = sqrtf(fmaxf(s - a * a, 0.0f)) + eps;
+    } else {
+      avg = sqrtf(s) + eps;
+    }
+    float upd = grad / avg;
+    if (has_mom) {
+      float m = mu * mom[i] + upd;
+      mom[i] = m;
+      upd = m;
+    }
+    p[i] -= lr * upd;
+  }
+}
+
+__global__ void adam_bump_kernel(float* step) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) step[0] += 1.0f;
+}
+
+__global__ void adam_step_kernel(
+    float* __restrict__ p, const float* __restrict__ g,
+    float* __restrict__ m, float* __restrict__ v,
+    const float* __restrict__ step, int64_t n, float lr, float b1, float b2,
+    float eps, float wd) {
+  float t = step[0];
+  float bc1 = 1.0f - __powf(b1, t);
+  float bc2 = 1.0f - __powf(b2, t);
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride) {
+    float grad = g[i] + wd * p[i];
+    float mi = b1 * m[i] + (1.0f - b1) * grad;
+    float vi = b2 * v[i] + (1.0f - b2) * grad * grad;
+    m[i] = mi;
+    v[i] = vi;
+    p[i] -= lr * (mi / bc1) / (sqrtf(vi / bc2) + eps);
+  }
+}
+}  // namespace
+
+void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor sq,
+                  torch::Tensor ga, torch::Tensor mom, double lr, double alpha,
+                  double eps, double wd, double mu, bool centered,
+                  bool has_mom) {
+  int64_t n = p.numel();
+  hipLaunchKernelGGL(rmsprop_step_kernel, dim3(grid_for(n, 4)), dim3(kBlock), 0,
+                     cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+                     sq.data_ptr<float>(),
+                     centered ? ga.data_ptr<float>() : sq.data_ptr<float>(),
+                     has_mom ? mom.data_ptr<float>() : sq.data_ptr<float>(),
+                     n, (float)lr, (float)alpha, (float)eps, (float)wd,
+                     (float)mu, centered, has_mom);
+}
+
+void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
+               torch::Tensor v, torch::Tensor step, double lr, double b1,
+               double b2, double eps, double wd) {
+  int64_t n = p.numel();
+  hipLaunchKernelGGL(adam_bump_kernel, dim3(1), dim3(64), 0, cur_stream(),
+                     step.data_ptr<float>());
+  hipLaunchKernelGGL(adam_step_kernel, dim3(grid_for(n, 4)), dim3(kBlock), 0,
+                     cur_stream(), p.data_ptr<float>(), g.data_ptr<float>(),
+                     m.data_ptr<float>(), v.data_ptr<float>(),
+                     step.data_ptr<float>(), n, (float)lr, (float)b1,
+                     (float)b2, (float)eps, (float)wd);
 }

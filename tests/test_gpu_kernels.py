@@ -242,3 +242,49 @@ def test_policy_objective_fused_vs_torch(ext):
     (-obj_ref).backward()
     assert torch.allclose(g_fused, logits2.grad, atol=1e-5), (
         (g_fused - logits2.grad).abs().max())
+
+
+def test_fused_rmsprop_matches_torch(ext):
+    torch.manual_seed(21)
+    n = 100_000
+    p = torch.randn(n, device=DEV)
+    g = torch.randn(n, device=DEV)
+    from distributed_rl_amd.parallel.flat_optim import FlatRMSprop
+
+    opt = FlatRMSprop([p], [g], lr=6.25e-5, alpha=0.95, eps=1.5e-7,
+                      weight_decay=0.0, momentum=0.0, centered=True)
+    p_ref = torch.nn.Parameter(p.clone())
+    ref = torch.optim.RMSprop([p_ref], lr=6.25e-5, alpha=0.95, eps=1.5e-7,
+                              centered=True, foreach=True)
+    for it in range(5):
+        gcur = torch.randn(n, device=DEV, generator=torch.Generator(DEV
+                           ).manual_seed(it))
+        g.copy_(gcur)
+        p_ref.grad = gcur.clone()
+        opt.step()
+        ref.step()
+    torch.cuda.synchronize()
+    assert torch.allclose(p, p_ref.detach(), atol=1e-6), (
+        (p - p_ref.detach()).abs().max())
+
+
+def test_fused_adam_matches_torch(ext):
+    torch.manual_seed(22)
+    n = 50_000
+    p = torch.randn(n, device=DEV)
+    g = torch.randn(n, device=DEV)
+    from distributed_rl_amd.parallel.flat_optim import FlatAdam
+
+    opt = FlatAdam([p], [g], lr=1e-4, eps=1e-3)
+    p_ref = torch.nn.Parameter(p.clone())
+    ref = torch.optim.Adam([p_ref], lr=1e-4, eps=1e-3, foreach=True)
+    for it in range(5):
+        gcur = torch.randn(n, device=DEV, generator=torch.Generator(DEV
+                           ).manual_seed(100 + it))
+        g.copy_(gcur)
+        p_ref.grad = gcur.clone()
+        opt.step()
+        ref.step()
+    torch.cuda.synchronize()
+    assert torch.allclose(p, p_ref.detach(), atol=1e-6), (
+        (p - p_ref.detach()).abs().max())
