@@ -74,8 +74,9 @@ def dequant_frames_nhwc(x_u8: torch.Tensor) -> torch.Tensor:
     """(N,4,H,W) uint8 -> channels_last bf16, fused /255 (K1 NHWC variant)."""
     if _use_hip(x_u8):
         out = torch.empty(
-            x_u8.shape, dtype=torch.bfloat16, device=x_u8.device
-        ).to(memory_format=torch.channels_last)
+            x_u8.shape, dtype=torch.bfloat16, device=x_u8.device,
+            memory_format=torch.channels_last,
+        )
         hip_ext().dequant_nhwc(x_u8.contiguous(), out)
         return out
     return torch_ref.dequant_frames(x_u8, torch.float32).to(
@@ -104,8 +105,9 @@ class _FusedConvFn(torch.autograd.Function):
         P = (H - KH) // stride + 1
         Q = (W - KW) // stride + 1
         out = torch.empty(
-            N, COUT, P, Q, dtype=torch.bfloat16, device=x.device
-        ).to(memory_format=torch.channels_last)
+            N, COUT, P, Q, dtype=torch.bfloat16, device=x.device,
+            memory_format=torch.channels_last,
+        )
         ext.conv_fwd(
             x, weight,
             bias if bias is not None else torch.empty(0, device=x.device),
