@@ -39,8 +39,12 @@ class _ManualLSTMSeq(torch.autograd.Function):
         tanhc = torch.empty(T, B, H, device=dev)
         hs[0] = h0
         cs[0] = c0
+        w_hh_t = w_hh.t()
         for t in range(T):
-            gates = torch.addmm(xp[t].reshape(B, 4 * H), hs[t], w_hh.t())
+            # accumulate the hh product INTO the hoisted projection buffer —
+            # in-place addmm_ avoids hipBLASLt's beta-copy of the C operand
+            gates = xp[t].reshape(B, 4 * H)
+            gates.addmm_(hs[t], w_hh_t)
             ext.lstm_cell_fwd(gates, cs[t], hs[t + 1], cs[t + 1], acts[t],
                               tanhc[t])
         ctx.save_for_backward(x, hs, cs, acts, tanhc, w_ih, w_hh)
@@ -59,10 +63,12 @@ class _ManualLSTMSeq(torch.autograd.Function):
         dh = gh_T.contiguous().clone()
         dc = gc_T.contiguous().clone()
         dc_next = torch.empty(B, H, device=dev)
+        gout = gout.contiguous()
+        empty = torch.empty(0, device=dev)
         for t in range(T - 1, -1, -1):
-            dh = dh + gout[t]
-            ext.lstm_cell_bwd(dh.contiguous(), dc, acts[t], tanhc[t], cs[t],
-                              dgates_all[t], dc_next)
+            # the per-step output grad is folded into the cell-bwd kernel
+            ext.lstm_cell_bwd(dh.contiguous(), gout[t], dc, acts[t], tanhc[t],
+                              cs[t], dgates_all[t], dc_next)
             dc, dc_next = dc_next, dc
             dh = dgates_all[t].mm(w_hh)
         dg_flat = dgates_all.reshape(T * B, 4 * H)

@@ -541,9 +541,10 @@ void dueling_bwd(torch::Tensor g, torch::Tensor gadv, torch::Tensor gval);
 void lstm_cell_fwd(torch::Tensor gates, torch::Tensor c_prev,
                    torch::Tensor h_out, torch::Tensor c_out,
                    torch::Tensor acts, torch::Tensor tanhc);
-void lstm_cell_bwd(torch::Tensor dh, torch::Tensor dc_in, torch::Tensor acts,
-                   torch::Tensor tanhc, torch::Tensor c_prev,
-                   torch::Tensor dgates, torch::Tensor dc_prev);
+void lstm_cell_bwd(torch::Tensor dh, torch::Tensor gout_t, torch::Tensor dc_in,
+                   torch::Tensor acts, torch::Tensor tanhc,
+                   torch::Tensor c_prev, torch::Tensor dgates,
+                   torch::Tensor dc_prev);
 void policy_loss_fwd(torch::Tensor logits, torch::Tensor act, torch::Tensor adv,
                      double er, torch::Tensor obj_out, torch::Tensor ent_out,
                      torch::Tensor logpa_out, torch::Tensor pi_save,
@@ -701,7 +702,8 @@ __global__ void lstm_cell_fwd_kernel(
 }
 
 __global__ void lstm_cell_bwd_kernel(
-    const float* __restrict__ dh,      // (B, H) total upstream dL/dh_t
+    const float* __restrict__ dh,      // (B, H) recurrent dL/dh_t
+    const float* __restrict__ gout_t,  // (B, H) per-step output grad or null
     const float* __restrict__ dc_in,   // (B, H) dL/dc_t from t+1
     const float* __restrict__ acts,    // (B, 4H) saved post-activations
     const float* __restrict__ tanhc,   // (B, H)
@@ -717,7 +719,7 @@ __global__ void lstm_cell_bwd_kernel(
     const float* a4 = acts + b * 4 * H;
     float i = a4[j], f = a4[H + j], g = a4[2 * H + j], o = a4[3 * H + j];
     float tc = tanhc[idx];
-    float dhv = dh[idx];
+    float dhv = dh[idx] + (gout_t ? gout_t[idx] : 0.0f);
     float do_ = dhv * tc;
     float dct = dc_in[idx] + dhv * o * (1.0f - tc * tc);
     float di = dct * g;
@@ -744,12 +746,14 @@ void lstm_cell_fwd(torch::Tensor gates, torch::Tensor c_prev,
                      tanhc.data_ptr<float>(), B * H, (int)H);
 }
 
-void lstm_cell_bwd(torch::Tensor dh, torch::Tensor dc_in, torch::Tensor acts,
-                   torch::Tensor tanhc, torch::Tensor c_prev,
-                   torch::Tensor dgates, torch::Tensor dc_prev) {
+void lstm_cell_bwd(torch::Tensor dh, torch::Tensor gout_t, torch::Tensor dc_in,
+                   torch::Tensor acts, torch::Tensor tanhc,
+                   torch::Tensor c_prev, torch::Tensor dgates,
+                   torch::Tensor dc_prev) {
   int64_t B = dh.size(0), H = dh.size(1);
+  const float* gp = gout_t.numel() ? gout_t.data_ptr<float>() : nullptr;
   hipLaunchKernelGGL(lstm_cell_bwd_kernel, dim3(grid_for(B * H)), dim3(kBlock),
-                     0, cur_stream(), dh.data_ptr<float>(),
+                     0, cur_stream(), dh.data_ptr<float>(), gp,
                      dc_in.data_ptr<float>(), acts.data_ptr<float>(),
                      tanhc.data_ptr<float>(), c_prev.data_ptr<float>(),
                      dgates.data_ptr<float>(), dc_prev.data_ptr<float>(),
