@@ -39,14 +39,24 @@ class _ManualLSTMSeq(torch.autograd.Function):
         tanhc = torch.empty(T, B, H, device=dev)
         hs[0] = h0
         cs[0] = c0
-        w_hh_t = w_hh.t()
-        for t in range(T):
-            # accumulate the hh product INTO the hoisted projection buffer —
-            # in-place addmm_ avoids hipBLASLt's beta-copy of the C operand
-            gates = xp[t].reshape(B, 4 * H)
-            gates.addmm_(hs[t], w_hh_t)
-            ext.lstm_cell_fwd(gates, cs[t], hs[t + 1], cs[t + 1], acts[t],
-                              tanhc[t])
+        w_hh_c = w_hh.contiguous()
+        # one-kernel fused step (fp32 MFMA hh-GEMM + cell) when the geometry
+        # allows; otherwise in-place addmm_ + fused cell kernel
+        fused = bool(ext.lstm_step_fused(
+            xp[0].reshape(B, 4 * H), hs[0], cs[0], w_hh_c, hs[1], cs[1],
+            acts[0], tanhc[0]))
+        if fused:
+            for t in range(1, T):
+                ext.lstm_step_fused(xp[t].reshape(B, 4 * H), hs[t], cs[t],
+                                    w_hh_c, hs[t + 1], cs[t + 1], acts[t],
+                                    tanhc[t])
+        else:
+            w_hh_t = w_hh.t()
+            for t in range(T):
+                gates = xp[t].reshape(B, 4 * H)
+                gates.addmm_(hs[t], w_hh_t)
+                ext.lstm_cell_fwd(gates, cs[t], hs[t + 1], cs[t + 1], acts[t],
+                                  tanhc[t])
         ctx.save_for_backward(x, hs, cs, acts, tanhc, w_ih, w_hh)
         ctx.dims = (T, B, IN, H)
         return hs[1:].clone(), hs[T].clone(), cs[T].clone()

@@ -558,6 +558,10 @@ void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor sq,
 void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                torch::Tensor v, torch::Tensor step, double lr, double b1,
                double b2, double eps, double wd);
+bool lstm_step_fused(torch::Tensor xp_t, torch::Tensor h_in, torch::Tensor c_in,
+                     torch::Tensor w_hh, torch::Tensor h_out,
+                     torch::Tensor c_out, torch::Tensor acts,
+                     torch::Tensor tanhc);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -584,6 +588,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("policy_loss_bwd", &policy_loss_bwd, "fused IMPALA policy obj bwd (K9)");
   m.def("rmsprop_step", &rmsprop_step, "fused flat centered RMSprop (K12)");
   m.def("adam_step", &adam_step, "fused flat Adam (K12)");
+  m.def("lstm_step_fused", &lstm_step_fused,
+        "one-kernel LSTM timestep: fp32-MFMA hh GEMM + cell (K5)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -937,4 +943,123 @@ void adam_step(torch::Tensor p, torch::Tensor g, torch::Tensor m,
                      m.data_ptr<float>(), v.data_ptr<float>(),
                      step.data_ptr<float>(), n, (float)lr, (float)b1,
                      (float)b2, (float)eps, (float)wd);
+}
+
+// K5 (full step): fused LSTM timestep — gates = xp_t + h_in @ W_hh^T on
+// fp32 MFMA (v_mfma_f32_16x16x4_f32, exact fp32) + the cell nonlinearity,
+// ONE kernel per timestep (vs hipBLASLt addmm ~8 us + cell ~5 us).
+//   Decomposition: grid = H/16 blocks; block = 4 waves; wave w computes the
+//   16-column slice of GATE w (torch gate order i,f,g,o along 4H), so after
+//   one barrier the block holds all 4 gates for its 16 h-columns and
+//   finishes the cell in-block. h_in (B x H) is staged once in LDS.
+//   Requires B <= 32 (R2D2 cfg batch); caller falls back otherwise.
+namespace {
+constexpr int kLstmMaxB = 32;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+template <int H>
+__global__ __launch_bounds__(256) void lstm_step_fused_kernel(
+    const float* __restrict__ xp_t,   // (B, 4H) hoisted input projection
+    const float* __restrict__ h_in,   // (B, H)
+    const float* __restrict__ c_in,   // (B, H)
+    const float* __restrict__ w_hh,   // (4H, H) row-major (torch layout)
+    float* __restrict__ h_out, float* __restrict__ c_out,
+    float* __restrict__ acts,         // (B, 4H)
+    float* __restrict__ tanhc,        // (B, H)
+    int B) {
+  constexpr int PAD = 4;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* h_lds = reinterpret_cast<float*>(smem);           // [32][H+PAD]
+  float* gbuf = h_lds + 32 * (H + PAD);                    // [4][32][16]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;   // = gate index
+  const int s16 = blockIdx.x * 16;  // h-column slice base
+
+  // ---- stage h_in (zero-padded rows) ----
+  for (int i = tid * 4; i < 32 * H; i += 256 * 4) {
+    int row = i / H;
+    int k = i - row * H;
+    float4 v = {0.f, 0.f, 0.f, 0.f};
+    if (row < B) v = *reinterpret_cast<const float4*>(h_in + row * H + k);
+    *reinterpret_cast<float4*>(h_lds + row * (H + PAD) + k) = v;
+  }
+  __syncthreads();
+
+  // ---- MFMA: 2 row-fragments x 16 cols, K = H in 16-wide groups ----
+  const int col = wave * H + s16 + (lane & 15);  // gate column in 4H
+  f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+  const int ksub = (lane >> 4) * 4;
+  const float* wrow = w_hh + (int64_t)col * H;
+#pragma unroll 4
+  for (int kb = 0; kb < H; kb += 16) {
+    float4 b4 = *reinterpret_cast<const float4*>(wrow + kb + ksub);
+    float4 a0 = *reinterpret_cast<const float4*>(
+        h_lds + (lane & 15) * (H + PAD) + kb + ksub);
+    float4 a1 = *reinterpret_cast<const float4*>(
+        h_lds + (16 + (lane & 15)) * (H + PAD) + kb + ksub);
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      acc[0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0[j], b4[j], acc[0], 0, 0, 0);
+      acc[1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1[j], b4[j], acc[1], 0, 0, 0);
+    }
+  }
+  // ---- add xp, park gates in LDS ----
+  const int crow = (lane >> 4) * 4;
+#pragma unroll
+  for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = rf * 16 + crow + r;
+      int rclamp = row < B ? row : 0;
+      float v = acc[rf][r] + xp_t[(int64_t)rclamp * 4 * H + col];
+      gbuf[(wave * 32 + row) * 16 + (lane & 15)] = v;
+    }
+  __syncthreads();
+
+  // ---- cell phase: 32 rows x 16 cols, 2 elems per thread ----
+  for (int e = tid; e < 32 * 16; e += 256) {
+    int row = e / 16;
+    int hc = e - row * 16;
+    if (row >= B) continue;
+    float gi = gbuf[(0 * 32 + row) * 16 + hc];
+    float gf = gbuf[(1 * 32 + row) * 16 + hc];
+    float gg = gbuf[(2 * 32 + row) * 16 + hc];
+    float go = gbuf[(3 * 32 + row) * 16 + hc];
+    float i_ = 1.0f / (1.0f + __expf(-gi));
+    float f_ = 1.0f / (1.0f + __expf(-gf));
+    float g_ = tanhf(gg);
+    float o_ = 1.0f / (1.0f + __expf(-go));
+    int64_t hidx = (int64_t)row * H + s16 + hc;
+    float c = f_ * c_in[hidx] + i_ * g_;
+    float tc = tanhf(c);
+    h_out[hidx] = o_ * tc;
+    c_out[hidx] = c;
+    tanhc[hidx] = tc;
+    float* a4 = acts + (int64_t)row * 4 * H + s16 + hc;
+    a4[0] = i_;
+    a4[H] = f_;
+    a4[2 * H] = g_;
+    a4[3 * H] = o_;
+  }
+}
+}  // namespace
+
+// returns false if the geometry is unsupported (caller uses the 2-kernel path)
+bool lstm_step_fused(torch::Tensor xp_t, torch::Tensor h_in, torch::Tensor c_in,
+                     torch::Tensor w_hh, torch::Tensor h_out,
+                     torch::Tensor c_out, torch::Tensor acts,
+                     torch::Tensor tanhc) {
+  int B = (int)h_in.size(0), H = (int)h_in.size(1);
+  if (B > kLstmMaxB || H != 512) return false;
+  constexpr int HH = 512;
+  int lds = (32 * (HH + 4) + 4 * 32 * 16) * sizeof(float);
+  hipLaunchKernelGGL(lstm_step_fused_kernel<HH>, dim3(HH / 16), dim3(256), lds,
+                     cur_stream(), xp_t.data_ptr<float>(),
+                     h_in.data_ptr<float>(), c_in.data_ptr<float>(),
+                     w_hh.data_ptr<float>(), h_out.data_ptr<float>(),
+                     c_out.data_ptr<float>(), acts.data_ptr<float>(),
+                     tanhc.data_ptr<float>(), B);
+  return true;
 }
