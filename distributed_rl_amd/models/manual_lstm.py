@@ -39,12 +39,18 @@ class _ManualLSTMSeq(torch.autograd.Function):
         tanhc = torch.empty(T, B, H, device=dev)
         hs[0] = h0
         cs[0] = c0
-        w_hh_c = w_hh.contiguous()
-        # one-kernel fused step (fp32 MFMA hh-GEMM + cell) when the geometry
-        # allows; otherwise in-place addmm_ + fused cell kernel
-        fused = bool(ext.lstm_step_fused(
-            xp[0].reshape(B, 4 * H), hs[0], cs[0], w_hh_c, hs[1], cs[1],
-            acts[0], tanhc[0]))
+        # The one-kernel fused step (lstm_step_fused: fp32-MFMA hh GEMM +
+        # cell) measures SLOWER than addmm_ + cell at H=512/B=32 — its
+        # 32-block grid underfills the chip (profiles/). Kept behind
+        # DRL_LSTM_FUSED=1 as the starting point for a persistent-grid K5.
+        import os as _os
+
+        fused = False
+        if _os.environ.get("DRL_LSTM_FUSED") == "1":
+            w_hh_c = w_hh.contiguous()
+            fused = bool(ext.lstm_step_fused(
+                xp[0].reshape(B, 4 * H), hs[0], cs[0], w_hh_c, hs[1], cs[1],
+                acts[0], tanhc[0]))
         if fused:
             for t in range(1, T):
                 ext.lstm_step_fused(xp[t].reshape(B, 4 * H), hs[t], cs[t],
