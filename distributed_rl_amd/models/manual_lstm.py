@@ -39,24 +39,18 @@ class _ManualLSTMSeq(torch.autograd.Function):
         tanhc = torch.empty(T, B, H, device=dev)
         hs[0] = h0
         cs[0] = c0
-        # The one-kernel fused step (lstm_step_fused: fp32-MFMA hh GEMM +
-        # cell) measures SLOWER than addmm_ + cell at H=512/B=32 — its
-        # 32-block grid underfills the chip (profiles/). Kept behind
-        # DRL_LSTM_FUSED=1 as the starting point for a persistent-grid K5.
+        # K5 persistent path: the whole recurrence in ONE kernel launch
+        # with agent-scope grid barriers between steps (the per-step loop is
+        # kernel-latency-floor bound — profiles/). Falls back to in-place
+        # addmm_ + fused cell kernel for unsupported geometries.
         import os as _os
 
-        fused = False
-        if _os.environ.get("DRL_LSTM_FUSED") == "1":
-            w_hh_c = w_hh.contiguous()
-            fused = bool(ext.lstm_step_fused(
-                xp[0].reshape(B, 4 * H), hs[0], cs[0], w_hh_c, hs[1], cs[1],
-                acts[0], tanhc[0]))
-        if fused:
-            for t in range(1, T):
-                ext.lstm_step_fused(xp[t].reshape(B, 4 * H), hs[t], cs[t],
-                                    w_hh_c, hs[t + 1], cs[t + 1], acts[t],
-                                    tanhc[t])
-        else:
+        used = False
+        if _os.environ.get("DRL_LSTM_PERSISTENT", "1") == "1":
+            ctr = torch.zeros(1, dtype=torch.int32, device=dev)
+            used = bool(ext.lstm_seq_persistent(
+                xp, hs, cs, w_hh.contiguous(), acts, tanhc, ctr))
+        if not used:
             w_hh_t = w_hh.t()
             for t in range(T):
                 gates = xp[t].reshape(B, 4 * H)

@@ -562,6 +562,9 @@ bool lstm_step_fused(torch::Tensor xp_t, torch::Tensor h_in, torch::Tensor c_in,
                      torch::Tensor w_hh, torch::Tensor h_out,
                      torch::Tensor c_out, torch::Tensor acts,
                      torch::Tensor tanhc);
+bool lstm_seq_persistent(torch::Tensor xp, torch::Tensor hs, torch::Tensor cs,
+                         torch::Tensor w_hh, torch::Tensor acts,
+                         torch::Tensor tanhc, torch::Tensor ctr);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -590,6 +593,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("adam_step", &adam_step, "fused flat Adam (K12)");
   m.def("lstm_step_fused", &lstm_step_fused,
         "one-kernel LSTM timestep: fp32-MFMA hh GEMM + cell (K5)");
+  m.def("lstm_seq_persistent", &lstm_seq_persistent,
+        "whole-sequence persistent LSTM: grid-resident, agent-scope step "
+        "barriers (K5)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -1061,5 +1067,145 @@ bool lstm_step_fused(torch::Tensor xp_t, torch::Tensor h_in, torch::Tensor c_in,
                      w_hh.data_ptr<float>(), h_out.data_ptr<float>(),
                      c_out.data_ptr<float>(), acts.data_ptr<float>(),
                      tanhc.data_ptr<float>(), B);
+  return true;
+}
+
+// K5 (persistent): the whole T-step recurrence in ONE launch. The graphed
+// per-step loop is kernel-latency-floor-bound (~4-5 us/kernel x 2 kernels x
+// T steps, profiles/); here the 32 blocks stay resident and synchronize
+// with an agent-scope release/acquire grid barrier between timesteps
+// (guide §6 G16: every storing wave drains vmcnt, one lane releases +
+// arrives on a monotonic counter, consumers poll relaxed then acquire).
+// Grid = H/16 = 32 blocks — trivially co-resident on 256 CUs; spins are
+// bounded so a mis-launch exits instead of wedging the GPU.
+namespace {
+
+#define DRL_RLX_AGENT __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT
+typedef unsigned int __attribute__((address_space(1))) gu32_t;
+
+template <int H>
+__global__ __launch_bounds__(256) void lstm_seq_persistent_kernel(
+    const float* __restrict__ xp,     // (T, B, 4H)
+    float* __restrict__ hs,           // (T+1, B, H); hs[0] pre-filled
+    float* __restrict__ cs,           // (T+1, B, H); cs[0] pre-filled
+    const float* __restrict__ w_hh,   // (4H, H)
+    float* __restrict__ acts,         // (T, B, 4H)
+    float* __restrict__ tanhc,        // (T, B, H)
+    unsigned int* __restrict__ ctr,   // zeroed before launch
+    int B, int T) {
+  constexpr int PAD = 4;
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* h_lds = reinterpret_cast<float*>(smem);  // [32][H+PAD]
+  float* gbuf = h_lds + 32 * (H + PAD);           // [4][32][16]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;  // gate index
+  const int s16 = blockIdx.x * 16;
+  const int col = wave * H + s16 + (lane & 15);
+  const float* wrow = w_hh + (int64_t)col * H;
+  const int ksub = (lane >> 4) * 4;
+  const unsigned nblocks = gridDim.x;
+
+  for (int t = 0; t < T; ++t) {
+    const float* h_in = hs + (int64_t)t * B * H;
+    const float* c_in = cs + (int64_t)t * B * H;
+    const float* xp_t = xp + (int64_t)t * B * 4 * H;
+    // ---- stage h (zero-padded rows) ----
+    for (int i = tid * 4; i < 32 * H; i += 256 * 4) {
+      int row = i / H;
+      int k = i - row * H;
+      float4 v = {0.f, 0.f, 0.f, 0.f};
+      if (row < B) v = *reinterpret_cast<const float4*>(h_in + row * H + k);
+      *reinterpret_cast<float4*>(h_lds + row * (H + PAD) + k) = v;
+    }
+    __syncthreads();
+    // ---- hh GEMM on fp32 MFMA ----
+    f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+#pragma unroll 4
+    for (int kb = 0; kb < H; kb += 16) {
+      float4 b4 = *reinterpret_cast<const float4*>(wrow + kb + ksub);
+      float4 a0 = *reinterpret_cast<const float4*>(
+          h_lds + (lane & 15) * (H + PAD) + kb + ksub);
+      float4 a1 = *reinterpret_cast<const float4*>(
+          h_lds + (16 + (lane & 15)) * (H + PAD) + kb + ksub);
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        acc[0] = __builtin_amdgcn_mfma_f32_16x16x4f32(a0[j], b4[j], acc[0], 0, 0, 0);
+        acc[1] = __builtin_amdgcn_mfma_f32_16x16x4f32(a1[j], b4[j], acc[1], 0, 0, 0);
+      }
+    }
+    const int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int rf = 0; rf < 2; ++rf)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int row = rf * 16 + crow + r;
+        int rclamp = row < B ? row : 0;
+        gbuf[(wave * 32 + row) * 16 + (lane & 15)] =
+            acc[rf][r] + xp_t[(int64_t)rclamp * 4 * H + col];
+      }
+    __syncthreads();
+    // ---- cell + publish h/c for step t+1 ----
+    float* h_out = hs + (int64_t)(t + 1) * B * H;
+    float* c_out = cs + (int64_t)(t + 1) * B * H;
+    for (int e = tid; e < 32 * 16; e += 256) {
+      int row = e / 16;
+      int hc = e - row * 16;
+      if (row >= B) continue;
+      float gi = gbuf[(0 * 32 + row) * 16 + hc];
+      float gf = gbuf[(1 * 32 + row) * 16 + hc];
+      float gg = gbuf[(2 * 32 + row) * 16 + hc];
+      float go = gbuf[(3 * 32 + row) * 16 + hc];
+      float i_ = 1.0f / (1.0f + __expf(-gi));
+      float f_ = 1.0f / (1.0f + __expf(-gf));
+      float g_ = tanhf(gg);
+      float o_ = 1.0f / (1.0f + __expf(-go));
+      int64_t hidx = (int64_t)row * H + s16 + hc;
+      float c = f_ * c_in[hidx] + i_ * g_;
+      float tc = tanhf(c);
+      h_out[hidx] = o_ * tc;
+      c_out[hidx] = c;
+      tanhc[(int64_t)t * B * H + hidx] = tc;
+      float* a4 = acts + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + s16 + hc;
+      a4[0] = i_;
+      a4[H] = f_;
+      a4[2 * H] = g_;
+      a4[3 * H] = o_;
+    }
+    // ---- grid barrier (G16 recipe): drain stores, release, arrive, poll,
+    // acquire. Monotonic counter: target = nblocks * (t+1).
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // every storing wave
+    __syncthreads();
+    if (tid == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");  // G16 pitfall 12
+      __hip_atomic_fetch_add((gu32_t*)ctr, 1u, DRL_RLX_AGENT);
+      const unsigned target = nblocks * (unsigned)(t + 1);
+      unsigned spins = 0;
+      while (__hip_atomic_load((gu32_t*)ctr, DRL_RLX_AGENT) < target) {
+        __builtin_amdgcn_s_sleep(4);
+        if (++spins > 5000000u) break;  // bounded: exit instead of wedging
+      }
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+  }
+}
+}  // namespace
+
+bool lstm_seq_persistent(torch::Tensor xp, torch::Tensor hs, torch::Tensor cs,
+                         torch::Tensor w_hh, torch::Tensor acts,
+                         torch::Tensor tanhc, torch::Tensor ctr) {
+  int T = (int)xp.size(0), B = (int)xp.size(1), H = (int)hs.size(2);
+  if (B > kLstmMaxB || H != 512) return false;
+  constexpr int HH = 512;
+  int lds = (32 * (HH + 4) + 4 * 32 * 16) * sizeof(float);
+  hipLaunchKernelGGL(lstm_seq_persistent_kernel<HH>, dim3(HH / 16), dim3(256),
+                     lds, cur_stream(), xp.data_ptr<float>(),
+                     hs.data_ptr<float>(), cs.data_ptr<float>(),
+                     w_hh.data_ptr<float>(), acts.data_ptr<float>(),
+                     tanhc.data_ptr<float>(),
+                     (unsigned int*)ctr.data_ptr(), B, T);
   return true;
 }
