@@ -39,14 +39,18 @@ class _ManualLSTMSeq(torch.autograd.Function):
         tanhc = torch.empty(T, B, H, device=dev)
         hs[0] = h0
         cs[0] = c0
-        # K5 persistent path: the whole recurrence in ONE kernel launch
-        # with agent-scope grid barriers between steps (the per-step loop is
-        # kernel-latency-floor bound — profiles/). Falls back to in-place
-        # addmm_ + fused cell kernel for unsupported geometries.
+        # Default: in-place addmm_ + fused cell kernel. Two alternatives
+        # were built and MEASURED SLOWER at H=512/B=32 (profiles/): the
+        # one-kernel fused step (DRL_LSTM_FUSED) and the whole-sequence
+        # persistent kernel with agent-scope grid barriers
+        # (DRL_LSTM_PERSISTENT) — both are issue-bound on a 32-block grid
+        # (128 waves on 1024 SIMDs); the hipBLASLt addmm keeps more of the
+        # chip busy. Revisit with a wider decomposition (split-K + 2-phase
+        # barrier) next round.
         import os as _os
 
         used = False
-        if _os.environ.get("DRL_LSTM_PERSISTENT", "1") == "1":
+        if _os.environ.get("DRL_LSTM_PERSISTENT", "0") == "1":
             ctr = torch.zeros(1, dtype=torch.int32, device=dev)
             used = bool(ext.lstm_seq_persistent(
                 xp, hs, cs, w_hh.contiguous(), acts, tanhc, ctr))
