@@ -58,7 +58,10 @@ class ApexLearner(LearnerBase):
             from ..parallel.precision import MixedPrecisionTrainer
 
             self.model.to(memory_format=torch.channels_last)
-            self.mp = MixedPrecisionTrainer(self.model)
+            # pin the two dueling streams' first-layer params adjacent in
+            # the flat buffer so both streams run as ONE fused GEMM
+            order = self._dueling_param_order()
+            self.mp = MixedPrecisionTrainer(self.model, param_order=order)
             self.net = self.mp.compute  # bf16 forward/backward model
             import copy as _copy
 
@@ -66,9 +69,11 @@ class ApexLearner(LearnerBase):
             for p in self.target.parameters():
                 p.requires_grad_(False)
             # flat target param buffer so hard-sync is ONE device copy
+            # (SAME param ordering as the compute flat buffer)
             from ..parallel.precision import _to_flat, _view_like
 
-            tp = [p for p in self.target.parameters()]
+            tnamed = dict(self.target.named_parameters())
+            tp = [tnamed[n] for n in self.mp._name_order]
             self.flat_tparam = torch.empty(
                 sum(p.numel() for p in tp), dtype=torch.bfloat16,
                 device=self.device,
@@ -80,6 +85,7 @@ class ApexLearner(LearnerBase):
                     self.flat_tparam[off : off + n].copy_(_to_flat(p))
                     p.data = _view_like(self.flat_tparam[off : off + n], p)
                     off += n
+            self._build_fast_forward(order)
         else:
             self.net = self.model
             self.target = self.build_model()
@@ -121,6 +127,70 @@ class ApexLearner(LearnerBase):
             torch.cuda.Stream(self.device) if self.device.type == "cuda" else None
         )
         self._staging: Dict[str, torch.Tensor] = {}
+
+    # ------------------------------------------------------------------
+    # fused dueling-stream forward: both 3136->512 stream GEMMs run as ONE
+    # (B,3136)x(3136,1024) GEMM through adjacent flat-buffer views
+    # ------------------------------------------------------------------
+    def _dueling_param_order(self) -> List[str]:
+        d = getattr(self.model, "_dueling", {})
+        if len(d) != 1:
+            return []
+        (a_node, v_node), = d.values()
+        a_mod = self.model.nodes.get(a_node)
+        v_mod = self.model.nodes.get(v_node)
+        try:
+            ok = (a_mod.body[0].in_features == v_mod.body[0].in_features
+                  and a_mod.body[0].out_features == v_mod.body[0].out_features
+                  and len(a_mod.body) == 4)
+        except (AttributeError, IndexError, TypeError):
+            ok = False
+        if not ok:
+            return []
+        self._fast_nodes = (a_node, v_node)
+        return [
+            f"nodes.{a_node}.body.0.weight", f"nodes.{v_node}.body.0.weight",
+            f"nodes.{a_node}.body.0.bias", f"nodes.{v_node}.body.0.bias",
+        ]
+
+    def _build_fast_forward(self, order: List[str]):
+        self._fast_fwd = None
+        if not order:
+            return
+        import torch.nn.functional as F
+
+        a_node, v_node = self._fast_nodes
+        a_mod = self.net.nodes[a_node]
+        hidden = a_mod.body[0].out_features
+        in_f = a_mod.body[0].in_features
+        # leaf views over the flat compute buffers so autograd deposits
+        # grads straight into flat_cgrad
+        w1 = self.mp.flat_view(order[:2]).view(2 * hidden, in_f
+                                               ).detach().requires_grad_(True)
+        w1.grad = self.mp.flat_view(order[:2], "grad").view_as(w1)
+        b1 = self.mp.flat_view(order[2:]).detach().requires_grad_(True)
+        b1.grad = self.mp.flat_view(order[2:], "grad").view_as(b1)
+        a_head = self.net.nodes[a_node].body[2]
+        v_head = self.net.nodes[v_node].body[2]
+        cnn_node = next(
+            n for n, c in self.cfg.model_info.items()
+            if str(c.get("netCat", "")).upper() == "CNN2D"
+        )
+        cnn = self.net.nodes[cnn_node]
+
+        def fast_fwd(x):
+            feat = cnn(x)
+            h = F.relu(F.linear(feat.to(w1.dtype), w1, b1))
+            adv = a_head(h[:, :hidden])
+            val = v_head(h[:, hidden:])
+            return ops.dueling_head(adv.float(), val.float())
+
+        self._fast_fwd = fast_fwd
+
+    def _online_q(self, x):
+        if getattr(self, "_fast_fwd", None) is not None:
+            return self._fast_fwd(x)
+        return self.net.forward([x])[0]
 
     # ------------------------------------------------------------------
     # ingest: transport -> pinned staging -> device replay (side stream)
@@ -198,9 +268,9 @@ class ApexLearner(LearnerBase):
         dones = data["done"].to(self.device)
         weights = weights.to(self.device)
 
-        q_s = self.net.forward([s])[0]
+        q_s = self._online_q(s) if cuda else self.net.forward([s])[0]
         with torch.no_grad():
-            q_sp_on = self.net.forward([sp])[0]
+            q_sp_on = self._online_q(sp) if cuda else self.net.forward([sp])[0]
             q_sp_tg = self.target.forward([sp])[0]
 
         loss, prio = ops.nstep_dqn_loss(
@@ -291,9 +361,9 @@ class ApexLearner(LearnerBase):
             s = data["state"].permute(0, 3, 1, 2)
             sp = data["next_state"].permute(0, 3, 1, 2)
             actions = data["action"].long()
-            q_s = self.net.forward([s])[0]
+            q_s = self._online_q(s)
             with torch.no_grad():
-                q_sp_on = self.net.forward([sp])[0]
+                q_sp_on = self._online_q(sp)
                 q_sp_tg = self.target.forward([sp])[0]
             loss, prio = ops.nstep_dqn_loss(
                 q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions,

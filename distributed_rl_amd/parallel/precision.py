@@ -76,13 +76,21 @@ class MixedPrecisionTrainer:
     def __init__(self, master: torch.nn.Module,
                  group: Optional["dist.ProcessGroup"] = None,
                  compute_dtype: torch.dtype = torch.bfloat16,
-                 keep_fp32: Optional[Callable[[str, torch.Tensor], bool]] = None):
+                 keep_fp32: Optional[Callable[[str, torch.Tensor], bool]] = None,
+                 param_order: Optional[list] = None):
         self.master = master
         dev = next(master.parameters()).device
         self.compute = copy.deepcopy(master)
         # cast only the non-pinned params of the compute replica
         named_master = [(n, p) for n, p in master.named_parameters()
                         if p.requires_grad]
+        if param_order:
+            # callers may pin chosen params to the FRONT of the flat buffers
+            # (in the given order) so adjacent params can be viewed as one
+            # fused tensor (see flat_view)
+            rank = {n: i for i, n in enumerate(param_order)}
+            named_master.sort(key=lambda np_: rank.get(np_[0], len(rank)))
+        self._name_order = [n for n, _ in named_master]
         named_compute = dict(self.compute.named_parameters())
         lo_c, lo_m, hi_c, hi_m = [], [], [], []
         for name, mp_ in named_master:
@@ -102,6 +110,29 @@ class MixedPrecisionTrainer:
             self.groups.append(_Group(torch.float32, hi_c, hi_m, dev))
         self.pg = group
         self.world = dist.get_world_size(group) if dist.is_initialized() else 1
+
+    def flat_view(self, names: list, of: str = "param") -> torch.Tensor:
+        """Contiguous flat view spanning the given (adjacent) params in the
+        group that holds them. ``of``: 'param' or 'grad'."""
+        named = dict(self.master.named_parameters())
+        ident = {id(p): n for n, p in named.items()}
+        for g in self.groups:
+            offs = {}
+            off = 0
+            for cp, mp_ in zip(g.c_params, g.m_params):
+                n = ident.get(id(mp_))
+                if n is not None:
+                    offs[n] = (off, mp_.numel())
+                off += cp.numel()
+            if all(n in offs for n in names):
+                start = min(offs[n][0] for n in names)
+                end = max(offs[n][0] + offs[n][1] for n in names)
+                assert end - start == sum(offs[n][1] for n in names), (
+                    f"params {names} are not adjacent in the flat buffer"
+                )
+                buf = g.flat_cparam if of == "param" else g.flat_cgrad
+                return buf[start:end]
+        raise KeyError(f"params {names} not found in one group")
 
     # -- convenience views (single-group fast paths used by callers) -------
     @property
