@@ -137,8 +137,10 @@ class ApexLearner(LearnerBase):
         if len(d) != 1:
             return []
         (a_node, v_node), = d.values()
-        a_mod = self.model.nodes.get(a_node)
-        v_mod = self.model.nodes.get(v_node)
+        if a_node not in self.model.nodes or v_node not in self.model.nodes:
+            return []
+        a_mod = self.model.nodes[a_node]
+        v_mod = self.model.nodes[v_node]
         try:
             ok = (a_mod.body[0].in_features == v_mod.body[0].in_features
                   and a_mod.body[0].out_features == v_mod.body[0].out_features
