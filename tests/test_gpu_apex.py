@@ -64,7 +64,7 @@ def test_fast_forward_matches_dag():
     from distributed_rl_amd.config import load_config
 
     learner = ApexLearner(load_config("ape_x"), device=DEV, enable_tb=False,
-                          batch_size=16, replay_capacity=256)
+                          batch_size=16, replay_capacity=1024)
     assert learner._fast_fwd is not None
     x = torch.randint(0, 256, (8, 84, 84, 4), dtype=torch.uint8,
                       device=DEV).permute(0, 3, 1, 2)
