@@ -428,12 +428,23 @@ class ApexLearner(LearnerBase):
         self.wait_memory(warmup_items)
         self.publish_weights(include_target=True)
         last_loss = None
+        stepper = None  # hipGraph-captured once the replay ring is full
         while self.step_count < max_steps:
             t0 = time.perf_counter()
             self.ingest()
             self.time_block("ingest", time.perf_counter() - t0)
             t0 = time.perf_counter()
-            stats = self.step()
+            if stepper is None and self.device.type == "cuda" \
+                    and len(self.replay) >= self.replay.capacity:
+                # n_valid is baked into the captured sample kernel; once the
+                # ring is full it stays at capacity, so capture is safe now
+                try:
+                    stepper = self.make_graphed_step()
+                except Exception as e:  # pragma: no cover
+                    print(f"[APE_X] graph capture failed ({e}); staying eager",
+                          flush=True)
+                    stepper = self.step
+            stats = (stepper or self.step)()
             self.time_block("train", time.perf_counter() - t0)
             last_loss = stats["loss"]
             if self.step_count % self.LOG_EVERY == 0:
