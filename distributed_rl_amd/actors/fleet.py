@@ -7,7 +7,6 @@ from __future__ import annotations
 
 import multiprocessing as mp
 import os
-import signal
 import time
 from typing import List, Optional
 

@@ -126,7 +126,6 @@ class ApexLearner(LearnerBase):
         self._ingest_stream = (
             torch.cuda.Stream(self.device) if self.device.type == "cuda" else None
         )
-        self._staging: Dict[str, torch.Tensor] = {}
 
     # ------------------------------------------------------------------
     # fused dueling-stream forward: both 3136->512 stream GEMMs run as ONE

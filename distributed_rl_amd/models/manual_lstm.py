@@ -78,7 +78,6 @@ class _ManualLSTMSeq(torch.autograd.Function):
         dc = gc_T.contiguous().clone()
         dc_next = torch.empty(B, H, device=dev)
         gout = gout.contiguous()
-        empty = torch.empty(0, device=dev)
         for t in range(T - 1, -1, -1):
             # the per-step output grad is folded into the cell-bwd kernel
             ext.lstm_cell_bwd(dh.contiguous(), gout[t], dc, acts[t], tanhc[t],

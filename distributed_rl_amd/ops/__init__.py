@@ -143,10 +143,12 @@ class _FusedConvFn(torch.autograd.Function):
             # accumulated by atomics, cast to the bf16 channels_last grad.
             ws = torch.zeros(COUT, KH * KW * C, dtype=torch.float32,
                              device=x.device)
-            ext.conv_wrw(x, gout, ws, stride)
+            gb_ws = (torch.zeros(COUT, dtype=torch.float32, device=x.device)
+                     if need_b else torch.empty(0, device=x.device))
+            ext.conv_wrw(x, gout, ws, gb_ws, stride)
             gw = ws.view(COUT, KH, KW, C).permute(0, 3, 1, 2).to(torch.bfloat16)
             if need_b:
-                gb = gout.sum(dim=(0, 2, 3)).to(torch.bfloat16)
+                gb = gb_ws.to(torch.bfloat16)
         if need_x or (need_w and not use_own_wrw):
             if x.dtype == torch.uint8:
                 # NHWC u8 -> bf16 for the aten path (own wrw reads u8 directly)

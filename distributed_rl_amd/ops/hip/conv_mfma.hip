@@ -320,7 +320,7 @@ bool conv_fwd_variant(torch::Tensor in, torch::Tensor weight,
 }
 
 void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
-              int64_t stride);  // defined below
+              torch::Tensor gb_ws, int64_t stride);  // defined below
 bool conv_wrw_supported(int64_t H, int64_t W, int64_t C, int64_t KH,
                         int64_t KW, int64_t S, int64_t COUT, bool u8);
 void tr16_probe(torch::Tensor out, int64_t mode);
@@ -381,6 +381,7 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
     const void* __restrict__ in_v,      // (N,H,W,C) NHWC u8/bf16
     const __bf16* __restrict__ gout,    // (M, COUT) = NHWC grad (relu-masked)
     float* __restrict__ gw_ws,          // (COUT, K) fp32, pre-zeroed
+    float* __restrict__ gb_ws,          // (COUT) fp32, pre-zeroed, or null
     int batch, int mblocks) {
   using G = ConvGeom<H, W, C, KH, KW, S, COUT, U8IN>;
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -459,6 +460,13 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
       }
     }
     __syncthreads();
+    // ---- fused bias grad: ktile-0 blocks fold their gout chunk into gb
+    if (gb_ws && ktile == 0 && tid < COUT) {
+      float s = 0.0f;
+#pragma unroll 8
+      for (int m = 0; m < 32; ++m) s += (float)g_t[m * GSTRIDE + tid];
+      atomicAdd(&gb_ws[tid], s);
+    }
     // ---- fragments via tr16 reads + MFMA: wave w owns kelem block w
     bf16x8 bfrag = tr_frag<ASTRIDE>(a_t, lane, wave);
 #pragma unroll
@@ -484,7 +492,7 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
 struct WrwLaunch {
   int H, W, C, KH, KW, S, COUT;
   bool u8;
-  void (*fn)(const void*, const __bf16*, float*, int, int);
+  void (*fn)(const void*, const __bf16*, float*, float*, int, int);
   int lds_bytes;
   int ktiles;
 };
@@ -523,7 +531,7 @@ bool conv_wrw_supported(int64_t H, int64_t W, int64_t C, int64_t KH, int64_t KW,
 // in: (N,C,H,W) channels_last u8/bf16; gout: (N,COUT,P,Q) channels_last bf16
 // (already relu-masked); gw_ws: (COUT, KH*KW*C) fp32 pre-zeroed.
 void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
-              int64_t stride) {
+              torch::Tensor gb_ws, int64_t stride) {
   TORCH_CHECK(in.is_contiguous(at::MemoryFormat::ChannelsLast));
   TORCH_CHECK(gout.is_contiguous(at::MemoryFormat::ChannelsLast));
   TORCH_CHECK(gw_ws.scalar_type() == torch::kFloat32 && gw_ws.is_contiguous());
@@ -549,10 +557,12 @@ void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
   const int n_chunks = (M + 31) / 32;
   if (mb > n_chunks) mb = n_chunks;
   if (mb < 1) mb = 1;
+  float* gb = (gb_ws.defined() && gb_ws.numel() > 0)
+                  ? gb_ws.data_ptr<float>() : nullptr;
   hipLaunchKernelGGL(L->fn, dim3(L->ktiles, mb), dim3(256), L->lds_bytes,
                      (hipStream_t)at::cuda::getCurrentCUDAStream().stream(),
                      (const void*)in.data_ptr(), (const __bf16*)gout.data_ptr(),
-                     gw_ws.data_ptr<float>(), N, mb);
+                     gw_ws.data_ptr<float>(), gb, N, mb);
 }
 
 // tr16 semantics probe: fill LDS with 0..511, each lane passes base +
