@@ -143,9 +143,12 @@ def test_conv_wrw_vs_aten(ext, shape):
     gout = torch.randn(N, COUT, P, Q, device=DEV).to(torch.bfloat16).to(
         memory_format=torch.channels_last)
     ws = torch.zeros(COUT, KH * KW * C, dtype=torch.float32, device=DEV)
-    ext.conv_wrw(x, gout, ws, S)
+    gb_ws = torch.zeros(COUT, dtype=torch.float32, device=DEV)
+    ext.conv_wrw(x, gout, ws, gb_ws, S)
     torch.cuda.synchronize()
     gw = ws.view(COUT, KH, KW, C).permute(0, 3, 1, 2)
+    gb_ref = gout.float().sum(dim=(0, 2, 3))
+    assert torch.allclose(gb_ws, gb_ref, rtol=2e-2, atol=2e-2)
     _, gw_ref, _ = torch.ops.aten.convolution_backward(
         gout, x, torch.empty(COUT, C, KH, KW, device=DEV, dtype=torch.bfloat16
                              ).to(memory_format=torch.channels_last),
@@ -164,7 +167,7 @@ def test_conv_wrw_u8_input(ext):
     gout = torch.randn(N, COUT, P, Q, device=DEV).to(torch.bfloat16).to(
         memory_format=torch.channels_last)
     ws = torch.zeros(COUT, KH * KH * C, dtype=torch.float32, device=DEV)
-    ext.conv_wrw(x, gout, ws, S)
+    ext.conv_wrw(x, gout, ws, torch.empty(0, device=DEV), S)
     torch.cuda.synchronize()
     gw = ws.view(COUT, KH, KH, C).permute(0, 3, 1, 2)
     xf = (x.float() / 255.0).to(torch.bfloat16)

@@ -82,7 +82,7 @@ def bench_wrw():
         gout = torch.randn(N, COUT, P, Q, device=DEV).to(torch.bfloat16).to(
             memory_format=torch.channels_last)
         ws = torch.zeros(COUT, KH * KW * C, dtype=torch.float32, device=DEV)
-        us = timeit(lambda: (ws.zero_(), ext.conv_wrw(x, gout, ws, S)))
+        us = timeit(lambda: (ws.zero_(), ext.conv_wrw(x, gout, ws, torch.empty(0, device=DEV), S)))
         line = f"wrw {name}: {us:7.1f} us  {flops/us/1e6:7.1f} TF"
         if not u8:
             w = torch.empty(COUT, C, KH, KW, device=DEV, dtype=torch.bfloat16
