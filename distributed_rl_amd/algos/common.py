@@ -62,8 +62,15 @@ class LearnerBase:
     @property
     def writer(self):
         if self._writer is None and self._enable_tb:
-            from torch.utils.tensorboard import SummaryWriter
-
+            try:
+                from torch.utils.tensorboard import SummaryWriter
+            except ImportError:
+                # tensorboard is optional (absent from some images);
+                # console telemetry keeps working
+                print("[telemetry] tensorboard not importable; "
+                      "TB scalars disabled", flush=True)
+                self._enable_tb = False
+                return None
             path = self.cfg.log_dir(self.run_root, self.run_name)
             os.makedirs(path, exist_ok=True)
             self._writer = SummaryWriter(path)
