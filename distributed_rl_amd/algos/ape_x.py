@@ -126,6 +126,7 @@ class ApexLearner(LearnerBase):
         self._ingest_stream = (
             torch.cuda.Stream(self.device) if self.device.type == "cuda" else None
         )
+        self._staging = None  # lazily allocated pinned buffers
 
     # ------------------------------------------------------------------
     # fused dueling-stream forward: both 3136->512 stream GEMMs run as ONE
@@ -196,6 +197,22 @@ class ApexLearner(LearnerBase):
     # ------------------------------------------------------------------
     # ingest: transport -> pinned staging -> device replay (side stream)
     # ------------------------------------------------------------------
+    _STAGE_ROWS = 4096  # persistent pinned staging capacity (C2)
+
+    def _staging_buffers(self):
+        if self._staging is None:
+            from ..replay import make_apex_schema
+
+            wire = make_apex_schema()  # wire format is always NCHW u8
+            self._staging = {
+                name: torch.empty((self._STAGE_ROWS, *shape), dtype=dtype
+                                  ).pin_memory()
+                for name, (shape, dtype) in wire.items()
+            }
+            self._staging["__prio__"] = torch.empty(
+                self._STAGE_ROWS, dtype=torch.float32).pin_memory()
+        return self._staging
+
     def ingest(self) -> int:
         if self.transport is None:
             return 0
@@ -204,25 +221,44 @@ class ApexLearner(LearnerBase):
             return 0
         cols_np, prio_np = got
         n = len(prio_np)
-        cols = {}
-        for name, arr in cols_np.items():
-            t = torch.from_numpy(np.ascontiguousarray(arr))
-            if self.device.type == "cuda":
-                t = t.pin_memory()
-            cols[name] = t
-        prio = torch.from_numpy(np.ascontiguousarray(prio_np))
-        if self._ingest_stream is not None:
+        if self._ingest_stream is None:
+            cols = {k: torch.from_numpy(np.ascontiguousarray(v))
+                    for k, v in cols_np.items()}
+            self.replay.push(cols, torch.from_numpy(
+                np.ascontiguousarray(prio_np)))
+            return n
+        # pinned staging ring: one hipHostMalloc for the process lifetime,
+        # chunked numpy->pinned memcpy + async H2D on the side stream
+        stage = self._staging_buffers()
+        done = 0
+        while done < n:
+            k = min(self._STAGE_ROWS, n - done)
+            for name, arr in cols_np.items():
+                stage[name][:k].numpy()[:] = arr[done : done + k]
+            stage["__prio__"][:k].numpy()[:] = prio_np[done : done + k]
             with torch.cuda.stream(self._ingest_stream):
                 dev_cols = {
-                    k: v.to(self.device, non_blocking=True) for k, v in cols.items()
+                    name: stage[name][:k].to(self.device, non_blocking=True)
+                    for name in cols_np
                 }
                 if self._nhwc:
-                    for k in ("state", "next_state"):
-                        dev_cols[k] = dev_cols[k].permute(0, 2, 3, 1).contiguous()
-                self.replay.push(dev_cols, prio.to(self.device, non_blocking=True))
-            torch.cuda.current_stream(self.device).wait_stream(self._ingest_stream)
-        else:
-            self.replay.push(cols, prio)
+                    for kk in ("state", "next_state"):
+                        dev_cols[kk] = dev_cols[kk].permute(0, 2, 3, 1
+                                                            ).contiguous()
+                if self.state_dtype != torch.uint8:
+                    # fp16-compressed replay stores frames pre-scaled
+                    for kk in ("state", "next_state"):
+                        dev_cols[kk] = dev_cols[kk].to(self.state_dtype) / 255.0
+                self.replay.push(
+                    dev_cols,
+                    stage["__prio__"][:k].to(self.device, non_blocking=True),
+                )
+            # replay state is consumed by the compute stream; order it after
+            # the ingest stream, and keep the staging slice stable until the
+            # H2D completed (the wait covers it)
+            torch.cuda.current_stream(self.device).wait_stream(
+                self._ingest_stream)
+            done += k
         return n
 
     def push_experience(self, cols: Dict[str, torch.Tensor], prio: torch.Tensor):
