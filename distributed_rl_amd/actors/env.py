@@ -17,7 +17,7 @@ Both expose: reset() -> state(4,84,84 u8); step(a) -> (state, reward, done, info
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 

@@ -8,7 +8,7 @@ Each function documents the reference code whose math it reproduces
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 

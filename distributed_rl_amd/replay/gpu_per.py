@@ -12,7 +12,7 @@ train step is hipGraph-capturable.
 
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Optional, Tuple
 
 import torch
 

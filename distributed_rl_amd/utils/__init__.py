@@ -9,7 +9,7 @@ import logging
 import pickle
 import random
 from collections import deque
-from typing import Any, Dict, List, Optional, Sequence
+from typing import Any, Dict, List, Sequence
 
 
 def dumps(obj: Any) -> bytes:
