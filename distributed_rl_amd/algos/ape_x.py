@@ -616,6 +616,7 @@ class ApexPlayer:
         self.alpha = cfg.alpha
         self.action_n = cfg.action_size
         self.local = LocalBuffer(self.n_step, self.gamma)
+        self.report_all_rewards = bool(cfg.get("REPORT_ALL_REWARDS", True))
         self.pending: List[tuple] = []
         self.env_steps = 0
         self.weight_version = -1
@@ -709,8 +710,10 @@ class ApexPlayer:
                 self.pull_weights()
             self._staleness_gate()
             if done:
-                # eval telemetry gate per Player.py:272-277
-                if self.eps < 0.05 or True:
+                # reference gates reward telemetry to near-greedy actors
+                # (Player.py:272-277); report_all_rewards widens it so the
+                # learner's Reward scalar reflects the whole fleet
+                if self.eps < 0.05 or self.report_all_rewards:
                     self.transport.push_reward(self.idx, episode_reward, self.eps)
                 episode_reward = 0.0
                 self.local.clear()

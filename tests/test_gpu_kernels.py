@@ -288,3 +288,23 @@ def test_fused_adam_matches_torch(ext):
     torch.cuda.synchronize()
     assert torch.allclose(p, p_ref.detach(), atol=1e-6), (
         (p - p_ref.detach()).abs().max())
+
+
+def test_sumtree_deep_tree_large_capacity(ext):
+    """Tree depth / large-capacity path (BASELINE config 4 sizes the replay
+    for 288 GB HBM): 4M-slot tree, distribution still proportional."""
+    from distributed_rl_amd.replay.gpu_per import HipSumTreePER
+
+    cap = 4_000_000
+    per = HipSumTreePER(cap, {"x": ((), torch.float32)}, DEV)
+    n = 300_000
+    prios = torch.rand(n, device=DEV) + 0.01
+    per.push({"x": torch.zeros(n, device=DEV)}, prios)
+    torch.cuda.synchronize()
+    assert abs(per.total_priority - float(prios.sum())) / float(prios.sum()) < 1e-3
+    data, idx, w = per.sample(4096, beta=0.4)
+    torch.cuda.synchronize()
+    assert idx.max() < n and idx.min() >= 0
+    # high-priority items sampled more often than low: split-half check
+    hi = prios[idx] > prios.median()
+    assert hi.float().mean() > 0.55
