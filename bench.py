@@ -61,6 +61,8 @@ def build_apex(cfg, device, rank, world, args):
         cfg, device=device, rank=rank, world_size=world, enable_tb=False,
         batch_size=args.batch,
         replay_capacity=args.replay or cfg.replay_memory_len,
+        replay_state_dtype=(torch.float16 if args.replay_dtype == "fp16"
+                            else None),
     )
     attach_reducer(learner)
     cap = learner.replay.capacity
@@ -144,6 +146,10 @@ def main():
     ap.add_argument("--alg", "--cfg", dest="cfg", type=str, default="ape_x")
     ap.add_argument("--graph", type=str, default="auto",
                     choices=["auto", "on", "off"])
+    ap.add_argument("--replay-dtype", type=str, default="u8",
+                    choices=["u8", "fp16"],
+                    help="Ape-X replay frame storage (fp16 = BASELINE "
+                         "config-5 compression option)")
     args = ap.parse_args()
 
     rank, local_rank, world = init_distributed()
@@ -216,7 +222,8 @@ def main():
                 "seq_len": {"APE_X": 4, "IMPALA": cfg.unroll_step,
                             "R2D2": cfg.fixed_trajectory}[cfg.alg],
                 "parallelism": f"dp{max(n_gpus, 1)}",
-                "replay": ("gpu sum-tree PER" if has_cuda else "cpu PER")
+                "replay": (("gpu sum-tree PER" if has_cuda else "cpu PER")
+                           + ("/fp16" if args.replay_dtype == "fp16" else ""))
                 if cfg.alg != "IMPALA" else "uniform fifo",
                 "graph": bool(use_graph),
             },
