@@ -217,7 +217,7 @@ def main():
             if cfg.alg != "R2D2" else "fp32",
             "data": "synthetic",
             "config": {
-                "model": f"{cfg.alg.lower()} Atari net (cfg/{cfg.alg.lower()}.json)",
+                "model": f"{cfg.alg.lower()} net ({os.path.basename(cfg.path)})",
                 "global_batch": learner.batch_size * max(n_gpus, 1),
                 "seq_len": {"APE_X": 4, "IMPALA": cfg.unroll_step,
                             "R2D2": cfg.fixed_trajectory}[cfg.alg],
