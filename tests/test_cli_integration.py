@@ -55,3 +55,85 @@ def test_run_learner_and_actor_cli(tmp_path):
             actor.wait(20)
         except subprocess.TimeoutExpired:
             actor.kill()
+
+
+@pytest.mark.timeout(180)
+def test_clean_transport_cli(tmp_path):
+    """clean_transport.py unlinks a live session's shm segments."""
+    import numpy as np
+    from distributed_rl_amd.actors.transport import (
+        RecordCodec, TransportSession,
+    )
+    from distributed_rl_amd.replay import make_apex_schema
+
+    codec = RecordCodec(make_apex_schema())
+    sess = TransportSession(str(tmp_path / "t"), codec, num_rings=1,
+                            ring_slots=4, weight_capacity=1 << 16, create=True)
+    name = f"drl_{sess.session}_w"
+    out = subprocess.run(
+        [sys.executable, os.path.join(REPO, "clean_transport.py"),
+         "--transport-dir", str(tmp_path / "t")],
+        capture_output=True, text=True, timeout=120,
+    )
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "removed session" in out.stdout
+    from multiprocessing import shared_memory
+
+    with pytest.raises(FileNotFoundError):
+        shared_memory.SharedMemory(name=name)
+
+
+@pytest.mark.timeout(240)
+def test_run_replay_server_cli(tmp_path):
+    """run_replay_server.py serves batches over TCP (3-tier mode)."""
+    import socket
+    import time as _time
+
+    import numpy as np
+    import torch
+
+    from distributed_rl_amd.actors.tcp_transport import TcpActorEndpoint
+    from distributed_rl_amd.actors.transport import RecordCodec
+    from distributed_rl_amd.replay import make_apex_schema
+    from distributed_rl_amd.replay.server import RemoteReplay
+
+    # pick a free port
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    proc = subprocess.Popen(
+        [sys.executable, os.path.join(REPO, "run_replay_server.py"),
+         "--alg", "ape_x", "--port", str(port), "--capacity", "1024"],
+        stdout=subprocess.PIPE, stderr=subprocess.STDOUT, text=True,
+    )
+    try:
+        deadline = _time.time() + 60
+        codec = RecordCodec(make_apex_schema())
+        ep = None
+        while ep is None and _time.time() < deadline:
+            try:
+                ep = TcpActorEndpoint("127.0.0.1", port, codec)
+            except OSError:
+                _time.sleep(0.2)
+        assert ep is not None
+        n = 32
+        cols = {
+            "state": np.zeros((n, 4, 84, 84), np.uint8),
+            "action": np.arange(n, dtype=np.int32) % 6,
+            "reward": np.ones(n, np.float32),
+            "next_state": np.zeros((n, 4, 84, 84), np.uint8),
+            "done": np.zeros(n, np.float32),
+        }
+        ep.push(cols, np.ones(n, np.float32))
+        remote = RemoteReplay("127.0.0.1", port)
+        deadline = _time.time() + 30
+        while len(remote) < n and _time.time() < deadline:
+            _time.sleep(0.1)
+        data, idx, w = remote.sample(8, beta=0.4)
+        assert data["state"].shape == (8, 4, 84, 84)
+        remote.close()
+        ep.close()
+    finally:
+        proc.terminate()
+        proc.wait(15)

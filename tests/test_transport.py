@@ -109,3 +109,31 @@ def test_inproc_pipe():
     assert pipe.drain() is None
     pipe.publish({"count": 3})
     assert pipe.fetch()["count"] == 3
+
+
+def test_learner_endpoint_ring_partitioning(tmp_path):
+    """Learner-DP partitioning: rank r of world w drains rings i % w == r."""
+    codec = RecordCodec(make_apex_schema())
+    sess = TransportSession(str(tmp_path), codec, num_rings=4, ring_slots=8,
+                            weight_capacity=1 << 16, create=True)
+    try:
+        n = 1
+        def push(i, action):
+            cols = {
+                "state": np.zeros((n, 4, 84, 84), np.uint8),
+                "action": np.full(n, action, np.int32),
+                "reward": np.zeros(n, np.float32),
+                "next_state": np.zeros((n, 4, 84, 84), np.uint8),
+                "done": np.zeros(n, np.float32),
+            }
+            ActorEndpoint(sess, i).push(cols, np.ones(n, np.float32))
+        for i in range(4):
+            push(i, action=i)
+        le0 = LearnerEndpoint(sess, rank=0, world_size=2)
+        le1 = LearnerEndpoint(sess, rank=1, world_size=2)
+        c0, _ = le0.drain()
+        c1, _ = le1.drain()
+        assert sorted(c0["action"].tolist()) == [0, 2]
+        assert sorted(c1["action"].tolist()) == [1, 3]
+    finally:
+        sess.close()
