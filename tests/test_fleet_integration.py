@@ -89,3 +89,37 @@ def test_impala_fleet_roundtrip(tmp_path):
     finally:
         fleet.stop()
         session.close()
+
+
+@pytest.mark.timeout(240)
+def test_r2d2_fleet_roundtrip(tmp_path):
+    cfg_path, cfg = _small_cfg_file(
+        tmp_path, alg="r2d2", BATCHSIZE=2, FIXED_TRAJECTORY=16, MEM=4,
+        REPLAY_MEMORY_LEN=128, BUFFER_SIZE=4,
+    )
+    tdir = str(tmp_path / "transport")
+    schema, with_prio = get_wire_schema(cfg)
+    codec = RecordCodec(schema, with_priority=with_prio)
+    session = TransportSession(tdir, codec, num_rings=2, ring_slots=64,
+                               create=True)
+    fleet = ActorFleet(cfg_path, 2, tdir, env_kind="synthetic",
+                       max_env_steps=120, respawn_on_exit=False)
+    try:
+        endpoint = LearnerEndpoint(session)
+        learner = get_learner_cls("R2D2")(
+            cfg, device="cpu", transport=endpoint, enable_tb=False,
+        )
+        learner.publish_weights(include_target=True)
+        fleet.start()
+        got = 0
+        t0 = time.time()
+        while got < 4 and time.time() - t0 < 180:
+            got += learner.ingest()
+            time.sleep(0.1)
+        assert got >= 4, f"only {got} sequences arrived"
+        stats = learner.step()
+        assert float(stats["loss"]) == float(stats["loss"])
+        fleet.join(60)
+    finally:
+        fleet.stop()
+        session.close()
