@@ -402,6 +402,7 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
   f32x4 acc[CBLK];
 #pragma unroll
   for (int cb = 0; cb < CBLK; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
+  float gb_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
 
   // staging assignment: 8-elem pieces, plain row-major destination
   const int sm = tid >> 3;            // sample row 0..31
@@ -443,7 +444,10 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
       }
       *reinterpret_cast<bf16x8*>(a_t + sm * ASTRIDE + sk8) = v;
     }
-    // ---- stage gout chunk -> g_t[m][cout] (coalesced, zero-padded)
+    // ---- stage gout chunk -> g_t[m][cout] (coalesced, zero-padded);
+    // ktile-0 blocks also fold the bias grad into per-thread registers
+    // (one atomic per thread at kernel end — per-chunk atomics serialized
+    // badly on the small gb array, profiles/)
     {
       constexpr int PIECES = COUT / 8;
       for (int piece = tid; piece < 32 * PIECES; piece += 256) {
@@ -456,17 +460,14 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
 #pragma unroll
           for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
         }
+        if (gb_ws && ktile == 0) {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) gb_acc[j] += (float)v[j];
+        }
         *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
       }
     }
     __syncthreads();
-    // ---- fused bias grad: ktile-0 blocks fold their gout chunk into gb
-    if (gb_ws && ktile == 0 && tid < COUT) {
-      float s = 0.0f;
-#pragma unroll 8
-      for (int m = 0; m < 32; ++m) s += (float)g_t[m * GSTRIDE + tid];
-      atomicAdd(&gb_ws[tid], s);
-    }
     // ---- fragments via tr16 reads + MFMA: wave w owns kelem block w
     bf16x8 bfrag = tr_frag<ASTRIDE>(a_t, lane, wave);
 #pragma unroll
@@ -478,7 +479,16 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue: atomic-accumulate partials into the fp32 workspace
+  // ---- epilogue: atomic-accumulate partials into the fp32 workspaces
+  if (gb_ws && ktile == 0) {
+    constexpr int PIECES = COUT / 8;
+    if (tid < 32 * PIECES) {
+      const int c0 = (tid % PIECES) * 8;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (gb_acc[j] != 0.0f) atomicAdd(&gb_ws[c0 + j], gb_acc[j]);
+    }
+  }
   const int kout = ktile * 64 + wave * 16 + (lane & 15);
 #pragma unroll
   for (int cb = 0; cb < CBLK; ++cb)
