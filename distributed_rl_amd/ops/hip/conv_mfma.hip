@@ -450,21 +450,38 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
     // badly on the small gb array, profiles/)
     {
       constexpr int PIECES = COUT / 8;
-      for (int piece = tid; piece < 32 * PIECES; piece += 256) {
-        const int m = m0 + piece / PIECES;
-        const int c0 = (piece % PIECES) * 8;
-        bf16x8 v;
-        if (m < M) {
-          v = *reinterpret_cast<const bf16x8*>(gout + (int64_t)m * COUT + c0);
-        } else {
+      // the gb condition is hoisted OUTSIDE the load loop: a per-element
+      // runtime select inside an unrolled load loop makes hipcc branch
+      // around each load and wait vmcnt(0) per element (guide trap 4c —
+      // measured 2.4x whole-step regression before hoisting)
+      if (gb_ws && ktile == 0) {
+        for (int piece = tid; piece < 32 * PIECES; piece += 256) {
+          const int m = m0 + piece / PIECES;
+          const int c0 = (piece % PIECES) * 8;
+          bf16x8 v;
+          if (m < M) {
+            v = *reinterpret_cast<const bf16x8*>(gout + (int64_t)m * COUT + c0);
+          } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
-        }
-        if (gb_ws && ktile == 0) {
+            for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
+          }
 #pragma unroll
           for (int j = 0; j < 8; ++j) gb_acc[j] += (float)v[j];
+          *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
         }
-        *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
+      } else {
+        for (int piece = tid; piece < 32 * PIECES; piece += 256) {
+          const int m = m0 + piece / PIECES;
+          const int c0 = (piece % PIECES) * 8;
+          bf16x8 v;
+          if (m < M) {
+            v = *reinterpret_cast<const bf16x8*>(gout + (int64_t)m * COUT + c0);
+          } else {
+#pragma unroll
+            for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
+          }
+          *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
+        }
       }
     }
     __syncthreads();
