@@ -381,7 +381,6 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
     const void* __restrict__ in_v,      // (N,H,W,C) NHWC u8/bf16
     const __bf16* __restrict__ gout,    // (M, COUT) = NHWC grad (relu-masked)
     float* __restrict__ gw_ws,          // (COUT, K) fp32, pre-zeroed
-    float* __restrict__ gb_ws,          // (COUT) fp32, pre-zeroed, or null
     int batch, int mblocks) {
   using G = ConvGeom<H, W, C, KH, KW, S, COUT, U8IN>;
   extern __shared__ __attribute__((aligned(16))) char smem[];
@@ -402,7 +401,6 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
   f32x4 acc[CBLK];
 #pragma unroll
   for (int cb = 0; cb < CBLK; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
-  float gb_acc[8] = {0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f, 0.f};
 
   // staging assignment: 8-elem pieces, plain row-major destination
   const int sm = tid >> 3;            // sample row 0..31
@@ -450,38 +448,17 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
     // badly on the small gb array, profiles/)
     {
       constexpr int PIECES = COUT / 8;
-      // the gb condition is hoisted OUTSIDE the load loop: a per-element
-      // runtime select inside an unrolled load loop makes hipcc branch
-      // around each load and wait vmcnt(0) per element (guide trap 4c —
-      // measured 2.4x whole-step regression before hoisting)
-      if (gb_ws && ktile == 0) {
-        for (int piece = tid; piece < 32 * PIECES; piece += 256) {
-          const int m = m0 + piece / PIECES;
-          const int c0 = (piece % PIECES) * 8;
-          bf16x8 v;
-          if (m < M) {
-            v = *reinterpret_cast<const bf16x8*>(gout + (int64_t)m * COUT + c0);
-          } else {
+      for (int piece = tid; piece < 32 * PIECES; piece += 256) {
+        const int m = m0 + piece / PIECES;
+        const int c0 = (piece % PIECES) * 8;
+        bf16x8 v;
+        if (m < M) {
+          v = *reinterpret_cast<const bf16x8*>(gout + (int64_t)m * COUT + c0);
+        } else {
 #pragma unroll
-            for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
-          }
-#pragma unroll
-          for (int j = 0; j < 8; ++j) gb_acc[j] += (float)v[j];
-          *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
+          for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
         }
-      } else {
-        for (int piece = tid; piece < 32 * PIECES; piece += 256) {
-          const int m = m0 + piece / PIECES;
-          const int c0 = (piece % PIECES) * 8;
-          bf16x8 v;
-          if (m < M) {
-            v = *reinterpret_cast<const bf16x8*>(gout + (int64_t)m * COUT + c0);
-          } else {
-#pragma unroll
-            for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
-          }
-          *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
-        }
+        *reinterpret_cast<bf16x8*>(g_t + (piece / PIECES) * GSTRIDE + c0) = v;
       }
     }
     __syncthreads();
@@ -496,16 +473,7 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
     __syncthreads();
   }
 
-  // ---- epilogue: atomic-accumulate partials into the fp32 workspaces
-  if (gb_ws && ktile == 0) {
-    constexpr int PIECES = COUT / 8;
-    if (tid < 32 * PIECES) {
-      const int c0 = (tid % PIECES) * 8;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        if (gb_acc[j] != 0.0f) atomicAdd(&gb_ws[c0 + j], gb_acc[j]);
-    }
-  }
+  // ---- epilogue: atomic-accumulate partials into the fp32 workspace
   const int kout = ktile * 64 + wave * 16 + (lane & 15);
 #pragma unroll
   for (int cb = 0; cb < CBLK; ++cb)
@@ -519,7 +487,7 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
 struct WrwLaunch {
   int H, W, C, KH, KW, S, COUT;
   bool u8;
-  void (*fn)(const void*, const __bf16*, float*, float*, int, int);
+  void (*fn)(const void*, const __bf16*, float*, int, int);
   int lds_bytes;
   int ktiles;
 };
@@ -543,6 +511,33 @@ static const WrwLaunch kWrwLaunches[] = {
     make_wrw<84, 84, 4, 8, 8, 4, 16, false>(),
     make_wrw<20, 20, 16, 4, 4, 2, 32, false>(),
 };
+
+// bias grad: column sum of the (M, COUT) relu-masked gout — torch's
+// strided bf16 reduce on channels_last costs ~11 us; this is one pass with
+// a block-level LDS reduction and COUT atomics per block.
+template <int COUT>
+__global__ __launch_bounds__(256) void colsum_bf16_kernel(
+    const __bf16* __restrict__ x, int64_t M, float* __restrict__ out) {
+  __shared__ float part[256 > COUT ? 256 : COUT];
+  constexpr int GROUPS = 256 / COUT;  // COUT in {16,32,64}
+  const int tid = threadIdx.x;
+  const int col = tid % COUT;
+  const int grp = tid / COUT;
+  float acc = 0.0f;
+  if (grp < GROUPS) {
+    for (int64_t m = blockIdx.x * GROUPS + grp; m < M;
+         m += (int64_t)gridDim.x * GROUPS)
+      acc += (float)x[m * COUT + col];
+  }
+  part[tid] = acc;
+  __syncthreads();
+  if (tid < COUT) {
+    float s = 0.0f;
+#pragma unroll
+    for (int g = 0; g < GROUPS; ++g) s += part[g * COUT + col];
+    atomicAdd(&out[col], s);
+  }
+}
 
 }  // namespace
 
@@ -584,12 +579,33 @@ void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
   const int n_chunks = (M + 31) / 32;
   if (mb > n_chunks) mb = n_chunks;
   if (mb < 1) mb = 1;
-  float* gb = (gb_ws.defined() && gb_ws.numel() > 0)
-                  ? gb_ws.data_ptr<float>() : nullptr;
   hipLaunchKernelGGL(L->fn, dim3(L->ktiles, mb), dim3(256), L->lds_bytes,
                      (hipStream_t)at::cuda::getCurrentCUDAStream().stream(),
                      (const void*)in.data_ptr(), (const __bf16*)gout.data_ptr(),
-                     gw_ws.data_ptr<float>(), gb, N, mb);
+                     gw_ws.data_ptr<float>(), N, mb);
+  if (gb_ws.defined() && gb_ws.numel() > 0) {
+    const int64_t Mrows = (int64_t)N * P * Q;
+    auto st = (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+    switch (COUT) {
+      case 16:
+        hipLaunchKernelGGL(colsum_bf16_kernel<16>, dim3(256), dim3(256), 0, st,
+                           (const __bf16*)gout.data_ptr(), Mrows,
+                           gb_ws.data_ptr<float>());
+        break;
+      case 32:
+        hipLaunchKernelGGL(colsum_bf16_kernel<32>, dim3(256), dim3(256), 0, st,
+                           (const __bf16*)gout.data_ptr(), Mrows,
+                           gb_ws.data_ptr<float>());
+        break;
+      case 64:
+        hipLaunchKernelGGL(colsum_bf16_kernel<64>, dim3(256), dim3(256), 0, st,
+                           (const __bf16*)gout.data_ptr(), Mrows,
+                           gb_ws.data_ptr<float>());
+        break;
+      default:
+        TORCH_CHECK(false, "no colsum instantiation for COUT=", COUT);
+    }
+  }
 }
 
 // tr16 semantics probe: fill LDS with 0..511, each lane passes base +
