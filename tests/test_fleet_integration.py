@@ -123,3 +123,44 @@ def test_r2d2_fleet_roundtrip(tmp_path):
     finally:
         fleet.stop()
         session.close()
+
+
+@pytest.mark.timeout(240)
+def test_fleet_respawns_killed_actor(tmp_path):
+    """Failure handling: a killed actor process is respawned
+    by the supervisor and the fleet returns to full strength."""
+    cfg_path, cfg = _small_cfg_file(tmp_path)
+    tdir = str(tmp_path / "transport")
+    schema, with_prio = get_wire_schema(cfg)
+    codec = RecordCodec(schema, with_priority=with_prio)
+    session = TransportSession(tdir, codec, num_rings=2, ring_slots=64,
+                               create=True)
+    fleet = ActorFleet(cfg_path, 2, tdir, env_kind="synthetic",
+                       max_env_steps=1 << 30, respawn_on_exit=True)
+    try:
+        fleet.start()
+        t0 = time.time()
+        while fleet.alive_count() < 2 and time.time() - t0 < 60:
+            time.sleep(0.2)
+        assert fleet.alive_count() == 2
+        victim = fleet.procs[0]
+        victim.kill()
+        victim.join(20)
+        # one supervision pass replaces it
+        import threading
+
+        th = threading.Thread(target=fleet.supervise, kwargs={"poll_s": 0.5},
+                              daemon=True)
+        th.start()
+        t0 = time.time()
+        while time.time() - t0 < 60:
+            p0 = fleet.procs[0]
+            if p0 is not None and p0.is_alive() and p0.pid != victim.pid:
+                break
+            time.sleep(0.2)
+        p0 = fleet.procs[0]
+        assert p0 is not None and p0.is_alive() and p0.pid != victim.pid
+    finally:
+        fleet.respawn = False
+        fleet.stop()
+        session.close()
