@@ -126,8 +126,12 @@ class SpscRing:
         head, tail = self.head, self.tail
         free = self.slots - (head - tail)
         n = min(len(rows), free)
-        for i in range(n):
-            self._data[(head + i) % self.slots] = raw[i]
+        if n > 0:
+            start = head % self.slots
+            first = min(n, self.slots - start)
+            self._data[start : start + first] = raw[:first]
+            if n > first:
+                self._data[: n - first] = raw[first:n]
         # payload before head (publish)
         self._set_u64(0, head + n)
         if n < len(rows):
@@ -141,8 +145,11 @@ class SpscRing:
         if n <= 0:
             return None
         out = np.empty((n, self.record_size), dtype=np.uint8)
-        for i in range(n):
-            out[i] = self._data[(tail + i) % self.slots]
+        start = tail % self.slots
+        first = min(n, self.slots - start)
+        out[:first] = self._data[start : start + first]
+        if n > first:
+            out[first:] = self._data[: n - first]
         self._set_u64(8, tail + n)
         return out
 
