@@ -264,6 +264,18 @@ class ImpalaLearner(LearnerBase):
                 )
                 self.log_scalar("training_Time", dt)
                 self.log_scalar("Norm of Gradient", float(self.model.calculateNorm()))
+            if mean_r is not None:
+                self._last_reward = mean_r
+            if self.rank == 0 and self.step_count % 500 == 0:
+                print(
+                    f"[IMPALA] step={self.step_count} "
+                    f"loss={float(stats['loss']):.5f} "
+                    f"entropy={float(stats['entropy']):.3f} "
+                    f"value={float(stats['value']):.3f} "
+                    f"reward={getattr(self, '_last_reward', float('nan')):.1f} "
+                    f"replay={len(self.replay)}",
+                    flush=True,
+                )
             if self.step_count % self.CKPT_EVERY == 0:
                 self.save_checkpoint()
         return None
