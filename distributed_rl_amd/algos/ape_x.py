@@ -310,9 +310,10 @@ class ApexLearner(LearnerBase):
             q_sp_on = self._online_q(sp) if cuda else self.net.forward([sp])[0]
             q_sp_tg = self.target.forward([sp])[0]
 
-        loss, prio = ops.nstep_dqn_loss(
-            q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions, rewards,
+        loss, prio, qmean = ops.nstep_dqn_loss(
+            q_s, q_sp_on, q_sp_tg, actions, rewards,
             dones, weights, self.gamma, self.n_step, self.alpha,
+            with_value_stat=True,
         )
         if self.mp is not None:
             self.mp.zero_grads()
@@ -327,7 +328,7 @@ class ApexLearner(LearnerBase):
                 self.reducer.all_reduce()
             self.optim.step()
         self.replay.update(idx, prio)
-        return {"loss": loss.detach(), "value": q_s.detach().float().max(1).values.mean()}
+        return {"loss": loss.detach(), "value": qmean}
 
     def sync_target(self):
         """Hard target sync (APE_X/Learner.py:204-208, tau=1)."""
@@ -402,15 +403,14 @@ class ApexLearner(LearnerBase):
             with torch.no_grad():
                 q_sp_on = self._online_q(sp)
                 q_sp_tg = self.target.forward([sp])[0]
-            loss, prio = ops.nstep_dqn_loss(
-                q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions,
+            loss, prio, qmean = ops.nstep_dqn_loss(
+                q_s, q_sp_on, q_sp_tg, actions,
                 data["reward"], data["done"], s_w, self.gamma, self.n_step,
-                self.alpha,
+                self.alpha, with_value_stat=True,
             )
             mp.zero_grads()
             loss.backward()
-            static_out = {"loss": loss.detach(),
-                          "value": q_s.detach().float().max(1).values.mean()}
+            static_out = {"loss": loss.detach(), "value": qmean}
         g2 = torch.cuda.CUDAGraph()
         with torch.cuda.graph(g2, pool=g1.pool()):
             mp.flat_mgrad.copy_(mp.flat_cgrad)

@@ -222,41 +222,49 @@ class _DQNLossFn(torch.autograd.Function):
         B, A = q_s.shape
         dev = q_s.device
         loss = torch.zeros((), dtype=torch.float32, device=dev)
+        qmean = torch.zeros((), dtype=torch.float32, device=dev)
         prio = torch.empty(B, dtype=torch.float32, device=dev)
         coef = torch.empty(B, dtype=torch.float32, device=dev)
         ext.dqn_loss_fwd(
             q_s.contiguous(), q_sp_on.contiguous(), q_sp_tg.contiguous(),
             actions.contiguous(), rewards.contiguous(), dones.contiguous(),
             weights.contiguous(), float(gamma_n), float(alpha),
-            loss.view(1), prio, coef,
+            loss.view(1), prio, coef, qmean.view(1),
         )
         ctx.save_for_backward(coef, actions)
         ctx.shape = (B, A)
-        ctx.mark_non_differentiable(prio)
-        return loss, prio
+        ctx.in_dtype = q_s.dtype
+        ctx.mark_non_differentiable(prio, qmean)
+        return loss, prio, qmean
 
     @staticmethod
-    def backward(ctx, gout, _gprio):
+    def backward(ctx, gout, _gprio, _gqm):
         coef, actions = ctx.saved_tensors
         B, A = ctx.shape
-        grad_q = torch.empty(B, A, dtype=torch.float32, device=coef.device)
+        grad_q = torch.empty(B, A, dtype=ctx.in_dtype, device=coef.device)
         hip_ext().dqn_loss_bwd(coef, actions, gout.reshape(1).contiguous(), grad_q)
         return grad_q, None, None, None, None, None, None, None, None
 
 
 def nstep_dqn_loss(q_s, q_sp_on, q_sp_tg, actions, rewards, dones, weights,
-                   gamma: float, n_step: int, alpha: float):
-    """Returns (loss, new_priorities). q_s must be fp32 (cast before call)."""
+                   gamma: float, n_step: int, alpha: float,
+                   with_value_stat: bool = False):
+    """Returns (loss, new_priorities[, mean-of-row-max value stat]).
+    On GPU q tensors may be bf16 or fp32 (the kernel reads both)."""
     if _use_hip(q_s):
-        return _DQNLossFn.apply(
-            q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions.long(),
+        loss, prio, qmean = _DQNLossFn.apply(
+            q_s, q_sp_on, q_sp_tg, actions.long(),
             rewards.float(), dones.float(), weights.float(),
             gamma ** n_step, alpha,
         )
-    return torch_ref.nstep_dqn_loss(
+        return (loss, prio, qmean) if with_value_stat else (loss, prio)
+    loss, prio = torch_ref.nstep_dqn_loss(
         q_s.float(), q_sp_on.float(), q_sp_tg.float(), actions, rewards.float(),
         dones.float(), weights.float(), gamma, n_step, alpha
     )
+    if with_value_stat:
+        return loss, prio, q_s.detach().float().max(1).values.mean()
+    return loss, prio
 
 
 # ---------------------------------------------------------------------------

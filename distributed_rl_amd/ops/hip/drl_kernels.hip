@@ -212,25 +212,38 @@ __global__ void bump_seed_kernel(unsigned long long* seed) {
 //   prio   = (|td| + 1e-7)^alpha ;  loss = 0.5*mean(w * td^2)
 // ---------------------------------------------------------------------------
 
+template <typename T>
+__device__ __forceinline__ float drl_ld(const T* p, int64_t i) {
+  return (float)p[i];
+}
+
+template <typename T>
 __global__ void dqn_loss_fwd_kernel(
-    const float* __restrict__ q_s, const float* __restrict__ q_sp_on,
-    const float* __restrict__ q_sp_tg, const int64_t* __restrict__ act,
+    const T* __restrict__ q_s, const T* __restrict__ q_sp_on,
+    const T* __restrict__ q_sp_tg, const int64_t* __restrict__ act,
     const float* __restrict__ rew, const float* __restrict__ done,
     const float* __restrict__ w, int B, int A, float gamma_n, float alpha,
     float* __restrict__ loss_out /*pre-zeroed scalar*/,
-    float* __restrict__ prio_out, float* __restrict__ grad_coef) {
+    float* __restrict__ prio_out, float* __restrict__ grad_coef,
+    float* __restrict__ qmax_out /*pre-zeroed scalar: mean of row maxes*/) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
-  float contrib = 0.0f;
+  float contrib = 0.0f, qm = 0.0f;
   if (i < B) {
-    const float* row = q_sp_on + (int64_t)i * A;
+    const T* row = q_sp_on + (int64_t)i * A;
     int a_star = 0;
-    float best = row[0];
+    float best = drl_ld(row, 0);
     for (int a = 1; a < A; ++a) {
-      float v = row[a];
+      float v = drl_ld(row, a);
       if (v > best) { best = v; a_star = a; }
     }
-    float target = rew[i] + gamma_n * q_sp_tg[(int64_t)i * A + a_star] * (1.0f - done[i]);
-    float q = q_s[(int64_t)i * A + act[i]];
+    float target = rew[i] + gamma_n * drl_ld(q_sp_tg, (int64_t)i * A + a_star)
+                                * (1.0f - done[i]);
+    float q = drl_ld(q_s, (int64_t)i * A + act[i]);
+    // value telemetry: mean over rows of max_a Q(s,a)
+    const T* srow = q_s + (int64_t)i * A;
+    float smax = drl_ld(srow, 0);
+    for (int a = 1; a < A; ++a) smax = fmaxf(smax, drl_ld(srow, a));
+    qm = smax / B;
     float raw = target - q;
     float td = fminf(1.0f, fmaxf(-1.0f, raw));
     prio_out[i] = __powf(fabsf(td) + 1e-7f, alpha);
@@ -240,19 +253,26 @@ __global__ void dqn_loss_fwd_kernel(
     contrib = 0.5f * w[i] * td * td * invB;
   }
   // wave reduce then one atomic per wave (guide G12)
-  for (int off = 32; off > 0; off >>= 1) contrib += __shfl_down(contrib, off, 64);
-  if ((threadIdx.x & 63) == 0 && contrib != 0.0f) atomicAdd(loss_out, contrib);
+  for (int off = 32; off > 0; off >>= 1) {
+    contrib += __shfl_down(contrib, off, 64);
+    qm += __shfl_down(qm, off, 64);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    if (contrib != 0.0f) atomicAdd(loss_out, contrib);
+    if (qm != 0.0f) atomicAdd(qmax_out, qm);
+  }
 }
 
+template <typename T>
 __global__ void dqn_loss_bwd_kernel(const float* __restrict__ grad_coef,
                                     const int64_t* __restrict__ act,
                                     const float* __restrict__ gout,
-                                    int B, int A, float* __restrict__ grad_q) {
+                                    int B, int A, T* __restrict__ grad_q) {
   int j = blockIdx.x * blockDim.x + threadIdx.x;
   if (j >= B * A) return;
   int i = j / A;
   int a = j - i * A;
-  grad_q[j] = (a == (int)act[i]) ? -grad_coef[i] * gout[0] : 0.0f;
+  grad_q[j] = (T)((a == (int)act[i]) ? -grad_coef[i] * gout[0] : 0.0f);
 }
 
 // ---------------------------------------------------------------------------
@@ -467,24 +487,47 @@ void dqn_loss_fwd(torch::Tensor q_s, torch::Tensor q_sp_on, torch::Tensor q_sp_t
                   torch::Tensor act, torch::Tensor rew, torch::Tensor done,
                   torch::Tensor w, double gamma_n, double alpha,
                   torch::Tensor loss_out, torch::Tensor prio_out,
-                  torch::Tensor grad_coef) {
+                  torch::Tensor grad_coef, torch::Tensor qmax_out) {
   int B = (int)q_s.size(0), A = (int)q_s.size(1);
-  hipLaunchKernelGGL(dqn_loss_fwd_kernel, dim3(ceil_div(B, kBlock)), dim3(kBlock), 0,
-                     cur_stream(), q_s.data_ptr<float>(), q_sp_on.data_ptr<float>(),
-                     q_sp_tg.data_ptr<float>(), act.data_ptr<int64_t>(),
-                     rew.data_ptr<float>(), done.data_ptr<float>(),
-                     w.data_ptr<float>(), B, A, (float)gamma_n, (float)alpha,
-                     loss_out.data_ptr<float>(), prio_out.data_ptr<float>(),
-                     grad_coef.data_ptr<float>());
+  if (q_s.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(dqn_loss_fwd_kernel<__bf16>, dim3(ceil_div(B, kBlock)),
+                       dim3(kBlock), 0, cur_stream(),
+                       (const __bf16*)q_s.data_ptr(),
+                       (const __bf16*)q_sp_on.data_ptr(),
+                       (const __bf16*)q_sp_tg.data_ptr(),
+                       act.data_ptr<int64_t>(), rew.data_ptr<float>(),
+                       done.data_ptr<float>(), w.data_ptr<float>(), B, A,
+                       (float)gamma_n, (float)alpha, loss_out.data_ptr<float>(),
+                       prio_out.data_ptr<float>(), grad_coef.data_ptr<float>(),
+                       qmax_out.data_ptr<float>());
+  } else {
+    hipLaunchKernelGGL(dqn_loss_fwd_kernel<float>, dim3(ceil_div(B, kBlock)),
+                       dim3(kBlock), 0, cur_stream(), q_s.data_ptr<float>(),
+                       q_sp_on.data_ptr<float>(), q_sp_tg.data_ptr<float>(),
+                       act.data_ptr<int64_t>(), rew.data_ptr<float>(),
+                       done.data_ptr<float>(), w.data_ptr<float>(), B, A,
+                       (float)gamma_n, (float)alpha, loss_out.data_ptr<float>(),
+                       prio_out.data_ptr<float>(), grad_coef.data_ptr<float>(),
+                       qmax_out.data_ptr<float>());
+  }
 }
 
 void dqn_loss_bwd(torch::Tensor grad_coef, torch::Tensor act, torch::Tensor gout,
                   torch::Tensor grad_q) {
   int B = (int)grad_q.size(0), A = (int)grad_q.size(1);
-  hipLaunchKernelGGL(dqn_loss_bwd_kernel, dim3(ceil_div((int64_t)B * A, kBlock)),
-                     dim3(kBlock), 0, cur_stream(), grad_coef.data_ptr<float>(),
-                     act.data_ptr<int64_t>(), gout.data_ptr<float>(), B, A,
-                     grad_q.data_ptr<float>());
+  if (grad_q.scalar_type() == torch::kBFloat16) {
+    hipLaunchKernelGGL(dqn_loss_bwd_kernel<__bf16>,
+                       dim3(ceil_div((int64_t)B * A, kBlock)), dim3(kBlock), 0,
+                       cur_stream(), grad_coef.data_ptr<float>(),
+                       act.data_ptr<int64_t>(), gout.data_ptr<float>(), B, A,
+                       (__bf16*)grad_q.data_ptr());
+  } else {
+    hipLaunchKernelGGL(dqn_loss_bwd_kernel<float>,
+                       dim3(ceil_div((int64_t)B * A, kBlock)), dim3(kBlock), 0,
+                       cur_stream(), grad_coef.data_ptr<float>(),
+                       act.data_ptr<int64_t>(), gout.data_ptr<float>(), B, A,
+                       grad_q.data_ptr<float>());
+  }
 }
 
 void value_rescale(torch::Tensor x, torch::Tensor y, double eps) {

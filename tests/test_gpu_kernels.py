@@ -308,3 +308,30 @@ def test_sumtree_deep_tree_large_capacity(ext):
     # high-priority items sampled more often than low: split-half check
     hi = prios[idx] > prios.median()
     assert hi.float().mean() > 0.55
+
+
+def test_dqn_loss_bf16_inputs(ext):
+    """The fused loss reads bf16 Q tensors directly (no cast kernels)."""
+    from distributed_rl_amd import ops
+    from distributed_rl_amd.ops import torch_ref
+
+    torch.manual_seed(31)
+    B, A = 128, 6
+    q_s = torch.randn(B, A, device=DEV).to(torch.bfloat16).requires_grad_(True)
+    q_on = torch.randn(B, A, device=DEV).to(torch.bfloat16)
+    q_tg = torch.randn(B, A, device=DEV).to(torch.bfloat16)
+    act = torch.randint(0, A, (B,), device=DEV)
+    rew = torch.randn(B, device=DEV)
+    done = torch.zeros(B, device=DEV)
+    w = torch.rand(B, device=DEV)
+    loss, prio, qmean = ops.nstep_dqn_loss(q_s, q_on, q_tg, act, rew, done, w,
+                                           0.99, 3, 0.6, with_value_stat=True)
+    loss.backward()
+    assert q_s.grad.dtype == torch.bfloat16
+    loss_ref, prio_ref = torch_ref.nstep_dqn_loss(
+        q_s.detach().float().cpu(), q_on.float().cpu(), q_tg.float().cpu(),
+        act.cpu(), rew.cpu(), done.cpu(), w.cpu(), 0.99, 3, 0.6)
+    assert abs(loss.item() - loss_ref.item()) < 1e-3
+    assert torch.allclose(prio.cpu(), prio_ref, atol=1e-3)
+    qm_ref = q_s.detach().float().max(1).values.mean()
+    assert abs(qmean.item() - qm_ref.item()) < 1e-3

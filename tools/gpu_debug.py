@@ -63,10 +63,11 @@ def main():
     loss = torch.zeros(1, device=dev)
     pr = torch.empty(B, device=dev)
     coef = torch.empty(B, device=dev)
+    qm = torch.zeros(1, device=dev)
     ext.dqn_loss_fwd(q_s, torch.randn(B, A, device=dev),
                      torch.randn(B, A, device=dev), act,
                      torch.randn(B, device=dev), torch.zeros(B, device=dev),
-                     torch.rand(B, device=dev), 0.97, 0.6, loss, pr, coef)
+                     torch.rand(B, device=dev), 0.97, 0.6, loss, pr, coef, qm)
     sync()
     print("  fwd ok loss=", loss.item(), flush=True)
     gq = torch.empty(B, A, device=dev)
