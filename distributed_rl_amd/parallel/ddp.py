@@ -79,12 +79,17 @@ def attach_reducer(learner, group=None) -> Optional[FlatGradReducer]:
         return None
     if getattr(learner, "mp", None) is not None:
         # mixed-precision path reduces its own flat bf16 grads; just make
-        # replicas start identical
+        # replicas start identical (including the target replica)
         learner.mp.broadcast_master()
         if hasattr(learner, "flat_tparam"):
             learner.flat_tparam.copy_(learner.mp.flat_cparam)
+        elif getattr(learner, "target", None) is not None:
+            learner.target.load_state_dict(learner.model.state_dict())
         return None
-    for model in (learner.model, learner.target):
+    models = [learner.model]
+    if getattr(learner, "target", None) is not None:
+        models.append(learner.target)
+    for model in models:
         for p in model.state_dict().values():
             if p.is_floating_point():
                 dist.broadcast(p, src=0, group=group)
