@@ -119,7 +119,7 @@ class TorchPER(ReplayBase):
                 torch.arange(k, device=self.device, dtype=torch.float32)
                 + torch.rand(k, device=self.device, generator=self.generator)
             ) / k * total
-            idx = torch.searchsorted(cdf, u.contiguous()).clamp(max=self.capacity - 1)
+            idx = torch.searchsorted(cdf, u.contiguous()).clamp(max=n - 1)
         else:
             idx = torch.multinomial(p, k, replacement=True, generator=self.generator)
         probs = p.index_select(0, idx) / total
