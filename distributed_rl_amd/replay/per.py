@@ -176,8 +176,6 @@ class FifoReplay(ReplayBase):
     def sample(self, k: int, beta: float = 0.0, with_data: bool = True):
         n = len(self)
         idx = torch.randint(0, n, (k,), device=self.device, generator=self.generator)
-        if self.count > self.capacity:
-            pass  # all slots valid once wrapped
         data = self.gather(idx) if with_data else None
         w = torch.ones(k, device=self.device)
         return data, idx, w
