@@ -141,10 +141,14 @@ class _FusedConvFn(torch.autograd.Function):
         if use_own_wrw:
             # hand-written MFMA weight-grad (conv_mfma.hip): fp32 workspace
             # accumulated by atomics, cast to the bf16 channels_last grad.
-            ws = torch.zeros(COUT, KH * KW * C, dtype=torch.float32,
-                             device=x.device)
-            gb_ws = (torch.zeros(COUT, dtype=torch.float32, device=x.device)
-                     if need_b else torch.empty(0, device=x.device))
+            # single zero-fill for both workspaces (one FillFunctor launch
+            # instead of two — these run every step inside the graph)
+            K = KH * KW * C
+            flat_ws = torch.zeros(COUT * (K + 1), dtype=torch.float32,
+                                  device=x.device)
+            ws = flat_ws[: COUT * K].view(COUT, K)
+            gb_ws = (flat_ws[COUT * K :] if need_b
+                     else torch.empty(0, device=x.device))
             ext.conv_wrw(x, gout, ws, gb_ws, stride)
             gw = ws.view(COUT, KH, KW, C).permute(0, 3, 1, 2).to(torch.bfloat16)
             if need_b:
@@ -221,15 +225,15 @@ class _DQNLossFn(torch.autograd.Function):
         ext = hip_ext()
         B, A = q_s.shape
         dev = q_s.device
-        loss = torch.zeros((), dtype=torch.float32, device=dev)
-        qmean = torch.zeros((), dtype=torch.float32, device=dev)
+        acc2 = torch.zeros(2, dtype=torch.float32, device=dev)  # one fill
+        loss, qmean = acc2[0], acc2[1]
         prio = torch.empty(B, dtype=torch.float32, device=dev)
         coef = torch.empty(B, dtype=torch.float32, device=dev)
         ext.dqn_loss_fwd(
             q_s.contiguous(), q_sp_on.contiguous(), q_sp_tg.contiguous(),
             actions.contiguous(), rewards.contiguous(), dones.contiguous(),
             weights.contiguous(), float(gamma_n), float(alpha),
-            loss.view(1), prio, coef, qmean.view(1),
+            acc2[0:1], prio, coef, acc2[1:2],
         )
         ctx.save_for_backward(coef, actions)
         ctx.shape = (B, A)
@@ -324,6 +328,34 @@ def vtrace(behavior_log_prob, target_log_prob, rewards, values, bootstrap_value,
     )
 
 
+def vtrace_bt(mu_probs, target_log_prob, rewards, values, bootstrap_value,
+              not_done, gamma, rho_bar=1.0, c_bar=1.0, lam=1.0):
+    """V-trace on (B, T) row-major tensors with raw behavior probabilities.
+
+    Same math as :func:`vtrace` but the learner's natural layout goes straight
+    in and comes straight out — on GPU this removes the 4 input transposes, the
+    mu.log() pass and the 2 output transposes per IMPALA step. Returns
+    (vs, pg_adv), both (B, T)."""
+    if _use_hip(rewards):
+        B, T = rewards.shape
+        dev = rewards.device
+        vs = torch.empty(B, T, dtype=torch.float32, device=dev)
+        pg_adv = torch.empty_like(vs)
+        hip_ext().vtrace_bt(
+            mu_probs.float().contiguous(), target_log_prob.float().contiguous(),
+            rewards.float().contiguous(), values.float().contiguous(),
+            bootstrap_value.float().contiguous(), not_done.float().contiguous(),
+            gamma, rho_bar, c_bar, lam, vs, pg_adv,
+        )
+        return vs, pg_adv
+    vs_T, pg_T, _ = torch_ref.vtrace(
+        mu_probs.log().t().contiguous(), target_log_prob.t().contiguous(),
+        rewards.t().contiguous(), values.t().contiguous(), bootstrap_value,
+        not_done, gamma, rho_bar, c_bar, lam
+    )
+    return vs_T.t(), pg_T.t()
+
+
 # ---------------------------------------------------------------------------
 # K9 — fused IMPALA policy objective
 # ---------------------------------------------------------------------------
@@ -335,9 +367,8 @@ def policy_softmax_stats(logits, actions):
     ext = hip_ext()
     N, A = logits.shape
     dev = logits.device
-    dummy = torch.zeros(N, device=dev)
-    obj = torch.zeros(1, device=dev)
-    ent = torch.zeros(1, device=dev)
+    zbuf = torch.zeros(N + 2, device=dev)  # one fill for all three
+    obj, ent, dummy = zbuf[0:1], zbuf[1:2], zbuf[2:]
     logpa = torch.empty(N, device=dev)
     pi = torch.empty(N, A, device=dev)
     H = torch.empty(N, device=dev)

@@ -359,6 +359,44 @@ __global__ void vtrace_kernel(const float* __restrict__ b_logp,
   }
 }
 
+// (B,T)-layout variant: inputs/outputs row-major (B, T) so the learner needs
+// no .t().contiguous() round-trips, and behavior probabilities mu enter raw
+// (log taken in-kernel) — replaces 7 transpose/log launches per IMPALA step.
+// Uncoalesced per-lane stride T is irrelevant here: the whole working set is
+// a few KB and lives in L2.
+__global__ void vtrace_bt_kernel(const float* __restrict__ mu,
+                                 const float* __restrict__ t_logp,
+                                 const float* __restrict__ rew,
+                                 const float* __restrict__ values,
+                                 const float* __restrict__ boot,
+                                 const float* __restrict__ not_done, int T,
+                                 int B, float gamma, float rho_bar,
+                                 float c_bar, float lam,
+                                 float* __restrict__ vs,
+                                 float* __restrict__ pg_adv) {
+  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  const int64_t base = (int64_t)b * T;
+  float bv = boot[b] * not_done[b];
+  float acc = 0.0f;
+  for (int t = T - 1; t >= 0; --t) {
+    int64_t o = base + t;
+    float vtp1 = (t == T - 1) ? bv : values[o + 1];
+    float rho = expf(t_logp[o]) / mu[o];
+    float rho_c = fminf(rho, rho_bar);
+    float c = lam * fminf(rho, c_bar);
+    float delta = rho_c * (rew[o] + gamma * vtp1 - values[o]);
+    acc = delta + gamma * c * acc;
+    vs[o] = values[o] + acc;
+  }
+  for (int t = 0; t < T; ++t) {
+    int64_t o = base + t;
+    float vstp1 = (t == T - 1) ? bv : vs[o + 1];
+    float rho_c = fminf(expf(t_logp[o]) / mu[o], rho_bar);
+    pg_adv[o] = rho_c * (rew[o] + gamma * vstp1 - values[o]);
+  }
+}
+
 // ---------------------------------------------------------------------------
 // K11: fused grad-norm + clip over a flat fp32 buffer (two launches, zero
 // host sync: the scale is computed on-device from the sq-sum scalar).
@@ -366,14 +404,28 @@ __global__ void vtrace_kernel(const float* __restrict__ b_logp,
 
 __global__ void sq_sum_kernel(const float* __restrict__ x, int64_t n,
                               float* __restrict__ out /*pre-zeroed*/) {
+  // float4 loads (16 B/lane) + one LDS block reduce -> ONE atomic per block
+  // (was: scalar loads + 4 wave atomics/block; measured 19.7 us on a 6.8 MB
+  // flat grad — ~350 GB/s, load-width bound).
   float acc = 0.0f;
+  const int64_t n4 = n >> 2;
+  const float4* __restrict__ x4 = (const float4*)x;
   int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n; i += stride) {
-    float v = x[i];
-    acc += v * v;
+  int64_t gid = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  for (int64_t i = gid; i < n4; i += stride) {
+    float4 v = x4[i];
+    acc += v.x * v.x + v.y * v.y + v.z * v.z + v.w * v.w;
   }
+  for (int64_t i = (n4 << 2) + gid; i < n; i += stride) acc += x[i] * x[i];
   for (int off = 32; off > 0; off >>= 1) acc += __shfl_down(acc, off, 64);
-  if ((threadIdx.x & 63) == 0) atomicAdd(out, acc);
+  __shared__ float wsum[kBlock / 64];
+  if ((threadIdx.x & 63) == 0) wsum[threadIdx.x >> 6] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float s = 0.0f;
+    for (int w = 0; w < kBlock / 64; ++w) s += wsum[w];
+    atomicAdd(out, s);
+  }
 }
 
 __global__ void clip_scale_kernel(float* __restrict__ x, int64_t n,
@@ -567,6 +619,20 @@ void vtrace(torch::Tensor b_logp, torch::Tensor t_logp, torch::Tensor rew,
                      rho_out.data_ptr<float>());
 }
 
+void vtrace_bt(torch::Tensor mu, torch::Tensor t_logp, torch::Tensor rew,
+               torch::Tensor values, torch::Tensor boot, torch::Tensor not_done,
+               double gamma, double rho_bar, double c_bar, double lam,
+               torch::Tensor vs, torch::Tensor pg_adv) {
+  int B = (int)rew.size(0), T = (int)rew.size(1);
+  hipLaunchKernelGGL(vtrace_bt_kernel, dim3(ceil_div(B, kBlock)), dim3(kBlock),
+                     0, cur_stream(), mu.data_ptr<float>(),
+                     t_logp.data_ptr<float>(), rew.data_ptr<float>(),
+                     values.data_ptr<float>(), boot.data_ptr<float>(),
+                     not_done.data_ptr<float>(), T, B, (float)gamma,
+                     (float)rho_bar, (float)c_bar, (float)lam,
+                     vs.data_ptr<float>(), pg_adv.data_ptr<float>());
+}
+
 void grad_clip(torch::Tensor flat, double max_norm, torch::Tensor sqsum_buf) {
   int64_t n = flat.numel();
   hipLaunchKernelGGL(sq_sum_kernel, dim3(grid_for(n, 8)), dim3(kBlock), 0,
@@ -624,6 +690,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("inv_value_rescale", &inv_value_rescale, "R2D2 h^-1(x) (K6)");
   m.def("seq_priority", &seq_priority, "R2D2 eta-mix sequence priority (K7)");
   m.def("vtrace", &vtrace, "IMPALA V-trace reversed scan (K8)");
+  m.def("vtrace_bt", &vtrace_bt,
+        "V-trace on (B,T) row-major tensors, mu probs raw (K8, no transposes)");
   m.def("grad_clip", &grad_clip, "fused global grad-norm clip (K11)");
   m.def("relu_mask_bwd", &relu_mask_bwd, "dst = gout * (out > 0), bf16");
   m.def("dueling_fwd", &dueling_fwd, "fused (A+V)-mean(A) (K3)");

@@ -135,6 +135,29 @@ def test_vtrace_matches_ref(ext):
     assert torch.allclose(rho.cpu(), rho_r, atol=1e-5)
 
 
+def test_vtrace_bt_matches_ref(ext):
+    """(B,T)-layout V-trace (raw mu probs, in-kernel log) vs the T-major
+    oracle composition."""
+    from distributed_rl_amd import ops
+
+    torch.manual_seed(2)
+    B, T = 48, 20
+    mu = torch.rand(B, T, device=DEV).clamp_min(1e-3)
+    tlogp = -torch.rand(B, T, device=DEV)
+    rew = torch.randn(B, T, device=DEV)
+    val = torch.randn(B, T, device=DEV)
+    boot = torch.randn(B, device=DEV)
+    nd = (torch.rand(B, device=DEV) < 0.8).float()
+    vs, pg = ops.vtrace_bt(mu, tlogp, rew, val, boot, nd, 0.99,
+                           rho_bar=1.0, c_bar=1.0, lam=0.95)
+    vs_c, pg_c = ops.vtrace_bt(mu.cpu(), tlogp.cpu(), rew.cpu(), val.cpu(),
+                               boot.cpu(), nd.cpu(), 0.99,
+                               rho_bar=1.0, c_bar=1.0, lam=0.95)
+    assert vs.shape == (B, T) and pg.shape == (B, T)
+    assert torch.allclose(vs.cpu(), vs_c, atol=1e-4)
+    assert torch.allclose(pg.cpu(), pg_c, atol=1e-4)
+
+
 def test_value_rescale_matches_ref(ext):
     from distributed_rl_amd import ops
     from distributed_rl_amd.ops import torch_ref
