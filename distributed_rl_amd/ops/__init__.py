@@ -393,8 +393,57 @@ class _PolicyObjFn(torch.autograd.Function):
         pi, H, actions, adv = ctx.saved_tensors
         dlogits = torch.empty_like(pi)
         ext.policy_loss_bwd(pi, H, actions, adv,
-                            gobj.reshape(1).contiguous(), ctx.er, dlogits)
+                            gobj.reshape(1).contiguous(), ctx.er, 1.0, dlogits)
         return dlogits, None, None, None, None, None, None
+
+
+class _ImpalaLossFn(torch.autograd.Function):
+    """Whole IMPALA loss in one forward kernel:
+    loss = -(mean(logpa*adv) + er*mean_H) + 0.5*mean((v - vs)^2),
+    differentiable in (logits, v) with closed-form backward kernels
+    (IMPALA/Learner.py:95-119 math, one launch instead of ~10)."""
+
+    @staticmethod
+    def forward(ctx, logits, v_t, logpa, pi, H, actions, adv, vs, mean_H, er):
+        ext = hip_ext()
+        dev = logits.device
+        v_c = v_t.contiguous().view(-1)
+        vs_c = vs.contiguous().view(-1)
+        loss = torch.empty(1, device=dev)
+        obj = torch.empty(1, device=dev)
+        critic = torch.empty(1, device=dev)
+        ext.impala_loss_fwd(logpa, adv, mean_H.reshape(1).contiguous(),
+                            v_c, vs_c, er, loss, obj, critic)
+        ctx.save_for_backward(pi, H, actions, adv, v_c, vs_c)
+        ctx.er = er
+        ctx.vshape = v_t.shape
+        ctx.set_materialize_grads(False)
+        ctx.mark_non_differentiable(obj, critic)
+        return loss.squeeze(0), obj.squeeze(0), critic.squeeze(0)
+
+    @staticmethod
+    def backward(ctx, gloss, _gobj, _gcritic):
+        ext = hip_ext()
+        pi, H, actions, adv, v_c, vs_c = ctx.saved_tensors
+        if gloss is None:
+            return (None,) * 10
+        g = gloss.reshape(1).contiguous()
+        dlogits = torch.empty_like(pi)
+        ext.policy_loss_bwd(pi, H, actions, adv, g, ctx.er, -1.0, dlogits)
+        dv = torch.empty_like(v_c)
+        ext.impala_critic_bwd(v_c, vs_c, g, dv)
+        return (dlogits, dv.view(ctx.vshape), None, None, None, None, None,
+                None, None, None)
+
+
+def impala_fused_loss(logits, v_t, stats, actions, adv, vs, entropy_coef):
+    """GPU-only fused total loss; returns (loss, obj, critic) scalars with
+    loss differentiable in (logits, v_t). ``stats`` is the
+    policy_softmax_stats tuple computed for V-trace."""
+    logpa, pi, H, ent = stats
+    return _ImpalaLossFn.apply(logits, v_t, logpa, pi, H, actions.long(),
+                               adv.float().contiguous(), vs, ent,
+                               entropy_coef)
 
 
 def impala_policy_objective(logits, actions, adv, entropy_coef,

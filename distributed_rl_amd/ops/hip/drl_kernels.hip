@@ -660,7 +660,13 @@ void policy_loss_fwd(torch::Tensor logits, torch::Tensor act, torch::Tensor adv,
                      torch::Tensor H_save);
 void policy_loss_bwd(torch::Tensor pi_save, torch::Tensor H_save,
                      torch::Tensor act, torch::Tensor adv, torch::Tensor gout,
-                     double er, torch::Tensor dlogits);
+                     double er, double gsign, torch::Tensor dlogits);
+void impala_loss_fwd(torch::Tensor logpa, torch::Tensor adv,
+                     torch::Tensor mean_H, torch::Tensor v, torch::Tensor vs,
+                     double er, torch::Tensor loss_out, torch::Tensor obj_out,
+                     torch::Tensor critic_out);
+void impala_critic_bwd(torch::Tensor v, torch::Tensor vs, torch::Tensor gloss,
+                       torch::Tensor dv);
 void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor sq,
                   torch::Tensor ga, torch::Tensor mom, double lr, double alpha,
                   double eps, double wd, double mu, bool centered, bool has_mom);
@@ -700,6 +706,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_cell_bwd", &lstm_cell_bwd, "fused LSTM cell backward (K5)");
   m.def("policy_loss_fwd", &policy_loss_fwd, "fused IMPALA policy obj fwd (K9)");
   m.def("policy_loss_bwd", &policy_loss_bwd, "fused IMPALA policy obj bwd (K9)");
+  m.def("impala_loss_fwd", &impala_loss_fwd,
+        "fused IMPALA total loss: pg objective + entropy + critic MSE (K9)");
+  m.def("impala_critic_bwd", &impala_critic_bwd,
+        "IMPALA critic MSE backward: dv = g*(v-vs)/n (K9)");
   m.def("rmsprop_step", &rmsprop_step, "fused flat centered RMSprop (K12)");
   m.def("adam_step", &adam_step, "fused flat Adam (K12)");
   m.def("lstm_step_fused", &lstm_step_fused,
@@ -935,8 +945,8 @@ __global__ void policy_loss_fwd_kernel(
 __global__ void policy_loss_bwd_kernel(
     const float* __restrict__ pi_save, const float* __restrict__ H_save,
     const int64_t* __restrict__ act, const float* __restrict__ adv,
-    const float* __restrict__ gout,  // d(total)/d(obj): scalar (usually -1)
-    int N, int A, float er, float* __restrict__ dlogits) {
+    const float* __restrict__ gout,  // upstream grad scalar
+    int N, int A, float er, float gsign, float* __restrict__ dlogits) {
   int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
   if (k >= (int64_t)N * A) return;
   int i = (int)(k / A);
@@ -944,10 +954,63 @@ __global__ void policy_loss_bwd_kernel(
   float p = pi_save[k];
   float lp = __logf(fmaxf(p, 1e-30f));
   float invN = 1.0f / N;
-  // d obj / d logit_ij = [ adv_i*(1{j=a} - p) - er * p*(lp + H_i) ] / N
+  // d obj / d logit_ij = [ adv_i*(1{j=a} - p) - er * p*(lp + H_i) ] / N;
+  // gsign = +1 when the upstream node is obj itself, -1 when it is
+  // loss = -obj + critic (the fused-loss path).
   float d = adv[i] * (((int)act[i] == j ? 1.0f : 0.0f) - p)
             - er * p * (lp + H_save[i]);
-  dlogits[k] = gout[0] * d * invN;
+  dlogits[k] = gsign * gout[0] * d * invN;
+}
+
+// Fused IMPALA total loss: loss = -(mean(logpa*adv) + er*mean_H) +
+// 0.5*mean((v - vs)^2). mean_H arrives pre-reduced (policy_softmax_stats
+// already accumulated it). ONE block — n = B*T is a few thousand — with
+// plain stores at the end (no pre-zero fill, no atomics). Replaces the
+// ~10-launch torch composition (mul, 3 means, mse, neg, add, broadcasts).
+__global__ void impala_loss_fwd_kernel(
+    const float* __restrict__ logpa, const float* __restrict__ adv,
+    const float* __restrict__ mean_H, const float* __restrict__ v,
+    const float* __restrict__ vs, int64_t n, float er,
+    float* __restrict__ loss_out, float* __restrict__ obj_out,
+    float* __restrict__ critic_out) {
+  float s1 = 0.0f, s3 = 0.0f;
+  for (int64_t i = threadIdx.x; i < n; i += blockDim.x) {
+    s1 += logpa[i] * adv[i];
+    float d = v[i] - vs[i];
+    s3 += d * d;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    s1 += __shfl_down(s1, off, 64);
+    s3 += __shfl_down(s3, off, 64);
+  }
+  __shared__ float w1[kBlock / 64], w3[kBlock / 64];
+  if ((threadIdx.x & 63) == 0) {
+    w1[threadIdx.x >> 6] = s1;
+    w3[threadIdx.x >> 6] = s3;
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float a = 0.0f, c = 0.0f;
+    for (int w = 0; w < kBlock / 64; ++w) { a += w1[w]; c += w3[w]; }
+    float invn = 1.0f / (float)n;
+    float obj = a * invn + er * mean_H[0];
+    float critic = 0.5f * c * invn;
+    loss_out[0] = -obj + critic;
+    obj_out[0] = obj;
+    critic_out[0] = critic;
+  }
+}
+
+// d loss / d v = gloss * (v - vs) / n
+__global__ void impala_critic_bwd_kernel(const float* __restrict__ v,
+                                         const float* __restrict__ vs,
+                                         const float* __restrict__ gloss,
+                                         int64_t n, float* __restrict__ dv) {
+  float g = gloss[0] / (float)n;
+  int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    dv[i] = g * (v[i] - vs[i]);
 }
 }  // namespace
 
@@ -966,14 +1029,36 @@ void policy_loss_fwd(torch::Tensor logits, torch::Tensor act, torch::Tensor adv,
 
 void policy_loss_bwd(torch::Tensor pi_save, torch::Tensor H_save,
                      torch::Tensor act, torch::Tensor adv, torch::Tensor gout,
-                     double er, torch::Tensor dlogits) {
+                     double er, double gsign, torch::Tensor dlogits) {
   int N = (int)pi_save.size(0), A = (int)pi_save.size(1);
   hipLaunchKernelGGL(policy_loss_bwd_kernel,
                      dim3(ceil_div((int64_t)N * A, kBlock)), dim3(kBlock), 0,
                      cur_stream(), pi_save.data_ptr<float>(),
                      H_save.data_ptr<float>(), act.data_ptr<int64_t>(),
                      adv.data_ptr<float>(), gout.data_ptr<float>(), N, A,
-                     (float)er, dlogits.data_ptr<float>());
+                     (float)er, (float)gsign, dlogits.data_ptr<float>());
+}
+
+void impala_loss_fwd(torch::Tensor logpa, torch::Tensor adv,
+                     torch::Tensor mean_H, torch::Tensor v, torch::Tensor vs,
+                     double er, torch::Tensor loss_out, torch::Tensor obj_out,
+                     torch::Tensor critic_out) {
+  int64_t n = logpa.numel();
+  hipLaunchKernelGGL(impala_loss_fwd_kernel, dim3(1), dim3(kBlock), 0,
+                     cur_stream(), logpa.data_ptr<float>(),
+                     adv.data_ptr<float>(), mean_H.data_ptr<float>(),
+                     v.data_ptr<float>(), vs.data_ptr<float>(), n, (float)er,
+                     loss_out.data_ptr<float>(), obj_out.data_ptr<float>(),
+                     critic_out.data_ptr<float>());
+}
+
+void impala_critic_bwd(torch::Tensor v, torch::Tensor vs, torch::Tensor gloss,
+                       torch::Tensor dv) {
+  int64_t n = v.numel();
+  hipLaunchKernelGGL(impala_critic_bwd_kernel, dim3(grid_for(n, 4)),
+                     dim3(kBlock), 0, cur_stream(), v.data_ptr<float>(),
+                     vs.data_ptr<float>(), gloss.data_ptr<float>(), n,
+                     dv.data_ptr<float>());
 }
 
 // K12: fused optimizers over the flat master buffers. torch's capturable

@@ -158,6 +158,44 @@ def test_vtrace_bt_matches_ref(ext):
     assert torch.allclose(pg.cpu(), pg_c, atol=1e-4)
 
 
+def test_impala_fused_loss_matches_torch(ext):
+    """One-kernel total loss (pg obj + entropy + critic MSE) vs the plain
+    torch composition, values and both closed-form gradients."""
+    import torch.nn.functional as F
+    from distributed_rl_amd import ops
+
+    torch.manual_seed(3)
+    B, T, A = 16, 16, 6
+    N = B * T
+    logits = torch.randn(N, A, device=DEV, requires_grad=True)
+    v = torch.randn(B, T, device=DEV, requires_grad=True)
+    actions = torch.randint(0, A, (N,), device=DEV)
+    adv = torch.randn(N, device=DEV)
+    vs = torch.randn(B, T, device=DEV)
+    er = 0.01
+    stats = ops.policy_softmax_stats(logits.detach(), actions)
+    loss, obj, critic = ops.impala_fused_loss(logits, v, stats, actions, adv,
+                                              vs, er)
+    loss.backward()
+
+    logits2 = logits.detach().clone().requires_grad_(True)
+    v2 = v.detach().clone().requires_grad_(True)
+    log_pi = torch.log_softmax(logits2, -1)
+    pi = log_pi.exp()
+    entropy = -(pi * log_pi).sum(-1).mean()
+    logpa = log_pi.gather(1, actions.long().unsqueeze(1)).squeeze(1)
+    obj_r = (logpa * adv).mean() + er * entropy
+    critic_r = 0.5 * F.mse_loss(v2, vs)
+    loss_r = -obj_r + critic_r
+    loss_r.backward()
+
+    assert abs(loss.item() - loss_r.item()) < 1e-4
+    assert abs(obj.item() - obj_r.item()) < 1e-4
+    assert abs(critic.item() - critic_r.item()) < 1e-4
+    assert torch.allclose(logits.grad, logits2.grad, atol=1e-5)
+    assert torch.allclose(v.grad, v2.grad, atol=1e-6)
+
+
 def test_value_rescale_matches_ref(ext):
     from distributed_rl_amd import ops
     from distributed_rl_amd.ops import torch_ref
