@@ -130,32 +130,37 @@ class ImpalaLearner(LearnerBase):
         v_t = values[:, :T]
         bootstrap = values[:, T]
         if cuda:
-            # K9 fused path: one softmax-stats kernel feeds V-trace, then the
-            # objective with a closed-form backward kernel
-            logits_flat = logits[:, :T].reshape(B * T, A).contiguous()
+            # K9 fused path: one softmax-stats kernel feeds V-trace; the
+            # whole loss keeps only `out` in the autograd graph (the fused
+            # backward writes d out in a single launch), so every slice /
+            # reshape below is detached
+            logits_flat = (logits[:, :T].detach().reshape(B * T, A)
+                           .contiguous())
             actions_flat = actions.reshape(-1)
             stats9 = ops.policy_softmax_stats(logits_flat, actions_flat)
             log_pi_a_det = stats9[0].view(B, T)
+            v_t_det = v_t.detach().contiguous()
         else:
             log_pi = F.log_softmax(logits[:, :T], dim=-1)  # (B, T, A)
             log_pi_a = log_pi.gather(2, actions.unsqueeze(2)).squeeze(2)
             log_pi_a_det = log_pi_a.detach()
+            v_t_det = v_t.detach()
 
         with torch.no_grad():
             # (B,T)-native V-trace: no transpose/log round-trips on GPU
             vs, pg_adv = ops.vtrace_bt(
-                mu, log_pi_a_det, rewards, v_t.detach(), bootstrap.detach(),
+                mu, log_pi_a_det, rewards, v_t_det, bootstrap.detach(),
                 not_done, self.gamma, rho_bar=self.cfg.p_value,
                 c_bar=self.cfg.c_value, lam=self.cfg.c_lambda,
             )
 
         if cuda:
             # one-kernel total loss (pg objective + entropy bonus + critic
-            # MSE) with closed-form backward — replaces the ~10-launch torch
-            # reduction chain
+            # MSE) + one-kernel whole-head backward — replaces the
+            # ~10-launch torch reduction chain and both SliceBackwards
             loss, obj_actor, critic_loss = ops.impala_fused_loss(
-                logits_flat, v_t, stats9, actions_flat, pg_adv.reshape(-1),
-                vs, self.cfg.entropy_r,
+                out, v_t_det, stats9, actions_flat, pg_adv.reshape(-1),
+                vs, self.cfg.entropy_r, T,
             )
             entropy = stats9[3]
         else:

@@ -398,15 +398,17 @@ class _PolicyObjFn(torch.autograd.Function):
 
 
 class _ImpalaLossFn(torch.autograd.Function):
-    """Whole IMPALA loss in one forward kernel:
-    loss = -(mean(logpa*adv) + er*mean_H) + 0.5*mean((v - vs)^2),
-    differentiable in (logits, v) with closed-form backward kernels
-    (IMPALA/Learner.py:95-119 math, one launch instead of ~10)."""
+    """Whole IMPALA loss on the raw head output out = (B*(T+1), A+1):
+    loss = -(mean(logpa*adv) + er*mean_H) + 0.5*mean((v_t - vs)^2),
+    one forward kernel and ONE backward kernel that writes the full d out
+    (policy grad, critic grad and the zero bootstrap/T rows together) —
+    no SliceBackward zeros+copy+accumulate chains
+    (IMPALA/Learner.py:95-119 math)."""
 
     @staticmethod
-    def forward(ctx, logits, v_t, logpa, pi, H, actions, adv, vs, mean_H, er):
+    def forward(ctx, out, v_t, logpa, pi, H, actions, adv, vs, mean_H, er, T):
         ext = hip_ext()
-        dev = logits.device
+        dev = out.device
         v_c = v_t.contiguous().view(-1)
         vs_c = vs.contiguous().view(-1)
         loss = torch.empty(1, device=dev)
@@ -416,7 +418,8 @@ class _ImpalaLossFn(torch.autograd.Function):
                             v_c, vs_c, er, loss, obj, critic)
         ctx.save_for_backward(pi, H, actions, adv, v_c, vs_c)
         ctx.er = er
-        ctx.vshape = v_t.shape
+        ctx.T = T
+        ctx.out_shape = out.shape
         ctx.set_materialize_grads(False)
         ctx.mark_non_differentiable(obj, critic)
         return loss.squeeze(0), obj.squeeze(0), critic.squeeze(0)
@@ -426,24 +429,28 @@ class _ImpalaLossFn(torch.autograd.Function):
         ext = hip_ext()
         pi, H, actions, adv, v_c, vs_c = ctx.saved_tensors
         if gloss is None:
-            return (None,) * 10
+            return (None,) * 11
+        T = ctx.T
+        A = ctx.out_shape[1] - 1
+        B = ctx.out_shape[0] // (T + 1)
         g = gloss.reshape(1).contiguous()
-        dlogits = torch.empty_like(pi)
-        ext.policy_loss_bwd(pi, H, actions, adv, g, ctx.er, -1.0, dlogits)
-        dv = torch.empty_like(v_c)
-        ext.impala_critic_bwd(v_c, vs_c, g, dv)
-        return (dlogits, dv.view(ctx.vshape), None, None, None, None, None,
-                None, None, None)
+        dout = torch.empty(ctx.out_shape, dtype=torch.float32,
+                           device=v_c.device)
+        ext.impala_out_bwd(pi, H, actions, adv, v_c, vs_c, g, B, T, A,
+                           ctx.er, dout)
+        return (dout, None, None, None, None, None, None, None, None, None,
+                None)
 
 
-def impala_fused_loss(logits, v_t, stats, actions, adv, vs, entropy_coef):
-    """GPU-only fused total loss; returns (loss, obj, critic) scalars with
-    loss differentiable in (logits, v_t). ``stats`` is the
-    policy_softmax_stats tuple computed for V-trace."""
+def impala_fused_loss(out, v_t, stats, actions, adv, vs, entropy_coef, T):
+    """GPU-only fused total loss on the raw (B*(T+1), A+1) head output;
+    returns (loss, obj, critic) scalars with loss differentiable in ``out``.
+    ``v_t`` is the detached (B, T) value slice (also fed to V-trace);
+    ``stats`` is the policy_softmax_stats tuple computed for V-trace."""
     logpa, pi, H, ent = stats
-    return _ImpalaLossFn.apply(logits, v_t, logpa, pi, H, actions.long(),
+    return _ImpalaLossFn.apply(out, v_t, logpa, pi, H, actions.long(),
                                adv.float().contiguous(), vs, ent,
-                               entropy_coef)
+                               entropy_coef, T)
 
 
 def impala_policy_objective(logits, actions, adv, entropy_coef,

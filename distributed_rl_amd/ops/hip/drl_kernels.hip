@@ -667,6 +667,10 @@ void impala_loss_fwd(torch::Tensor logpa, torch::Tensor adv,
                      torch::Tensor critic_out);
 void impala_critic_bwd(torch::Tensor v, torch::Tensor vs, torch::Tensor gloss,
                        torch::Tensor dv);
+void impala_out_bwd(torch::Tensor pi_save, torch::Tensor H_save,
+                    torch::Tensor act, torch::Tensor adv, torch::Tensor v,
+                    torch::Tensor vs, torch::Tensor gloss, int64_t B,
+                    int64_t T, int64_t A, double er, torch::Tensor dout);
 void rmsprop_step(torch::Tensor p, torch::Tensor g, torch::Tensor sq,
                   torch::Tensor ga, torch::Tensor mom, double lr, double alpha,
                   double eps, double wd, double mu, bool centered, bool has_mom);
@@ -710,6 +714,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "fused IMPALA total loss: pg objective + entropy + critic MSE (K9)");
   m.def("impala_critic_bwd", &impala_critic_bwd,
         "IMPALA critic MSE backward: dv = g*(v-vs)/n (K9)");
+  m.def("impala_out_bwd", &impala_out_bwd,
+        "IMPALA whole-head backward: d loss/d out in one launch (K9)");
   m.def("rmsprop_step", &rmsprop_step, "fused flat centered RMSprop (K12)");
   m.def("adam_step", &adam_step, "fused flat Adam (K12)");
   m.def("lstm_step_fused", &lstm_step_fused,
@@ -1012,6 +1018,46 @@ __global__ void impala_critic_bwd_kernel(const float* __restrict__ v,
        i += stride)
     dv[i] = g * (v[i] - vs[i]);
 }
+
+// Whole-output IMPALA backward: writes d loss / d out for the raw network
+// head out = (B*(T+1), A+1) in ONE launch — policy grad into columns 0..A-1
+// for t<T, critic grad into column A for t<T, zeros elsewhere (t==T rows
+// feed only the detached bootstrap). Replaces two SliceBackward
+// zeros+copy+accumulate chains in autograd.
+__global__ void impala_out_bwd_kernel(
+    const float* __restrict__ pi_save,  // (B*T, A)
+    const float* __restrict__ H_save,   // (B*T,)
+    const int64_t* __restrict__ act,    // (B*T,)
+    const float* __restrict__ adv,      // (B*T,)
+    const float* __restrict__ v,        // (B*T,)
+    const float* __restrict__ vs,       // (B*T,)
+    const float* __restrict__ gloss, int B, int T, int A, float er,
+    float* __restrict__ dout) {
+  const int64_t total = (int64_t)B * (T + 1) * (A + 1);
+  int64_t k = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  const int64_t stride = (int64_t)gridDim.x * blockDim.x;
+  const float n_inv = 1.0f / (float)((int64_t)B * T);
+  for (; k < total; k += stride) {
+    int64_t r = k / (A + 1);
+    int j = (int)(k - r * (A + 1));
+    int b = (int)(r / (T + 1));
+    int t = (int)(r - (int64_t)b * (T + 1));
+    float g = 0.0f;
+    if (t < T) {
+      int64_t i = (int64_t)b * T + t;
+      if (j < A) {
+        float p = pi_save[i * A + j];
+        float lp = __logf(fmaxf(p, 1e-30f));
+        float d = adv[i] * (((int)act[i] == j ? 1.0f : 0.0f) - p)
+                  - er * p * (lp + H_save[i]);
+        g = -gloss[0] * d * n_inv;  // loss = -obj + critic
+      } else {
+        g = gloss[0] * (v[i] - vs[i]) * n_inv;
+      }
+    }
+    dout[k] = g;
+  }
+}
 }  // namespace
 
 void policy_loss_fwd(torch::Tensor logits, torch::Tensor act, torch::Tensor adv,
@@ -1059,6 +1105,19 @@ void impala_critic_bwd(torch::Tensor v, torch::Tensor vs, torch::Tensor gloss,
                      dim3(kBlock), 0, cur_stream(), v.data_ptr<float>(),
                      vs.data_ptr<float>(), gloss.data_ptr<float>(), n,
                      dv.data_ptr<float>());
+}
+
+void impala_out_bwd(torch::Tensor pi_save, torch::Tensor H_save,
+                    torch::Tensor act, torch::Tensor adv, torch::Tensor v,
+                    torch::Tensor vs, torch::Tensor gloss, int64_t B,
+                    int64_t T, int64_t A, double er, torch::Tensor dout) {
+  int64_t total = B * (T + 1) * (A + 1);
+  hipLaunchKernelGGL(impala_out_bwd_kernel, dim3(grid_for(total, 2)),
+                     dim3(kBlock), 0, cur_stream(), pi_save.data_ptr<float>(),
+                     H_save.data_ptr<float>(), act.data_ptr<int64_t>(),
+                     adv.data_ptr<float>(), v.data_ptr<float>(),
+                     vs.data_ptr<float>(), gloss.data_ptr<float>(), (int)B,
+                     (int)T, (int)A, (float)er, dout.data_ptr<float>());
 }
 
 // K12: fused optimizers over the flat master buffers. torch's capturable

@@ -159,27 +159,31 @@ def test_vtrace_bt_matches_ref(ext):
 
 
 def test_impala_fused_loss_matches_torch(ext):
-    """One-kernel total loss (pg obj + entropy + critic MSE) vs the plain
-    torch composition, values and both closed-form gradients."""
+    """One-kernel total loss (pg obj + entropy + critic MSE) + one-kernel
+    whole-head backward vs the plain torch composition on the raw
+    (B*(T+1), A+1) head output."""
     import torch.nn.functional as F
     from distributed_rl_amd import ops
 
     torch.manual_seed(3)
     B, T, A = 16, 16, 6
     N = B * T
-    logits = torch.randn(N, A, device=DEV, requires_grad=True)
-    v = torch.randn(B, T, device=DEV, requires_grad=True)
+    out = torch.randn(B * (T + 1), A + 1, device=DEV, requires_grad=True)
     actions = torch.randint(0, A, (N,), device=DEV)
     adv = torch.randn(N, device=DEV)
     vs = torch.randn(B, T, device=DEV)
     er = 0.01
-    stats = ops.policy_softmax_stats(logits.detach(), actions)
-    loss, obj, critic = ops.impala_fused_loss(logits, v, stats, actions, adv,
-                                              vs, er)
+    logits3 = out.view(B, T + 1, A + 1)[:, :T, :A]
+    logits_flat = logits3.detach().reshape(N, A).contiguous()
+    v_t = out.view(B, T + 1, A + 1)[:, :T, A].detach().contiguous()
+    stats = ops.policy_softmax_stats(logits_flat, actions)
+    loss, obj, critic = ops.impala_fused_loss(out, v_t, stats, actions, adv,
+                                              vs, er, T)
     loss.backward()
 
-    logits2 = logits.detach().clone().requires_grad_(True)
-    v2 = v.detach().clone().requires_grad_(True)
+    out2 = out.detach().clone().requires_grad_(True)
+    logits2 = out2.view(B, T + 1, A + 1)[:, :T, :A].reshape(N, A)
+    v2 = out2.view(B, T + 1, A + 1)[:, :T, A]
     log_pi = torch.log_softmax(logits2, -1)
     pi = log_pi.exp()
     entropy = -(pi * log_pi).sum(-1).mean()
@@ -192,8 +196,7 @@ def test_impala_fused_loss_matches_torch(ext):
     assert abs(loss.item() - loss_r.item()) < 1e-4
     assert abs(obj.item() - obj_r.item()) < 1e-4
     assert abs(critic.item() - critic_r.item()) < 1e-4
-    assert torch.allclose(logits.grad, logits2.grad, atol=1e-5)
-    assert torch.allclose(v.grad, v2.grad, atol=1e-6)
+    assert torch.allclose(out.grad, out2.grad, atol=1e-5)
 
 
 def test_value_rescale_matches_ref(ext):
