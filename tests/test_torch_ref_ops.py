@@ -111,3 +111,25 @@ def test_impala_loss_shapes_and_entropy():
     assert 0 < ent.item() <= math.log(A) + 1e-5
     total.backward()
     assert logits.grad is not None and values.grad is not None
+
+
+def test_vtrace_bt_cpu_matches_tmajor_composition():
+    """The (B,T)-layout wrapper (CPU fallback path) equals the T-major
+    oracle with explicit transposes and mu.log()."""
+    from distributed_rl_amd import ops
+
+    torch.manual_seed(4)
+    B, T = 6, 12
+    mu = torch.rand(B, T).clamp_min(1e-3)
+    tlogp = -torch.rand(B, T)
+    rew = torch.randn(B, T)
+    val = torch.randn(B, T)
+    boot = torch.randn(B)
+    nd = (torch.rand(B) < 0.7).float()
+    vs, pg = ops.vtrace_bt(mu, tlogp, rew, val, boot, nd, 0.97,
+                           rho_bar=1.0, c_bar=1.0, lam=0.9)
+    vs_r, pg_r, _ = R.vtrace(mu.log().t().contiguous(), tlogp.t().contiguous(),
+                             rew.t().contiguous(), val.t().contiguous(),
+                             boot, nd, 0.97, rho_bar=1.0, c_bar=1.0, lam=0.9)
+    assert torch.allclose(vs, vs_r.t(), atol=1e-5)
+    assert torch.allclose(pg, pg_r.t(), atol=1e-5)
