@@ -107,7 +107,9 @@ class ImpalaLearner(LearnerBase):
         self.replay.push(cols)
 
     # -- train -----------------------------------------------------------
-    def train_step(self, data) -> Dict[str, torch.Tensor]:
+    def _fwd_bwd(self, data) -> Dict[str, torch.Tensor]:
+        """Forward + V-trace + fused loss + backward (grads left in the
+        flat buffers). Collective-free, so it is hipGraph-capturable."""
         cuda = self.device.type == "cuda"
         T = self.unroll
         B = data["not_done"].shape[0]
@@ -173,17 +175,9 @@ class ImpalaLearner(LearnerBase):
         if self.mp is not None:
             self.mp.zero_grads()
             loss.backward()
-            self.mp.reduce_and_upcast()
-            ops.clip_flat_grad_(self.mp.flat_mgrad, 40.0, self._sqsum_buf)
-            self.optim.step()
-            self.mp.sync_compute_params()
         else:
             self.optim.zero_grad(set_to_none=False)
             loss.backward()
-            if self.reducer is not None:
-                self.reducer.all_reduce()
-            self.model.clippingNorm(40.0)
-            self.optim.step()
         return {
             "loss": loss.detach(),
             "obj_actor": obj_actor.detach(),
@@ -196,6 +190,26 @@ class ImpalaLearner(LearnerBase):
             "vs": vs,
             "pg_adv": pg_adv,
         }
+
+    def _optimize_mp(self):
+        """Upcast + clip + optimizer + param sync (post-collective stage;
+        hipGraph-capturable)."""
+        self.mp.upcast_grads()
+        ops.clip_flat_grad_(self.mp.flat_mgrad, 40.0, self._sqsum_buf)
+        self.optim.step()
+        self.mp.sync_compute_params()
+
+    def train_step(self, data) -> Dict[str, torch.Tensor]:
+        stats = self._fwd_bwd(data)
+        if self.mp is not None:
+            self.mp.allreduce_grads()
+            self._optimize_mp()
+        else:
+            if self.reducer is not None:
+                self.reducer.all_reduce()
+            self.model.clippingNorm(40.0)
+            self.optim.step()
+        return stats
 
     def _inner_step(self):
         data, _, _ = self.replay.sample(self.batch_size)
@@ -212,22 +226,47 @@ class ImpalaLearner(LearnerBase):
         return stats
 
     def make_graphed_step(self, warmup_iters: int = 3):
+        """hipGraph-capture the learner step. At world_size > 1 the step is
+        captured as TWO graphs with the RCCL all-reduce running eagerly
+        between them — collectives are never captured (same structure as
+        ApexLearner.make_graphed_step)."""
         assert self.device.type == "cuda"
         for g in self.optim.param_groups:
             g["capturable"] = True
+        split = self.mp is not None and self.mp.world > 1
         side = torch.cuda.Stream(self.device)
         side.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(side):
             for _ in range(warmup_iters):
                 self._inner_step()
         torch.cuda.current_stream(self.device).wait_stream(side)
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            static_out = self._inner_step()
-        self._graph = graph
+
+        if not split:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_out = self._inner_step()
+            self._graph = graph
+
+            def stepper():
+                graph.replay()
+                self._cadence()
+                return static_out
+
+            return stepper
+
+        g1 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g1):
+            data, _, _ = self.replay.sample(self.batch_size)
+            static_out = self._fwd_bwd(data)
+        g2 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g2, pool=g1.pool()):
+            self._optimize_mp()
+        self._graph = (g1, g2)
 
         def stepper():
-            graph.replay()
+            g1.replay()
+            self.mp.allreduce_grads()
+            g2.replay()
             self._cadence()
             return static_out
 

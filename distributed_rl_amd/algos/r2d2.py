@@ -212,7 +212,9 @@ class R2D2Learner(LearnerBase):
         return (h.to(self.device), c.to(self.device))
 
     # -- train -------------------------------------------------------------
-    def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
+    def _fwd_bwd(self, data, weights):
+        """Sequence passes + n-step targets + loss + backward + priority
+        (collective-free: hipGraph-capturable)."""
         B = data["done"].shape[0]
         T = self.T
         states = data["states"].to(self.device, non_blocking=True)  # (B,T,4,84,84)
@@ -250,23 +252,35 @@ class R2D2Learner(LearnerBase):
         if self.mp is not None:
             self.mp.zero_grads()
             loss.backward()
-            self.mp.reduce_and_upcast()
-            self.model.clippingNorm(40.0)  # joint norm over both dtype groups
-            self.optim.step()
-            self.mp.sync_compute_params()
         else:
             self.optim.zero_grad(set_to_none=False)
             loss.backward()
+        stats = {
+            "loss": loss.detach(),
+            "value": q_taken.detach().mean(),
+            "td_abs": td.detach().abs().mean(),
+        }
+        return stats, prio
+
+    def _optimize_mp(self):
+        """Upcast + clip + optimizer + param sync (post-collective stage)."""
+        self.mp.upcast_grads()
+        self.model.clippingNorm(40.0)  # joint norm over both dtype groups
+        self.optim.step()
+        self.mp.sync_compute_params()
+
+    def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
+        stats, prio = self._fwd_bwd(data, weights)
+        if self.mp is not None:
+            self.mp.allreduce_grads()
+            self._optimize_mp()
+        else:
             if self.reducer is not None:
                 self.reducer.all_reduce()
             self.model.clippingNorm(40.0)  # R2D2/Learner.py:208
             self.optim.step()
         self.replay.update(idx, prio)
-        return {
-            "loss": loss.detach(),
-            "value": q_taken.detach().mean(),
-            "td_abs": td.detach().abs().mean(),
-        }
+        return stats
 
     def _inner_step(self):
         data, idx, w = self.replay.sample(self.batch_size, self.beta)
@@ -288,24 +302,48 @@ class R2D2Learner(LearnerBase):
     def make_graphed_step(self, warmup_iters: int = 3):
         """hipGraph-capture the whole R2D2 step (PER sample -> burn-in +
         train + target sequence passes -> n-step rescaled targets -> loss ->
-        backward -> Adam -> sequence-priority update)."""
+        backward -> Adam -> sequence-priority update). At world_size > 1 the
+        step is captured as TWO graphs with the RCCL all-reduce (one per
+        dtype group) running eagerly between them — collectives are never
+        captured (same structure as ApexLearner.make_graphed_step)."""
         assert self.device.type == "cuda"
-        assert self.reducer is None, "graphed R2D2 step is single-replica"
         for g in self.optim.param_groups:
             g["capturable"] = True
+        split = self.mp is not None and self.mp.world > 1
         side = torch.cuda.Stream(self.device)
         side.wait_stream(torch.cuda.current_stream(self.device))
         with torch.cuda.stream(side):
             for _ in range(warmup_iters):
                 self._inner_step()
         torch.cuda.current_stream(self.device).wait_stream(side)
-        graph = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(graph):
-            static_out = self._inner_step()
-        self._graph = graph
+
+        if not split:
+            graph = torch.cuda.CUDAGraph()
+            with torch.cuda.graph(graph):
+                static_out = self._inner_step()
+            self._graph = graph
+
+            def stepper():
+                graph.replay()
+                self._cadence()
+                return static_out
+
+            return stepper
+
+        g1 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g1):
+            data, s_idx, s_w = self.replay.sample(self.batch_size, self.beta)
+            static_out, prio = self._fwd_bwd(data, s_w)
+        g2 = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g2, pool=g1.pool()):
+            self._optimize_mp()
+            self.replay.update(s_idx, prio)
+        self._graph = (g1, g2)
 
         def stepper():
-            graph.replay()
+            g1.replay()
+            self.mp.allreduce_grads()
+            g2.replay()
             self._cadence()
             return static_out
 

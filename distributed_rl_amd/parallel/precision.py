@@ -163,14 +163,24 @@ class MixedPrecisionTrainer:
         for g in self.groups:
             g.flat_cgrad.zero_()
 
-    def reduce_and_upcast(self):
-        """all-reduce compute grads (if distributed) then cast to fp32."""
-        for g in self.groups:
-            if self.world > 1:
+    def allreduce_grads(self):
+        """all-reduce the compute-dtype grad flats (eager; NEVER inside a
+        hipGraph capture — collectives are not capturable)."""
+        if self.world > 1:
+            for g in self.groups:
                 dist.all_reduce(g.flat_cgrad, group=self.pg)
+
+    def upcast_grads(self):
+        """compute-dtype grad flats -> fp32 master grads (+ 1/world)."""
+        for g in self.groups:
             g.flat_mgrad.copy_(g.flat_cgrad)
             if self.world > 1:
                 g.flat_mgrad.mul_(1.0 / self.world)
+
+    def reduce_and_upcast(self):
+        """all-reduce compute grads (if distributed) then cast to fp32."""
+        self.allreduce_grads()
+        self.upcast_grads()
 
     def sync_compute_params(self):
         """fp32 master -> compute replica (one cast kernel per group)."""
