@@ -665,8 +665,6 @@ void impala_loss_fwd(torch::Tensor logpa, torch::Tensor adv,
                      torch::Tensor mean_H, torch::Tensor v, torch::Tensor vs,
                      double er, torch::Tensor loss_out, torch::Tensor obj_out,
                      torch::Tensor critic_out);
-void impala_critic_bwd(torch::Tensor v, torch::Tensor vs, torch::Tensor gloss,
-                       torch::Tensor dv);
 void impala_out_bwd(torch::Tensor pi_save, torch::Tensor H_save,
                     torch::Tensor act, torch::Tensor adv, torch::Tensor v,
                     torch::Tensor vs, torch::Tensor gloss, int64_t B,
@@ -712,8 +710,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("policy_loss_bwd", &policy_loss_bwd, "fused IMPALA policy obj bwd (K9)");
   m.def("impala_loss_fwd", &impala_loss_fwd,
         "fused IMPALA total loss: pg objective + entropy + critic MSE (K9)");
-  m.def("impala_critic_bwd", &impala_critic_bwd,
-        "IMPALA critic MSE backward: dv = g*(v-vs)/n (K9)");
   m.def("impala_out_bwd", &impala_out_bwd,
         "IMPALA whole-head backward: d loss/d out in one launch (K9)");
   m.def("rmsprop_step", &rmsprop_step, "fused flat centered RMSprop (K12)");
@@ -1007,18 +1003,6 @@ __global__ void impala_loss_fwd_kernel(
   }
 }
 
-// d loss / d v = gloss * (v - vs) / n
-__global__ void impala_critic_bwd_kernel(const float* __restrict__ v,
-                                         const float* __restrict__ vs,
-                                         const float* __restrict__ gloss,
-                                         int64_t n, float* __restrict__ dv) {
-  float g = gloss[0] / (float)n;
-  int64_t stride = (int64_t)gridDim.x * blockDim.x;
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += stride)
-    dv[i] = g * (v[i] - vs[i]);
-}
-
 // Whole-output IMPALA backward: writes d loss / d out for the raw network
 // head out = (B*(T+1), A+1) in ONE launch — policy grad into columns 0..A-1
 // for t<T, critic grad into column A for t<T, zeros elsewhere (t==T rows
@@ -1096,15 +1080,6 @@ void impala_loss_fwd(torch::Tensor logpa, torch::Tensor adv,
                      v.data_ptr<float>(), vs.data_ptr<float>(), n, (float)er,
                      loss_out.data_ptr<float>(), obj_out.data_ptr<float>(),
                      critic_out.data_ptr<float>());
-}
-
-void impala_critic_bwd(torch::Tensor v, torch::Tensor vs, torch::Tensor gloss,
-                       torch::Tensor dv) {
-  int64_t n = v.numel();
-  hipLaunchKernelGGL(impala_critic_bwd_kernel, dim3(grid_for(n, 4)),
-                     dim3(kBlock), 0, cur_stream(), v.data_ptr<float>(),
-                     vs.data_ptr<float>(), gloss.data_ptr<float>(), n,
-                     dv.data_ptr<float>());
 }
 
 void impala_out_bwd(torch::Tensor pi_save, torch::Tensor H_save,
