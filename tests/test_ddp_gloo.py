@@ -76,3 +76,94 @@ def test_dp_replicas_stay_in_sync():
     assert abs(g[0] - g[1]) < 1e-6, g
     # and learning actually moved the weights
     assert abs(g[0] - g0[0]) > 1e-9
+
+
+def _worker_alg(rank, world, port, alg, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        torch.manual_seed(99 + rank)  # intentionally different init
+        from distributed_rl_amd.parallel import attach_reducer
+
+        g = torch.Generator().manual_seed(rank)
+        if alg == "impala":
+            from distributed_rl_amd.algos.impala import ImpalaLearner
+
+            raw = copy.deepcopy(load_config("impala").raw)
+            raw["BATCHSIZE"] = 4
+            raw["REPLAY_MEMORY_LEN"] = 64
+            cfg = Config(raw=raw)
+            learner = ImpalaLearner(cfg, device="cpu", enable_tb=False,
+                                    world_size=world, rank=rank)
+            B, T = 8, cfg.unroll_step
+            cols = {
+                "states": torch.randint(0, 255, (B, T + 1, 4, 84, 84),
+                                        dtype=torch.uint8, generator=g),
+                "actions": torch.randint(0, 6, (B, T), dtype=torch.int32,
+                                         generator=g),
+                "mu": torch.full((B, T), 1 / 6),
+                "rewards": torch.randn(B, T, generator=g),
+                "not_done": torch.ones(B),
+            }
+            attach_reducer(learner)
+            learner.push_trajectories(cols)
+        else:
+            from distributed_rl_amd.algos.r2d2 import R2D2Learner
+
+            raw = copy.deepcopy(load_config("r2d2").raw)
+            raw["BATCHSIZE"] = 2
+            raw["REPLAY_MEMORY_LEN"] = 16
+            raw["BUFFER_SIZE"] = 2
+            raw["N"] = 2
+            raw["FIXED_TRAJECTORY"] = 16
+            raw["MEM"] = 4
+            cfg = Config(raw=raw)
+            learner = R2D2Learner(cfg, device="cpu", enable_tb=False,
+                                  world_size=world, rank=rank)
+            B, T, H = 4, cfg.fixed_trajectory, 512
+            cols = {
+                "h0": torch.zeros(B, 2, H),
+                "states": torch.randint(0, 255, (B, T, 4, 84, 84),
+                                        dtype=torch.uint8, generator=g),
+                "actions": torch.randint(0, 6, (B, T), dtype=torch.int32,
+                                         generator=g),
+                "rewards": torch.randn(B, T, generator=g),
+                "done": torch.zeros(B),
+            }
+            attach_reducer(learner)
+            learner.push_sequences(cols, torch.rand(B, generator=g) + 0.1)
+        h0 = sum(p.double().sum().item() for p in learner.model.parameters())
+        for _ in range(2):
+            learner.step()
+        h = sum(p.double().sum().item() for p in learner.model.parameters())
+        gathered0 = [None] * world
+        gathered = [None] * world
+        dist.all_gather_object(gathered0, h0)
+        dist.all_gather_object(gathered, h)
+        if rank == 0:
+            result_q.put((gathered0, gathered))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+@pytest.mark.parametrize("alg,port", [("impala", 29632), ("r2d2", 29633)])
+def test_dp_replicas_stay_in_sync_impala_r2d2(alg, port):
+    """Same world-2 bit-sync property for the other two algorithms — guards
+    the staged train_step (_fwd_bwd / eager all-reduce / optimize) that the
+    split-graph capture relies on."""
+    ctx = tmp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_alg, args=(r, 2, port, alg, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    g0, g = q.get(timeout=240)
+    for p in procs:
+        p.join(30)
+    assert abs(g0[0] - g0[1]) < 1e-9, g0
+    assert abs(g[0] - g[1]) < 1e-6, g
+    assert abs(g[0] - g0[0]) > 1e-9
