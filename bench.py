@@ -155,6 +155,12 @@ def main():
     rank, local_rank, world = init_distributed()
     has_cuda = torch.cuda.is_available()
     device = f"cuda:{local_rank}" if has_cuda else "cpu"
+    if world == 1 and args.gpus > 1:
+        # N>1 must come through torchrun (one rank per GPU); a single process
+        # only ever drives one GPU — never multiply frames by an unused N.
+        print(f"# --gpus {args.gpus} without torchrun: clamping to 1",
+              file=sys.stderr)
+        args.gpus = 1
     n_gpus = world if world > 1 else (args.gpus if has_cuda else 0)
 
     cfg = load_config(args.cfg)
