@@ -43,6 +43,9 @@ class LearnerBase:
         self._enable_tb = enable_tb and rank == 0
         self._timing: Dict[str, float] = {}
         self._t_block = time.perf_counter()
+        # opt-in replay persistence (SURVEY §5.4 "PER state optional"):
+        # off by default — a full Ape-X buffer is multi-GB on disk
+        self.checkpoint_replay = os.environ.get("DRL_CKPT_REPLAY", "0") == "1"
 
     # -- model helpers ----------------------------------------------------
     def build_model(self) -> BaseAgent:
@@ -125,6 +128,8 @@ class LearnerBase:
             return ""
         d = self.checkpoint_dir()
         state = self.state_for_checkpoint()
+        if self.checkpoint_replay and getattr(self, "replay", None) is not None:
+            state["replay"] = self.replay.state_dict()
         # reference-format model-only file (contract: weight.pth)
         torch.save(state["model"], os.path.join(d, "weight.pth"))
         torch.save(state, os.path.join(d, "resume.pt"))
@@ -132,10 +137,12 @@ class LearnerBase:
 
     def resume(self, path: str) -> None:
         """Load either a reference-style weight.pth (model only) or a full
-        resume.pt."""
+        resume.pt (model + optimizer + step counter + optional replay)."""
         state = torch.load(path, map_location=self.device, weights_only=False)
         if isinstance(state, dict) and "model" in state and "step" in state:
             self.load_from_checkpoint(state)
+            if "replay" in state and getattr(self, "replay", None) is not None:
+                self.replay.load_state_dict(state["replay"])
         else:
             self.load_model_only(state)
 

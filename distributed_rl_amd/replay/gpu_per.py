@@ -51,6 +51,9 @@ class HipSumTreePER(ReplayBase):
     def update(self, idx: torch.Tensor, prios: torch.Tensor):
         self._set_priorities(idx, prios)
 
+    def _leaf_priorities(self):
+        return self.tree[self.P : self.P + self.capacity]
+
     @property
     def total_priority(self) -> float:
         return float(self.tree[1])

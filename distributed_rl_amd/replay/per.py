@@ -68,6 +68,36 @@ class ReplayBase:
     def gather(self, idx: torch.Tensor) -> Dict[str, torch.Tensor]:
         return {name: col.index_select(0, idx) for name, col in self.data.items()}
 
+    # -- checkpoint (optional; SURVEY §5.4 "PER state optional") ----------
+    def _leaf_priorities(self) -> torch.Tensor:
+        """Priorities of the filled region, ring order (subclass hook)."""
+        raise NotImplementedError
+
+    def state_dict(self) -> Dict[str, object]:
+        """Serializable replay contents (filled region only, CPU tensors).
+        Opt-in via the learner's ``checkpoint_replay`` flag — Ape-X's 100k
+        uint8 frames are ~5.6 GB on disk, so this is NOT saved by default."""
+        n = len(self)
+        return {
+            "n": n,
+            "write_pos": self.write_pos,
+            "count": self.count,
+            "data": {k: v[:n].to("cpu") for k, v in self.data.items()},
+            "priorities": self._leaf_priorities()[:n].to("cpu"),
+        }
+
+    def load_state_dict(self, state: Dict[str, object]) -> None:
+        n = int(state["n"])
+        if n > self.capacity:
+            raise ValueError("checkpointed replay larger than capacity")
+        for k, col in state["data"].items():
+            self.data[k][:n].copy_(col.to(self.device))
+        idx = torch.arange(n, device=self.device)
+        self._set_priorities(idx, state["priorities"].to(self.device,
+                                                         torch.float32))
+        self.write_pos = int(state["write_pos"])
+        self.count = int(state["count"])
+
     # -- priority machinery (implemented by subclasses) -------------------
     def _set_priorities(self, idx: torch.Tensor, prios: torch.Tensor):
         raise NotImplementedError
@@ -95,6 +125,9 @@ class TorchPER(ReplayBase):
 
     def _set_priorities(self, idx, prios):
         self.priorities.index_copy_(0, idx, prios)
+
+    def _leaf_priorities(self):
+        return self.priorities
 
     def update(self, idx, prios):
         idx = idx.to(self.device)
@@ -167,6 +200,9 @@ class FifoReplay(ReplayBase):
 
     def _set_priorities(self, idx, prios):
         pass
+
+    def _leaf_priorities(self):
+        return torch.ones(self.capacity, device=self.device)
 
     def push(self, columns: Dict[str, torch.Tensor], n: Optional[int] = None):
         first = next(iter(columns.values()))

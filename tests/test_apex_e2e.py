@@ -110,3 +110,34 @@ def test_checkpoint_resume(tmp_path):
     l3.resume(path)
     for p, q in zip(l3.model.parameters(), learner.model.parameters()):
         assert torch.equal(p, q)
+
+
+def test_checkpoint_resume_with_replay_state(tmp_path, monkeypatch):
+    """DRL_CKPT_REPLAY=1 persists the PER contents (data, priorities, ring
+    counters) through resume.pt (SURVEY §5.4 'PER state optional')."""
+    monkeypatch.setenv("DRL_CKPT_REPLAY", "1")
+    cfg = small_cfg()
+    learner = ApexLearner(cfg, device="cpu", enable_tb=False,
+                          run_root=str(tmp_path))
+    B = 64
+    cols = {
+        "state": torch.randint(0, 255, (B, 4, 84, 84), dtype=torch.uint8),
+        "action": torch.randint(0, 6, (B,), dtype=torch.int32),
+        "reward": torch.rand(B),
+        "next_state": torch.randint(0, 255, (B, 4, 84, 84), dtype=torch.uint8),
+        "done": torch.zeros(B),
+    }
+    learner.push_experience(cols, torch.rand(B) + 0.1)
+    learner.step()
+    path = learner.save_checkpoint()
+    resume_path = os.path.join(os.path.dirname(path), "resume.pt")
+    l2 = ApexLearner(cfg, device="cpu", enable_tb=False, run_root=str(tmp_path))
+    assert len(l2.replay) == 0
+    l2.resume(resume_path)
+    assert len(l2.replay) == len(learner.replay) == B
+    assert l2.replay.write_pos == learner.replay.write_pos
+    for k in cols:
+        assert torch.equal(l2.replay.data[k][:B], learner.replay.data[k][:B])
+    assert torch.allclose(l2.replay.priorities[:B],
+                          learner.replay.priorities[:B])
+    l2.step()  # stepping straight off the restored buffer works
