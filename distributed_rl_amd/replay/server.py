@@ -30,7 +30,7 @@ import numpy as np
 import torch
 
 from ..actors.transport import RecordCodec
-from ..actors.tcp_transport import _HDR, _recv, _send, K_EXP
+from ..actors.tcp_transport import _recv, _send, K_EXP
 from .per import TorchPER
 
 K_BATCH_REQ = 10

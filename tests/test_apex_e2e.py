@@ -2,7 +2,6 @@
 in-proc pipe -> replay -> learner train steps (SURVEY §7 stage 2)."""
 
 import copy
-import json
 import os
 
 import numpy as np

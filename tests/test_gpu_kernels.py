@@ -1,6 +1,5 @@
 """HIP kernel numerics vs the plain-PyTorch fp32 references (torch_ref)."""
 
-import numpy as np
 import pytest
 import torch
 

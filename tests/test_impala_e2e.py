@@ -7,9 +7,7 @@ import numpy as np
 import torch
 
 from distributed_rl_amd.actors.transport import InprocPipe
-from distributed_rl_amd.algos.impala import (
-    ImpalaLearner, ImpalaPlayer, make_impala_schema,
-)
+from distributed_rl_amd.algos.impala import ImpalaLearner, ImpalaPlayer
 from distributed_rl_amd.config import Config, load_config
 
 

@@ -2,7 +2,6 @@
 invariants — the pieces where silent corruption would be hardest to spot."""
 
 import numpy as np
-import pytest
 import torch
 from hypothesis import given, settings, strategies as st
 

@@ -8,7 +8,7 @@ import torch
 
 from distributed_rl_amd.actors.transport import InprocPipe
 from distributed_rl_amd.algos.r2d2 import (
-    R2D2Learner, R2D2Player, nstep_recurrent_targets, ETA,
+    R2D2Learner, R2D2Player, nstep_recurrent_targets,
 )
 from distributed_rl_amd.config import Config, load_config
 from distributed_rl_amd.ops import torch_ref

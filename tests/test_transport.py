@@ -1,5 +1,4 @@
 import numpy as np
-import pytest
 import torch
 
 from distributed_rl_amd.actors.transport import (
