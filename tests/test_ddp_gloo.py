@@ -167,3 +167,58 @@ def test_dp_replicas_stay_in_sync_impala_r2d2(alg, port):
     assert abs(g0[0] - g0[1]) < 1e-9, g0
     assert abs(g[0] - g[1]) < 1e-6, g
     assert abs(g[0] - g0[0]) > 1e-9
+
+
+def _worker_equiv(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["RANK"] = str(rank)
+    os.environ["WORLD_SIZE"] = str(world)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    try:
+        from distributed_rl_amd.models import BaseAgent
+        from distributed_rl_amd.parallel.ddp import FlatGradReducer
+
+        torch.manual_seed(7)  # same init everywhere
+        net = BaseAgent(load_config("ape_x").model_info)
+        reducer = FlatGradReducer(list(net.parameters()))
+        torch.manual_seed(11)  # same data stream everywhere
+        B = 8
+        x = torch.rand(world * B, 4, 84, 84)
+        y = torch.randn(world * B, 6)
+        xs, ys = x[rank * B:(rank + 1) * B], y[rank * B:(rank + 1) * B]
+        reducer.zero_()
+        out = net.forward([xs])[0]
+        # per-rank mean over B, averaged by the reducer == mean over world*B
+        ((out - ys) ** 2).mean().backward()
+        reducer.all_reduce()
+        gsum = float(reducer.flat.double().abs().sum())
+
+        # single-process oracle: full batch through an identical net
+        torch.manual_seed(7)
+        net1 = BaseAgent(load_config("ape_x").model_info)
+        out1 = net1.forward([x])[0]
+        ((out1 - y) ** 2).mean().backward()
+        ref = float(sum(p.grad.double().abs().sum() for p in net1.parameters()
+                        if p.grad is not None))
+        if rank == 0:
+            result_q.put((gsum, ref))
+    finally:
+        dist.destroy_process_group()
+
+
+@pytest.mark.timeout(300)
+def test_dp2_gradient_equals_single_large_batch():
+    """SURVEY §4 build implication: summed DP gradients must equal the
+    single-process large-batch gradient (here world=2 over gloo; the RCCL
+    path shares the code)."""
+    ctx = tmp.get_context("spawn")
+    q = ctx.Queue()
+    procs = [ctx.Process(target=_worker_equiv, args=(r, 2, 29634, q))
+             for r in range(2)]
+    for p in procs:
+        p.start()
+    gsum, ref = q.get(timeout=240)
+    for p in procs:
+        p.join(30)
+    assert abs(gsum - ref) / max(ref, 1e-12) < 1e-5, (gsum, ref)
