@@ -193,7 +193,16 @@ class LSTMNET(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         p = next(self.lstm.parameters())
-        if x.dtype != p.dtype:
+        # DRL_LSTM_BF16_IN=1 (round-2 prototype, default off): keep bf16
+        # trunk features bf16 and run the big input projection GEMMs in
+        # bf16 inside manual_lstm_seq — the recurrence/state stays fp32.
+        import os as _os
+
+        bf16_in = (
+            x.dtype == torch.bfloat16 and x.is_cuda and self.num_layers == 1
+            and _os.environ.get("DRL_LSTM_BF16_IN", "0") == "1"
+        )
+        if x.dtype != p.dtype and not bf16_in:
             # mixed-trunk path: bf16 conv features into the fp32 LSTM
             x = x.to(p.dtype)
         if x.dim() == 2:
@@ -202,8 +211,8 @@ class LSTMNET(nn.Module):
         if self._state is None or self._state[0].shape[1] != batch:
             self.zero_cell_state(batch)
         st = self._state
-        if st[0].dtype != x.dtype or st[0].device != x.device:
-            st = (st[0].to(x.device, x.dtype), st[1].to(x.device, x.dtype))
+        if st[0].dtype != p.dtype or st[0].device != x.device:
+            st = (st[0].to(x.device, p.dtype), st[1].to(x.device, p.dtype))
         if x.is_cuda and self.num_layers == 1:
             # K5-lite: fused-cell manual recurrence (hipGraph-capturable;
             # MIOpen's RNN path is not) — models/manual_lstm.py

@@ -32,7 +32,19 @@ class _ManualLSTMSeq(torch.autograd.Function):
         H = w_hh.shape[1]
         dev = x.device
         bias = b_ih + b_hh
-        xp = torch.addmm(bias, x.reshape(T * B, IN), w_ih.t()).view(T, B, 4 * H)
+        if x.dtype == torch.bfloat16:
+            # bf16 input projection (DRL_LSTM_BF16_IN): the (T*B, IN) GEMM
+            # runs on bf16 MFMA (fp32 accumulate inside hipBLASLt) instead
+            # of fp32 — ~8x the rate at these shapes — and the fp32 upcast
+            # happens on the 4H-wide gates instead of the IN-wide input.
+            # Recurrence, cell state and cell kernels stay fp32.
+            w_ih_c = w_ih.to(torch.bfloat16)
+            xp = torch.mm(x.reshape(T * B, IN), w_ih_c.t()).float()
+            xp.add_(bias)
+            xp = xp.view(T, B, 4 * H)
+        else:
+            xp = torch.addmm(bias, x.reshape(T * B, IN),
+                             w_ih.t()).view(T, B, 4 * H)
         hs = torch.empty(T + 1, B, H, device=dev)
         cs = torch.empty(T + 1, B, H, device=dev)
         acts = torch.empty(T, B, 4 * H, device=dev)
@@ -86,9 +98,17 @@ class _ManualLSTMSeq(torch.autograd.Function):
             dh = dgates_all[t].mm(w_hh)
         dg_flat = dgates_all.reshape(T * B, 4 * H)
         dw_hh = dg_flat.t().mm(hs[:-1].reshape(T * B, H))
-        dw_ih = dg_flat.t().mm(x.reshape(T * B, IN))
         db = dg_flat.sum(0)
-        dx = dg_flat.mm(w_ih).view(T, B, IN)
+        if x.dtype == torch.bfloat16:
+            # bf16 weight/input grads (fp32-accumulated inside the GEMM,
+            # rounded to bf16 on output — same precision class as the conv
+            # trunk's bf16 grads that feed the same fp32 master upcast)
+            dg_bf = dg_flat.to(torch.bfloat16)
+            dw_ih = dg_bf.t().mm(x.reshape(T * B, IN)).float()
+            dx = dg_bf.mm(w_ih.to(torch.bfloat16)).view(T, B, IN)
+        else:
+            dw_ih = dg_flat.t().mm(x.reshape(T * B, IN))
+            dx = dg_flat.mm(w_ih).view(T, B, IN)
         return dx, dh, dc, dw_ih, dw_hh, db, db
 
 
@@ -99,8 +119,9 @@ def manual_lstm_seq(x: torch.Tensor, state: Tuple[torch.Tensor, torch.Tensor],
     Returns (out (T,B,H), (h_T (1,B,H), c_T (1,B,H))) like nn.LSTM."""
     assert lstm.num_layers == 1 and not lstm.bidirectional
     h0, c0 = state
+    x_in = x if x.dtype == torch.bfloat16 else x.float()
     out, hT, cT = _ManualLSTMSeq.apply(
-        x.float(), h0[0].float(), c0[0].float(), lstm.weight_ih_l0,
+        x_in, h0[0].float(), c0[0].float(), lstm.weight_ih_l0,
         lstm.weight_hh_l0, lstm.bias_ih_l0, lstm.bias_hh_l0,
     )
     return out, (hT.unsqueeze(0), cT.unsqueeze(0))

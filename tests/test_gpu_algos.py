@@ -120,6 +120,43 @@ def test_manual_lstm_matches_nn_lstm():
     assert torch.allclose(gw, lstm.weight_hh_l0.grad, atol=1e-3)
 
 
+def test_manual_lstm_bf16_input_close_to_fp32():
+    """Round-2 prototype (DRL_LSTM_BF16_IN): bf16 input-projection GEMMs,
+    fp32 recurrence. Values/grads must track the fp32 path within bf16
+    rounding (outputs are tanh-bounded, so absolute tolerance is safe)."""
+    from distributed_rl_amd.models.manual_lstm import manual_lstm_seq
+
+    torch.manual_seed(13)
+    T, B, IN, H = 8, 4, 64, 32
+    lstm = torch.nn.LSTM(IN, H).to(DEV)
+    h0 = torch.randn(1, B, H, device=DEV)
+    c0 = torch.randn(1, B, H, device=DEV)
+    x32 = torch.randn(T, B, IN, device=DEV)
+    xbf = x32.to(torch.bfloat16).requires_grad_(True)
+    x = x32.clone().requires_grad_(True)
+
+    out, (hT, cT) = manual_lstm_seq(x, (h0, c0), lstm)
+    g = torch.randn_like(out)
+    out.backward(g)
+    gw_ih = lstm.weight_ih_l0.grad.clone()
+    gx = x.grad.clone()
+    for p in lstm.parameters():
+        p.grad = None
+
+    out_bf, (hT_bf, cT_bf) = manual_lstm_seq(xbf, (h0, c0), lstm)
+    assert out_bf.dtype == torch.float32  # recurrence stays fp32
+    out_bf.backward(g)
+    assert xbf.grad.dtype == torch.bfloat16
+
+    assert torch.allclose(out_bf, out, atol=0.05), (out_bf - out).abs().max()
+    assert torch.allclose(hT_bf, hT, atol=0.05)
+    assert torch.allclose(cT_bf, cT, atol=0.08)
+    assert torch.allclose(xbf.grad.float(), gx, atol=0.08,
+                          rtol=0.05), (xbf.grad.float() - gx).abs().max()
+    rel = (lstm.weight_ih_l0.grad - gw_ih).abs().max() / gw_ih.abs().max()
+    assert rel < 0.05, rel
+
+
 def test_r2d2_gpu_graphed_step():
     from distributed_rl_amd.algos.r2d2 import R2D2Learner
     from distributed_rl_amd.config import Config, load_config
