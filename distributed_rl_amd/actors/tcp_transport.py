@@ -119,7 +119,9 @@ class TcpTransportServer:
                     if blob is None or ver == have:
                         _send(conn, K_NONE, b"")
                     else:
-                        _send(conn, K_WEIGHTS, blob)
+                        # u64 version prefix lets the actor skip unchanged
+                        # snapshots on later polls (multi-MB state_dicts)
+                        _send(conn, K_WEIGHTS, struct.pack("<Q", ver) + blob)
                 elif kind == K_REWARD:
                     with self._lock:
                         self._rewards.append(pickle.loads(payload))
@@ -169,6 +171,7 @@ class TcpActorEndpoint:
         self.sock = socket.create_connection((host, port), timeout=60)
         self.sock.setsockopt(socket.IPPROTO_TCP, socket.TCP_NODELAY, 1)
         self._last_version_payload = None
+        self._have = -1  # server-side version of the cached snapshot
 
     def push(self, columns: Dict[str, np.ndarray],
              priorities: Optional[np.ndarray] = None):
@@ -176,14 +179,17 @@ class TcpActorEndpoint:
         _send(self.sock, K_EXP, rec.tobytes())
 
     def fetch(self):
-        _send(self.sock, K_FETCH, pickle.dumps(-1))
+        _send(self.sock, K_FETCH, pickle.dumps(self._have))
         msg = _recv(self.sock)
         if msg is None:
             return None
         kind, payload = msg
         if kind != K_WEIGHTS:
+            # K_NONE: nothing published yet, or our cached version is
+            # current — no multi-MB re-download
             return self._last_version_payload
-        self._last_version_payload = pickle.loads(payload)
+        (self._have,) = struct.unpack_from("<Q", payload)
+        self._last_version_payload = pickle.loads(payload[8:])
         return self._last_version_payload
 
     def push_reward(self, _idx, reward: float, eps: float = 0.0):

@@ -27,6 +27,15 @@ def test_tcp_roundtrip():
         le.publish({"count": 3, "state_dict": {"w": torch.ones(2)}})
         got = ep.fetch()
         assert got["count"] == 3
+        # unchanged version: second poll is served from the client cache
+        # (server answers K_NONE, no blob re-transfer)
+        have = ep._have
+        assert have >= 1
+        got2 = ep.fetch()
+        assert got2 is got and ep._have == have
+        le.publish({"count": 4, "state_dict": {"w": torch.ones(2)}})
+        got3 = ep.fetch()
+        assert got3["count"] == 4 and ep._have == have + 1
         n = 4
         cols = {
             "state": np.zeros((n, 4, 84, 84), np.uint8),
