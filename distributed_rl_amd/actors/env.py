@@ -30,8 +30,8 @@ class SyntheticEnv:
     noise whose mean encodes a hidden scalar state; reward = +1 when the
     action matches (hidden_state mod action_n), else small negative."""
 
-    def __init__(self, action_n: int = 6, seed: int = 0, episode_len: int = 512,
-                 frame_cost_flops: int = 0):
+    def __init__(self, action_n: int = 6, seed: int = 0,
+                 episode_len: int = 512):
         self.action_n = action_n
         self.rng = np.random.default_rng(seed)
         self.episode_len = episode_len
