@@ -70,17 +70,22 @@ def test_clean_transport_cli(tmp_path):
     sess = TransportSession(str(tmp_path / "t"), codec, num_rings=1,
                             ring_slots=4, weight_capacity=1 << 16, create=True)
     name = f"drl_{sess.session}_w"
-    out = subprocess.run(
-        [sys.executable, os.path.join(REPO, "clean_transport.py"),
-         "--transport-dir", str(tmp_path / "t")],
-        capture_output=True, text=True, timeout=120,
-    )
-    assert out.returncode == 0, out.stdout + out.stderr
-    assert "removed session" in out.stdout
-    from multiprocessing import shared_memory
+    try:
+        out = subprocess.run(
+            [sys.executable, os.path.join(REPO, "clean_transport.py"),
+             "--transport-dir", str(tmp_path / "t")],
+            capture_output=True, text=True, timeout=120,
+        )
+        assert out.returncode == 0, out.stdout + out.stderr
+        assert "removed session" in out.stdout
+        from multiprocessing import shared_memory
 
-    with pytest.raises(FileNotFoundError):
-        shared_memory.SharedMemory(name=name)
+        with pytest.raises(FileNotFoundError):
+            shared_memory.SharedMemory(name=name)
+    finally:
+        # release our mappings of the externally-unlinked segments so
+        # SharedMemory.__del__ never fires with live numpy views
+        sess.close()
 
 
 @pytest.mark.timeout(240)
