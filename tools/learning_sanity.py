@@ -56,7 +56,47 @@ def main(rounds=50, steps_per_round=600, train_per_round=60):
     return early, late
 
 
+def main_impala(rounds=30, steps_per_round=800, train_per_round=12):
+    """Same harness for IMPALA (on-policy; actors re-pull weights each
+    round, learner trains on the freshest unrolls)."""
+    from distributed_rl_amd.algos.impala import ImpalaLearner, ImpalaPlayer
+
+    torch.manual_seed(0)
+    raw = copy.deepcopy(load_config("impala").raw)
+    raw.update({"REPLAY_MEMORY_LEN": 256, "BATCHSIZE": 16, "N": 2})
+    cfg = Config(raw=raw)
+    pipe = InprocPipe()
+    learner = ImpalaLearner(cfg, device="cpu", transport=pipe, enable_tb=False)
+    learner.publish_weights()
+    players = [
+        ImpalaPlayer(cfg, idx=i, transport=pipe,
+                     env=SyntheticEnv(seed=i, episode_len=150))
+        for i in range(2)
+    ]
+    curve = []
+    for r in range(rounds):
+        for p in players:
+            p.run(max_env_steps=p.env_steps + steps_per_round)
+        learner.ingest()
+        if len(learner.replay) >= cfg.batch_size:
+            for _ in range(train_per_round):
+                learner.step()
+        rs = pipe.drain_rewards()
+        if rs:
+            curve.append(float(np.mean(rs)))
+            print(f"round {r}: mean_ep_reward {curve[-1]:8.2f} "
+                  f"steps {learner.step_count} replay {len(learner.replay)}",
+                  flush=True)
+    early = np.mean(curve[:5])
+    late = np.mean(curve[-5:])
+    print(f"EARLY {early:.2f} LATE {late:.2f}")
+    return early, late
+
+
 if __name__ == "__main__":
-    e, l = main()
+    if "--alg" in sys.argv and "impala" in sys.argv[sys.argv.index("--alg") + 1]:
+        e, l = main_impala()
+    else:
+        e, l = main()
     assert l > e + 10, f"no learning signal: {e} -> {l}"
     print("LEARNING OK")
