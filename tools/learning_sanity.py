@@ -93,9 +93,54 @@ def main_impala(rounds=30, steps_per_round=800, train_per_round=12):
     return early, late
 
 
+def main_r2d2(rounds=30, steps_per_round=700, train_per_round=10):
+    """R2D2 harness, scaled down for CPU (16-step sequences, 4-step
+    burn-in via cfg overrides; the 80/20 production shape only changes
+    sizes, not code paths)."""
+    from distributed_rl_amd.algos.r2d2 import R2D2Learner, R2D2Player
+
+    torch.manual_seed(0)
+    raw = copy.deepcopy(load_config("r2d2").raw)
+    raw.update({"REPLAY_MEMORY_LEN": 512, "BUFFER_SIZE": 32, "BATCHSIZE": 8,
+                "N": 2, "FIXED_TRAJECTORY": 16, "MEM": 4, "UNROLL_STEP": 3,
+                "TARGET_FREQUENCY": 100})
+    cfg = Config(raw=raw)
+    pipe = InprocPipe()
+    learner = R2D2Learner(cfg, device="cpu", transport=pipe, enable_tb=False)
+    learner.publish_weights(include_target=True)
+    players = [
+        R2D2Player(cfg, idx=i, transport=pipe,
+                   env=SyntheticEnv(seed=i, episode_len=150))
+        for i in range(2)
+    ]
+    players[1].eps = 0.02
+    curve = []
+    for r in range(rounds):
+        for p in players:
+            p.run(max_env_steps=p.env_steps + steps_per_round)
+        learner.ingest()
+        if len(learner.replay) > cfg.buffer_size:
+            for _ in range(train_per_round):
+                learner.step()
+        rs = pipe.drain_rewards()
+        if rs:
+            curve.append(float(np.mean(rs)))
+            print(f"round {r}: mean_ep_reward {curve[-1]:8.2f} "
+                  f"steps {learner.step_count} replay {len(learner.replay)}",
+                  flush=True)
+    early = np.mean(curve[:5])
+    late = np.mean(curve[-5:])
+    print(f"EARLY {early:.2f} LATE {late:.2f}")
+    return early, late
+
+
 if __name__ == "__main__":
-    if "--alg" in sys.argv and "impala" in sys.argv[sys.argv.index("--alg") + 1]:
+    alg = (sys.argv[sys.argv.index("--alg") + 1]
+           if "--alg" in sys.argv else "ape_x")
+    if "impala" in alg:
         e, l = main_impala()
+    elif "r2d2" in alg:
+        e, l = main_r2d2()
     else:
         e, l = main()
     assert l > e + 10, f"no learning signal: {e} -> {l}"
