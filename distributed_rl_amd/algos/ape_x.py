@@ -211,6 +211,8 @@ class ApexLearner(LearnerBase):
             }
             self._staging["__prio__"] = torch.empty(
                 self._STAGE_ROWS, dtype=torch.float32).pin_memory()
+            self._stage_evt = torch.cuda.Event()
+            self._stage_busy = False
         return self._staging
 
     def ingest(self) -> int:
@@ -233,6 +235,11 @@ class ApexLearner(LearnerBase):
         done = 0
         while done < n:
             k = min(self._STAGE_ROWS, n - done)
+            if self._stage_busy:
+                # the pinned buffers may still be read by a previous async
+                # H2D — wait_stream below only orders GPU streams, it does
+                # NOT protect the host-side refill from racing that copy
+                self._stage_evt.synchronize()
             for name, arr in cols_np.items():
                 stage[name][:k].numpy()[:] = arr[done : done + k]
             stage["__prio__"][:k].numpy()[:] = prio_np[done : done + k]
@@ -253,9 +260,10 @@ class ApexLearner(LearnerBase):
                     dev_cols,
                     stage["__prio__"][:k].to(self.device, non_blocking=True),
                 )
+            self._stage_evt.record(self._ingest_stream)
+            self._stage_busy = True
             # replay state is consumed by the compute stream; order it after
-            # the ingest stream, and keep the staging slice stable until the
-            # H2D completed (the wait covers it)
+            # the ingest stream
             torch.cuda.current_stream(self.device).wait_stream(
                 self._ingest_stream)
             done += k
