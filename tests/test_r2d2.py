@@ -120,3 +120,33 @@ def test_r2d2_checkpoint_roundtrip(tmp_path):
     assert l2.step_count == 1
     for a, b in zip(l2.model.parameters(), learner.model.parameters()):
         assert torch.equal(a, b)
+
+
+def test_nstep_recurrent_targets_random_geometries():
+    """The vectorized target math must match the brute force for every
+    (T, burn_in, n_step) combination, including truncated-tail windows."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(T=st.integers(6, 24), burn=st.integers(0, 8), n=st.integers(1, 8),
+           rescale=st.booleans(), seed=st.integers(0, 10_000))
+    def check(T, burn, n, rescale, seed):
+        if burn >= T - 1:
+            return
+        torch.manual_seed(seed)
+        B, A = 3, 4
+        q_on = torch.randn(T, B, A)
+        q_tg = torch.randn(T, B, A)
+        actions = torch.randint(0, A, (T, B))
+        rewards = torch.randn(T, B)
+        done = (torch.rand(B) < 0.5).float()
+        td, q_taken, targets = nstep_recurrent_targets(
+            q_on, q_tg, actions, rewards, done, burn_in=burn, n_step=n,
+            gamma=0.97, use_rescaling=rescale)
+        expect = brute_force_targets(q_on, q_tg, actions, rewards, done,
+                                     burn, n, 0.97, rescale)
+        assert targets.shape == expect.shape
+        assert torch.allclose(targets, expect, atol=1e-5), (
+            T, burn, n, rescale, (targets - expect).abs().max())
+
+    check()
