@@ -573,7 +573,10 @@ class LocalBuffer:
             r = 0.0
             for i in range(self.n):
                 r += (self.gamma ** i) * self.buf[i][2]
-            out.append((s0, a0, r, next_state, 0.0))
+            # a full window emitted AT the terminal step still ends the
+            # episode: its bootstrap state is post-terminal, so done rides
+            # the emission step's flag (APE_X/Player.py get_traj(done))
+            out.append((s0, a0, r, next_state, 1.0 if done else 0.0))
             self.buf.popleft()
         if done:
             # flush remaining with truncated returns, done=1

@@ -140,3 +140,35 @@ def test_checkpoint_resume_with_replay_state(tmp_path, monkeypatch):
     assert torch.allclose(l2.replay.priorities[:B],
                           learner.replay.priorities[:B])
     l2.step()  # stepping straight off the restored buffer works
+
+
+def test_local_buffer_nstep_random_episodes():
+    """Every emitted transition's return equals the literal
+    sum_{i<k} gamma^i r_{t+i}; exactly one transition per env step is
+    emitted; tail flush marks done and truncates the window."""
+    from hypothesis import given, settings, strategies as st
+
+    @settings(max_examples=40, deadline=None)
+    @given(n=st.integers(1, 6), ep_len=st.integers(1, 15),
+           seed=st.integers(0, 9999))
+    def check(n, ep_len, seed):
+        rng = np.random.default_rng(seed)
+        gamma = 0.9
+        rewards = rng.normal(size=ep_len)
+        lb = LocalBuffer(n_step=n, gamma=gamma)
+        out = []
+        for t in range(ep_len):
+            lb.append(np.full((1,), t, np.uint8), t, float(rewards[t]))
+            out += lb.emit_ready(np.full((1,), t + 1, np.uint8),
+                                 done=(t == ep_len - 1))
+        assert len(out) == ep_len  # one transition per env step, all flushed
+        for s0, a0, r, sn, d in out:
+            t = int(a0)
+            k = min(n, ep_len - t)
+            expect = sum(gamma ** i * rewards[t + i] for i in range(k))
+            assert abs(r - expect) < 1e-6, (t, n, ep_len)
+            # done=1 exactly when the window reaches the episode end (the
+            # bootstrap state is terminal or post-terminal)
+            assert d == (1.0 if t + n >= ep_len else 0.0)
+
+    check()
