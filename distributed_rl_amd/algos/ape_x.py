@@ -554,7 +554,14 @@ class ApexLearner(LearnerBase):
 class LocalBuffer:
     """n-step transition assembly (APE_X/Player.py:19-60 semantics: emit
     [s_t, a_t, sum gamma^i r, s_{t+n}, done]); one transition per env step
-    once warm, flushing the tail with done=1 at episode end."""
+    once warm, flushing the tail with done=1 at episode end.
+
+    Deliberate divergence: the reference consumes n steps per emit
+    (non-overlapping windows, APE_X/Player.py:56 ``del storage[:3n]``) and
+    DROPS the non-aligned tail at episode end; we emit the Ape-X paper's
+    per-step sliding windows (every (s_t, a_t) becomes a transition) and
+    flush the tail with truncated returns. Terminal-step emissions carry
+    done=1 like the reference's (Player.py:33-44)."""
 
     def __init__(self, n_step: int, gamma: float):
         self.n = n_step
