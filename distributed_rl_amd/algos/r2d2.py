@@ -451,6 +451,12 @@ class R2D2Player:
         self.weight_version = -1
         self._steps: List[tuple] = []  # (state, action, reward)
         self._hiddens: List[tuple] = []  # cell state BEFORE each step
+        # rolling last-T history for the terminal emission
+        # (R2D2/Player.py:37-47 emits storage[-3*FT:] with done=True)
+        from collections import deque as _deque
+
+        self._recent: _deque = _deque(maxlen=self.T)
+        self._recent_hid: _deque = _deque(maxlen=self.T)
 
     # -- inference ---------------------------------------------------------
     @torch.no_grad()
@@ -487,10 +493,7 @@ class R2D2Player:
         mix = ETA * td.abs().max() + (1 - ETA) * td.abs().mean()
         return float(mix ** self.alpha)
 
-    def _emit(self, done: bool):
-        T = self.T
-        steps = self._steps[:T]
-        h0_state = self._hiddens[0]
+    def _emit(self, steps, h0_state, done: bool):
         states_np = np.stack([s[0] for s in steps])
         actions_np = np.array([s[1] for s in steps], np.int32)
         rewards_np = np.array([s[2] for s in steps], np.float32)
@@ -528,18 +531,28 @@ class R2D2Player:
         episode_reward = 0.0
         while self.env_steps < max_env_steps:
             h = self.model.getCellState()
-            self._hiddens.append((h[0].clone(), h[1].clone()))
+            hid = (h[0].clone(), h[1].clone())
+            self._hiddens.append(hid)
+            self._recent_hid.append(hid)
             action = self.act(state)
             next_state, reward, done, info = self.env.step(action)
             episode_reward += reward
             self._steps.append((state, action, reward))
+            self._recent.append((state, action, reward))
             if len(self._steps) == self.T:
-                self._emit(done=done)
+                self._emit(self._steps[: self.T], self._hiddens[0], done)
                 if done:
                     self._steps, self._hiddens = [], []
                 else:
                     self._steps = self._steps[self.overlap:]
                     self._hiddens = self._hiddens[self.overlap:]
+            elif done and len(self._recent) == self.T:
+                # terminal emission: the LAST T steps ending at the terminal
+                # step (R2D2/Player.py:37-47) — without it, episodes whose
+                # length is not a window boundary never produce a done=1
+                # sequence and the value function never sees termination
+                self._emit(list(self._recent), self._recent_hid[0], True)
+                self._steps, self._hiddens = [], []
             elif done:
                 self._steps, self._hiddens = [], []
             state = next_state
@@ -550,4 +563,6 @@ class R2D2Player:
                 self.transport.push_reward(self.idx, episode_reward, self.eps)
                 episode_reward = 0.0
                 self._zero_hidden()
+                self._recent.clear()
+                self._recent_hid.clear()
                 state = self.env.reset()

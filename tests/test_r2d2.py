@@ -150,3 +150,23 @@ def test_nstep_recurrent_targets_random_geometries():
             T, burn, n, rescale, (targets - expect).abs().max())
 
     check()
+
+
+def test_r2d2_terminal_sequence_emitted_off_boundary():
+    """An episode whose length is NOT a window boundary still produces a
+    done=1 sequence covering the last T steps (R2D2/Player.py:37-47
+    terminal emission)."""
+    from distributed_rl_amd.actors.env import SyntheticEnv
+
+    cfg = small_cfg()
+    T = cfg.fixed_trajectory  # 16; overlap 8 -> boundaries at 16, 24, 32...
+    pipe = InprocPipe()
+    env = SyntheticEnv(seed=3, episode_len=T + 5)  # 21: off-boundary
+    player = R2D2Player(cfg, idx=0, transport=pipe, env=env)
+    player.run(max_env_steps=T + 5)  # exactly one episode
+    cols, prio = pipe.drain()
+    done = cols["done"].ravel()
+    assert done[-1] == 1.0, done  # terminal window emitted
+    # it covers the LAST T steps: its first frame equals the boundary
+    # window's frame at offset 5 (windows: [0,16) done=0, [5,21) done=1)
+    assert np.array_equal(cols["states"][-1][0], cols["states"][0][5])
