@@ -624,6 +624,11 @@ void vtrace_bt(torch::Tensor mu, torch::Tensor t_logp, torch::Tensor rew,
                double gamma, double rho_bar, double c_bar, double lam,
                torch::Tensor vs, torch::Tensor pg_adv) {
   int B = (int)rew.size(0), T = (int)rew.size(1);
+  for (auto* t : {&mu, &t_logp, &values, &vs, &pg_adv}) {
+    DRL_CHECK_CONTIG((*t));
+    TORCH_CHECK(t->numel() == (int64_t)B * T, "vtrace_bt shape mismatch");
+  }
+  TORCH_CHECK(boot.numel() == B && not_done.numel() == B);
   hipLaunchKernelGGL(vtrace_bt_kernel, dim3(ceil_div(B, kBlock)), dim3(kBlock),
                      0, cur_stream(), mu.data_ptr<float>(),
                      t_logp.data_ptr<float>(), rew.data_ptr<float>(),
@@ -1074,6 +1079,9 @@ void impala_loss_fwd(torch::Tensor logpa, torch::Tensor adv,
                      double er, torch::Tensor loss_out, torch::Tensor obj_out,
                      torch::Tensor critic_out) {
   int64_t n = logpa.numel();
+  TORCH_CHECK(adv.numel() == n && v.numel() == n && vs.numel() == n,
+              "impala_loss_fwd length mismatch");
+  for (auto* t : {&logpa, &adv, &v, &vs}) DRL_CHECK_CONTIG((*t));
   hipLaunchKernelGGL(impala_loss_fwd_kernel, dim3(1), dim3(kBlock), 0,
                      cur_stream(), logpa.data_ptr<float>(),
                      adv.data_ptr<float>(), mean_H.data_ptr<float>(),
@@ -1087,6 +1095,10 @@ void impala_out_bwd(torch::Tensor pi_save, torch::Tensor H_save,
                     torch::Tensor vs, torch::Tensor gloss, int64_t B,
                     int64_t T, int64_t A, double er, torch::Tensor dout) {
   int64_t total = B * (T + 1) * (A + 1);
+  TORCH_CHECK(dout.numel() == total && dout.is_contiguous(),
+              "impala_out_bwd dout shape mismatch");
+  TORCH_CHECK(pi_save.numel() == B * T * A && v.numel() == B * T &&
+              vs.numel() == B * T && act.numel() == B * T);
   hipLaunchKernelGGL(impala_out_bwd_kernel, dim3(grid_for(total, 2)),
                      dim3(kBlock), 0, cur_stream(), pi_save.data_ptr<float>(),
                      H_save.data_ptr<float>(), act.data_ptr<int64_t>(),
