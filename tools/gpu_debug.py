@@ -87,6 +87,39 @@ def main():
     sync()
     print("  ok", flush=True)
 
+    print("phase: vtrace_bt", flush=True)
+    Bv2, Tv2 = 32, 20
+    mu = torch.rand(Bv2, Tv2, device=dev).clamp_min(1e-3)
+    tlp = -torch.rand(Bv2, Tv2, device=dev)
+    vs_bt = torch.empty(Bv2, Tv2, device=dev)
+    pg_bt = torch.empty_like(vs_bt)
+    ext.vtrace_bt(mu, tlp, torch.randn(Bv2, Tv2, device=dev),
+                  torch.randn(Bv2, Tv2, device=dev),
+                  torch.randn(Bv2, device=dev), torch.ones(Bv2, device=dev),
+                  0.99, 1.0, 1.0, 1.0, vs_bt, pg_bt)
+    sync()
+    print("  ok", flush=True)
+
+    print("phase: impala_loss_fwd + out_bwd", flush=True)
+    Ni, Ai = Bv2 * Tv2, 6
+    logpa = -torch.rand(Ni, device=dev)
+    adv = torch.randn(Ni, device=dev)
+    mh = torch.rand(1, device=dev)
+    l3 = torch.empty(1, device=dev)
+    o3 = torch.empty(1, device=dev)
+    c3 = torch.empty(1, device=dev)
+    ext.impala_loss_fwd(logpa, adv, mh, vs_bt.view(-1), pg_bt.view(-1),
+                        0.01, l3, o3, c3)
+    sync()
+    pi_s = torch.softmax(torch.randn(Ni, Ai, device=dev), -1).contiguous()
+    H_s = torch.rand(Ni, device=dev)
+    acts = torch.randint(0, Ai, (Ni,), dtype=torch.int64, device=dev)
+    dout = torch.empty(Bv2 * (Tv2 + 1), Ai + 1, device=dev)
+    ext.impala_out_bwd(pi_s, H_s, acts, adv, vs_bt.view(-1), pg_bt.view(-1),
+                       torch.ones(1, device=dev), Bv2, Tv2, Ai, 0.01, dout)
+    sync()
+    print("  ok loss=", l3.item(), flush=True)
+
     print("phase: rescale", flush=True)
     xv = torch.linspace(-5, 5, 1024, device=dev)
     yv = torch.empty_like(xv)
