@@ -111,3 +111,14 @@ def test_impala_resnet_model():
 
     blocks = [m for m in net.modules() if isinstance(m, ResidualBlock)]
     assert len(blocks) == 6  # 3 sections x 2 blocks
+
+
+def test_reference_name_aliases():
+    """The reference's camelCase surface (baseline.baseAgent / getOptim)
+    resolves to the same objects."""
+    from distributed_rl_amd.models import (
+        BaseAgent, baseAgent, get_optim, getOptim,
+    )
+
+    assert baseAgent is BaseAgent
+    assert getOptim is get_optim
