@@ -63,8 +63,18 @@ def build_apex(cfg, device, rank, world, args):
         replay_capacity=args.replay or cfg.replay_memory_len,
         replay_state_dtype=(torch.float16 if args.replay_dtype == "fp16"
                             else None),
+        transport=getattr(args, "_transport", None),
     )
     attach_reducer(learner)
+    if (learner.device.type == "cuda"
+            and getattr(learner, "_fast_fwd", None) is None
+            and os.environ.get("DRL_ALLOW_SLOW_DUELING") != "1"):
+        # the fused dueling forward is worth ~15%; a silent fallback would
+        # quietly degrade the headline number
+        raise RuntimeError(
+            "dueling fast-forward inactive for this cfg — set "
+            "DRL_ALLOW_SLOW_DUELING=1 to bench the slow path"
+        )
     cap = learner.replay.capacity
     n = min(cap, 100_000 if learner.device.type == "cuda" else 2_048)
     prefill_apex(learner, n, seed=1234 + rank)
@@ -79,6 +89,7 @@ def build_impala(cfg, device, rank, world, args):
     learner = ImpalaLearner(
         cfg, device=device, rank=rank, world_size=world, enable_tb=False,
         batch_size=batch, replay_capacity=args.replay or 2048,
+        transport=getattr(args, "_transport", None),
     )
     attach_reducer(learner)
     dev = learner.replay.device
@@ -110,6 +121,7 @@ def build_r2d2(cfg, device, rank, world, args):
     learner = R2D2Learner(
         cfg, device=device, rank=rank, world_size=world, enable_tb=False,
         batch_size=batch, replay_capacity=args.replay or 1024,
+        transport=getattr(args, "_transport", None),
     )
     attach_reducer(learner)
     dev = learner.replay.device
@@ -150,6 +162,16 @@ def main():
                     choices=["u8", "fp16"],
                     help="Ape-X replay frame storage (fp16 = BASELINE "
                          "config-5 compression option)")
+    ap.add_argument("--with-actors", type=int, default=0,
+                    help="spawn K synthetic-env CPU actor processes feeding "
+                         "the replay DURING the timed region (BASELINE "
+                         "config-2 whole-node mode); 0 = learner-only")
+    ap.add_argument("--ingest-every", type=int, default=8,
+                    help="learner.ingest() cadence inside the timed loop "
+                         "(matches the production run loop)")
+    ap.add_argument("--burn-in", type=int, default=None,
+                    help="R2D2 burn-in override (BASELINE config 4 = 40)")
+    ap.add_argument("--transport-dir", default=None)
     args = ap.parse_args()
 
     rank, local_rank, world = init_distributed()
@@ -164,8 +186,47 @@ def main():
     n_gpus = world if world > 1 else (args.gpus if has_cuda else 0)
 
     cfg = load_config(args.cfg)
+    if args.burn_in is not None:
+        cfg.raw["MEM"] = int(args.burn_in)
+    if args.with_actors > 0 and args.replay == 0:
+        # size the ring to the prefill so it is exactly full at capture time:
+        # mid-bench pushes then wrap (overwrite-oldest), keeping the baked
+        # n_valid of the captured sample kernel exact
+        args.replay = 100_000
+
+    # --- optional whole-node mode: shm transport + synthetic actor fleet ---
+    session = fleet = None
+    if args.with_actors > 0:
+        import tempfile
+
+        from distributed_rl_amd.actors.fleet import ActorFleet
+        from distributed_rl_amd.actors.transport import (
+            LearnerEndpoint, RecordCodec, TransportSession,
+        )
+        from distributed_rl_amd.algos import get_wire_schema
+
+        schema, with_prio = get_wire_schema(cfg)
+        codec = RecordCodec(schema, with_priority=with_prio)
+        tdir = args.transport_dir or tempfile.mkdtemp(prefix="drl_bench_")
+        if rank == 0:
+            session = TransportSession(tdir, codec, num_rings=args.with_actors,
+                                       ring_slots=256, create=True)
+        if world > 1:
+            torch.distributed.barrier()
+        if rank != 0:
+            session = TransportSession(tdir, codec, num_rings=args.with_actors,
+                                       create=False)
+        args._transport = LearnerEndpoint(session, rank=rank, world_size=world)
+        if rank == 0:
+            fleet = ActorFleet(args.cfg, args.with_actors, tdir,
+                               env_kind="synthetic", respawn_on_exit=False)
+
     builders = {"APE_X": build_apex, "IMPALA": build_impala, "R2D2": build_r2d2}
     learner, frames_per_step = builders[cfg.alg](cfg, device, rank, world, args)
+
+    if fleet is not None:
+        learner.publish_weights(include_target=True)
+        fleet.start()
 
     use_graph = args.graph == "on" or (
         args.graph == "auto" and has_cuda and hasattr(learner, "make_graphed_step")
@@ -177,8 +238,13 @@ def main():
         except Exception as e:
             print(f"# graph capture failed ({e}); falling back to eager",
                   file=sys.stderr)
-            stepper = learner.step
+            stepper = (learner.make_pipelined_step() if world > 1
+                       and hasattr(learner, "make_pipelined_step")
+                       else learner.step)
             use_graph = False
+    elif world > 1 and hasattr(learner, "make_pipelined_step"):
+        # eager pipelined step: same overlap ordering without graphs
+        stepper = learner.make_pipelined_step()
 
     def barrier_sync():
         if world > 1:
@@ -186,14 +252,24 @@ def main():
         if has_cuda:
             torch.cuda.synchronize()
 
-    for _ in range(args.warmup):
+    ingest_on = args.with_actors > 0
+    ingested = 0
+    for i in range(args.warmup):
+        if ingest_on and i % args.ingest_every == 0:
+            learner.ingest()
         stepper()
     barrier_sync()
     t0 = time.perf_counter()
-    for _ in range(args.steps):
+    for i in range(args.steps):
+        if ingest_on and i % args.ingest_every == 0:
+            ingested += learner.ingest()
         stepper()
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    if fleet is not None:
+        fleet.stop()
+    if session is not None:
+        session.close()
     if world > 1:
         t = torch.tensor([elapsed], device=device if has_cuda else "cpu")
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
@@ -205,7 +281,8 @@ def main():
     metric_name = {
         "APE_X": "learner frames/sec (whole node), Ape-X DQN Atari CNN",
         "IMPALA": "learner frames/sec (whole node), IMPALA V-trace Atari CNN",
-        "R2D2": "learner frames/sec (whole node), R2D2 LSTM seq80 burn-in20",
+        "R2D2": f"learner frames/sec (whole node), R2D2 LSTM "
+                f"seq{cfg.fixed_trajectory} burn-in{cfg.burn_in}",
     }[cfg.alg]
     if rank == 0:
         out = {
@@ -228,6 +305,8 @@ def main():
                 "seq_len": {"APE_X": 4, "IMPALA": cfg.unroll_step,
                             "R2D2": cfg.fixed_trajectory}[cfg.alg],
                 "parallelism": f"dp{max(n_gpus, 1)}",
+                "actors": args.with_actors,
+                "ingested_rows": ingested if args.with_actors else None,
                 "replay": (("gpu sum-tree PER" if has_cuda else "cpu PER")
                            + ("/fp16" if args.replay_dtype == "fp16" else ""))
                 if cfg.alg != "IMPALA" else "uniform fifo",

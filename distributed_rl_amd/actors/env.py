@@ -139,16 +139,18 @@ class AtariEnv:
         return self._stack.copy(), total_r, done, {"pseudo_done": pseudo_done, **info}
 
 
-def make_env(kind: str = "auto", **kw):
+def make_env(kind: str = "auto", reward_clip: bool = True, **kw):
+    """reward_clip carries cfg USE_REWARD_CLIP (cfg/ape_x.json:25) to the
+    real-env path; the synthetic env has no unclipped rewards to clip."""
     if kind == "synthetic":
         return SyntheticEnv(**kw)
     if kind == "atari":
-        return AtariEnv(**kw)
+        return AtariEnv(reward_clip=reward_clip, **kw)
     # auto: atari when gym importable, else synthetic
     try:
         import gym  # noqa: F401
 
-        return AtariEnv(**kw)
+        return AtariEnv(reward_clip=reward_clip, **kw)
     except Exception:
         kw.pop("game", None)
         return SyntheticEnv(**kw)

@@ -256,10 +256,18 @@ class ApexLearner(LearnerBase):
                     # fp16-compressed replay stores frames pre-scaled
                     for kk in ("state", "next_state"):
                         dev_cols[kk] = dev_cols[kk].to(self.state_dtype) / 255.0
-                self.replay.push(
-                    dev_cols,
-                    stage["__prio__"][:k].to(self.device, non_blocking=True),
-                )
+                prio_dev = stage["__prio__"][:k].to(self.device,
+                                                    non_blocking=True)
+            # H2D copies + relayout overlap queued compute, but the ring/
+            # sum-tree MUTATION must not interleave with the compute
+            # stream's own tree ops (sample/priority rebuild): two
+            # concurrent level-by-level rebuilds can leave the root
+            # transiently inconsistent with the leaves, and a sample in
+            # that window could descend into an unwritten row
+            self._ingest_stream.wait_stream(
+                torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(self._ingest_stream):
+                self.replay.push(dev_cols, prio_dev)
             self._stage_evt.record(self._ingest_stream)
             self._stage_busy = True
             # replay state is consumed by the compute stream; order it after
@@ -285,6 +293,21 @@ class ApexLearner(LearnerBase):
     # train
     # ------------------------------------------------------------------
     def train_step(self, data, idx, weights) -> Dict[str, torch.Tensor]:
+        stats, prio = self._fwd_bwd(data, weights)
+        if self.mp is not None:
+            self.mp.allreduce_grads()
+            self._optimize_mp()
+        else:
+            if self.reducer is not None:
+                self.reducer.all_reduce()
+            self.optim.step()
+        self.replay.update(idx, prio)
+        return stats
+
+    def _fwd_bwd(self, data, weights):
+        """Sample-batch forward + fused TD loss + backward; grads are left
+        in the flat buffers (collective-free -> hipGraph-capturable).
+        Returns (stats, prio)."""
         cuda = self.device.type == "cuda"
         if cuda:
             # NHWC frames straight into the fused conv stack (u8: dequant
@@ -326,17 +349,17 @@ class ApexLearner(LearnerBase):
         if self.mp is not None:
             self.mp.zero_grads()
             loss.backward()
-            self.mp.reduce_and_upcast()  # bf16 all-reduce at world>1
-            self.optim.step()
-            self.mp.sync_compute_params()
         else:
             self.optim.zero_grad(set_to_none=False)
             loss.backward()
-            if self.reducer is not None:
-                self.reducer.all_reduce()
-            self.optim.step()
-        self.replay.update(idx, prio)
-        return {"loss": loss.detach(), "value": qmean}
+        return {"loss": loss.detach(), "value": qmean}, prio
+
+    def _optimize_mp(self):
+        """Upcast + optimizer + param sync (post-collective stage;
+        hipGraph-capturable)."""
+        self.mp.upcast_grads()
+        self.optim.step()
+        self.mp.sync_compute_params()
 
     def sync_target(self):
         """Hard target sync (APE_X/Learner.py:204-208, tau=1)."""
@@ -398,42 +421,80 @@ class ApexLearner(LearnerBase):
 
             return stepper
 
-        # ---- split capture: g1 = sample..backward, eager all-reduce,
-        # ---- g2 = upcast + optimizer + param sync + priority update
+        # ---- overlapped pipeline (north-star C1): 4 graphs
+        #   g_s   sample batch t+1 into static buffers
+        #   g_fb  fwd + fused loss + backward (+ save idx for the update)
+        #   g_upd priority update of batch t
+        #   g_opt upcast + optimizer + param sync
+        # Step t runs: g_fb | [comm stream: RCCL all-reduce(flat_cgrad)]
+        #              in parallel with [compute stream: g_upd, g_s] | join
+        #              | g_opt — the collective is hidden behind the
+        #              priority update + NEXT replay sample. Op order on the
+        #              replay (update-then-sample) is identical to the
+        #              sequential path, so trajectories are bit-equal.
         mp = self.mp
-        g1 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g1):
-            data, s_idx, s_w = self.replay.sample(self.batch_size, self.beta)
-            s = data["state"].permute(0, 3, 1, 2)
-            sp = data["next_state"].permute(0, 3, 1, 2)
-            actions = data["action"].long()
-            q_s = self._online_q(s)
-            with torch.no_grad():
-                q_sp_on = self._online_q(sp)
-                q_sp_tg = self.target.forward([sp])[0]
-            loss, prio, qmean = ops.nstep_dqn_loss(
-                q_s, q_sp_on, q_sp_tg, actions,
-                data["reward"], data["done"], s_w, self.gamma, self.n_step,
-                self.alpha, with_value_stat=True,
-            )
-            mp.zero_grads()
-            loss.backward()
-            static_out = {"loss": loss.detach(), "value": qmean}
-        g2 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g2, pool=g1.pool()):
-            mp.flat_mgrad.copy_(mp.flat_cgrad)
-            mp.flat_mgrad.mul_(1.0 / mp.world)
-            self.optim.step()
-            mp.sync_compute_params()
-            self.replay.update(s_idx, prio)
-        self._graph = (g1, g2)
+        idx_save = torch.empty(self.batch_size, dtype=torch.int64,
+                               device=self.device)
+        g_s = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_s):
+            s_data, s_idx, s_w = self.replay.sample(self.batch_size, self.beta)
+        g_fb = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_fb, pool=g_s.pool()):
+            static_out, prio = self._fwd_bwd(s_data, s_w)
+            idx_save.copy_(s_idx)
+        g_upd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_upd, pool=g_s.pool()):
+            self.replay.update(idx_save, prio)
+        g_opt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_opt, pool=g_s.pool()):
+            self._optimize_mp()
+        self._graph = (g_s, g_fb, g_upd, g_opt)
+        comm = torch.cuda.Stream(self.device)
+        g_s.replay()  # prime the first batch
 
         def stepper():
-            g1.replay()
-            torch.distributed.all_reduce(mp.flat_cgrad, group=mp.pg)
-            g2.replay()
+            cur = torch.cuda.current_stream(self.device)
+            g_fb.replay()
+            comm.wait_stream(cur)  # grads ready on the compute stream
+            with torch.cuda.stream(comm):
+                mp.allreduce_grads()  # RCCL on the comm stream
+            g_upd.replay()  # overlap window: priority update of batch t
+            g_s.replay()  # ... and sample of batch t+1
+            cur.wait_stream(comm)
+            g_opt.replay()
             self._cadence()
             return static_out
+
+        return stepper
+
+    def make_pipelined_step(self):
+        """Eager pipelined step for world_size > 1 without graph capture
+        (CPU/gloo rehearsal of the same overlap ordering; also the GPU
+        fallback if capture fails). The async all-reduce runs while the
+        priority update + next sample execute; op order on the replay is
+        identical to the sequential stepper."""
+        pending = [self.replay.sample(self.batch_size, self.beta)]
+
+        def stepper():
+            data, idx, w = pending[0]
+            stats, prio = self._fwd_bwd(data, w)
+            if self.mp is not None:
+                works = self.mp.allreduce_grads_async()
+                self.replay.update(idx, prio)
+                pending[0] = self.replay.sample(self.batch_size, self.beta)
+                for wk in works:
+                    wk.wait()
+                self._optimize_mp()
+            else:
+                work = (self.reducer.all_reduce_async()
+                        if self.reducer is not None else None)
+                self.replay.update(idx, prio)
+                pending[0] = self.replay.sample(self.batch_size, self.beta)
+                if self.reducer is not None:
+                    self.reducer.finish(work)
+                self.optim.step()
+            self._cadence()
+            return stats
 
         return stepper
 
@@ -500,12 +561,16 @@ class ApexLearner(LearnerBase):
 
     def _log_block(self, stats):
         rewards = self.transport.drain_rewards() if self.transport else []
-        mean_r = float(np.mean(rewards)) if rewards else -21.0  # ref placeholder
+        # no fabricated placeholder when nothing was drained (the reference
+        # writes -21.0, a Pong-specific constant — APE_X/Learner.py:231);
+        # skip the scalar instead, like the IMPALA learner does
+        mean_r = float(np.mean(rewards)) if rewards else float("nan")
         timing = self.flush_timing()
         loss = float(stats["loss"])
         value = float(stats["value"])
         norm = float(self.model.calculateNorm())
-        self.log_scalar("Reward", mean_r)
+        if rewards:
+            self.log_scalar("Reward", mean_r)
         self.log_scalar("value", value)
         self.log_scalar("norm", norm)
         self.log_scalar("loss", loss)
@@ -614,7 +679,10 @@ class ApexPlayer:
         self.cfg = cfg
         self.idx = idx
         self.transport = transport
-        self.env = env or make_env(env_kind, seed=seed if seed is not None else idx)
+        self.env = env or make_env(
+            env_kind, seed=seed if seed is not None else idx,
+            reward_clip=cfg.use_reward_clip,
+        )
         self.device = torch.device(cfg.actor_device)
         # bounded staleness (SURVEY §5.3 gap): if the learner's published
         # `count` hasn't advanced for this many env steps, the actor blocks

@@ -254,21 +254,61 @@ class ImpalaLearner(LearnerBase):
 
             return stepper
 
-        g1 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g1):
-            data, _, _ = self.replay.sample(self.batch_size)
-            static_out = self._fwd_bwd(data)
-        g2 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g2, pool=g1.pool()):
+        # ---- overlapped pipeline (north-star C1): the RCCL all-reduce on a
+        # comm stream runs in parallel with the NEXT batch's FIFO sample on
+        # the compute stream (see ApexLearner.make_graphed_step).
+        mp = self.mp
+        g_s = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_s):
+            s_data, _, _ = self.replay.sample(self.batch_size)
+        g_fb = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_fb, pool=g_s.pool()):
+            static_out = self._fwd_bwd(s_data)
+        g_opt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_opt, pool=g_s.pool()):
             self._optimize_mp()
-        self._graph = (g1, g2)
+        self._graph = (g_s, g_fb, g_opt)
+        comm = torch.cuda.Stream(self.device)
+        g_s.replay()  # prime the first batch
 
         def stepper():
-            g1.replay()
-            self.mp.allreduce_grads()
-            g2.replay()
+            cur = torch.cuda.current_stream(self.device)
+            g_fb.replay()
+            comm.wait_stream(cur)
+            with torch.cuda.stream(comm):
+                mp.allreduce_grads()
+            g_s.replay()  # next sample overlaps the collective
+            cur.wait_stream(comm)
+            g_opt.replay()
             self._cadence()
             return static_out
+
+        return stepper
+
+    def make_pipelined_step(self):
+        """Eager pipelined step for world_size > 1 without graph capture
+        (CPU/gloo rehearsal of the overlap ordering; GPU fallback)."""
+        pending = [self.replay.sample(self.batch_size)]
+
+        def stepper():
+            data, _, _ = pending[0]
+            stats = self._fwd_bwd(data)
+            if self.mp is not None:
+                works = self.mp.allreduce_grads_async()
+                pending[0] = self.replay.sample(self.batch_size)
+                for wk in works:
+                    wk.wait()
+                self._optimize_mp()
+            else:
+                work = (self.reducer.all_reduce_async()
+                        if self.reducer is not None else None)
+                pending[0] = self.replay.sample(self.batch_size)
+                if self.reducer is not None:
+                    self.reducer.finish(work)
+                self.model.clippingNorm(40.0)
+                self.optim.step()
+            self._cadence()
+            return stats
 
         return stepper
 
@@ -290,10 +330,21 @@ class ImpalaLearner(LearnerBase):
                 raise TimeoutError("IMPALA replay warmup stalled")
             time.sleep(0.01)
         self.publish_weights()
+        stepper = None  # hipGraph-captured once the FIFO ring is full
         while self.step_count < max_steps:
             self.ingest()
+            if stepper is None and self.device.type == "cuda" \
+                    and len(self.replay) >= self.replay.capacity:
+                # n_valid is baked into the captured sample kernel; once the
+                # ring is full it stays at capacity, so capture is safe now
+                try:
+                    stepper = self.make_graphed_step()
+                except Exception as e:  # pragma: no cover
+                    print(f"[IMPALA] graph capture failed ({e}); staying eager",
+                          flush=True)
+                    stepper = self.step
             t1 = time.perf_counter()
-            stats = self.step()
+            stats = (stepper or self.step)()
             dt = time.perf_counter() - t1
             rewards = self.transport.drain_rewards() if self.transport else []
             mean_r = float(np.mean(rewards)) if rewards else None
@@ -365,7 +416,10 @@ class ImpalaPlayer:
         self.cfg = cfg
         self.idx = idx
         self.transport = transport
-        self.env = env or make_env(env_kind, seed=seed if seed is not None else idx)
+        self.env = env or make_env(
+            env_kind, seed=seed if seed is not None else idx,
+            reward_clip=cfg.use_reward_clip,
+        )
         self.model = BaseAgent(cfg.model_info).to(cfg.actor_device).eval()
         self.unroll = cfg.unroll_step
         self.action_n = cfg.action_size

@@ -330,22 +330,70 @@ class R2D2Learner(LearnerBase):
 
             return stepper
 
-        g1 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g1):
-            data, s_idx, s_w = self.replay.sample(self.batch_size, self.beta)
-            static_out, prio = self._fwd_bwd(data, s_w)
-        g2 = torch.cuda.CUDAGraph()
-        with torch.cuda.graph(g2, pool=g1.pool()):
+        # ---- overlapped pipeline (north-star C1): all-reduce on a comm
+        # stream runs in parallel with the priority update of batch t + the
+        # PER sample of batch t+1 (see ApexLearner.make_graphed_step).
+        mp = self.mp
+        idx_save = torch.empty(self.batch_size, dtype=torch.int64,
+                               device=self.device)
+        g_s = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_s):
+            s_data, s_idx, s_w = self.replay.sample(self.batch_size, self.beta)
+        g_fb = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_fb, pool=g_s.pool()):
+            static_out, prio = self._fwd_bwd(s_data, s_w)
+            idx_save.copy_(s_idx)
+        g_upd = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_upd, pool=g_s.pool()):
+            self.replay.update(idx_save, prio)
+        g_opt = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g_opt, pool=g_s.pool()):
             self._optimize_mp()
-            self.replay.update(s_idx, prio)
-        self._graph = (g1, g2)
+        self._graph = (g_s, g_fb, g_upd, g_opt)
+        comm = torch.cuda.Stream(self.device)
+        g_s.replay()  # prime the first batch
 
         def stepper():
-            g1.replay()
-            self.mp.allreduce_grads()
-            g2.replay()
+            cur = torch.cuda.current_stream(self.device)
+            g_fb.replay()
+            comm.wait_stream(cur)
+            with torch.cuda.stream(comm):
+                mp.allreduce_grads()
+            g_upd.replay()
+            g_s.replay()
+            cur.wait_stream(comm)
+            g_opt.replay()
             self._cadence()
             return static_out
+
+        return stepper
+
+    def make_pipelined_step(self):
+        """Eager pipelined step for world_size > 1 without graph capture
+        (CPU/gloo rehearsal of the overlap ordering; GPU fallback)."""
+        pending = [self.replay.sample(self.batch_size, self.beta)]
+
+        def stepper():
+            data, idx, w = pending[0]
+            stats, prio = self._fwd_bwd(data, w)
+            if self.mp is not None:
+                works = self.mp.allreduce_grads_async()
+                self.replay.update(idx, prio)
+                pending[0] = self.replay.sample(self.batch_size, self.beta)
+                for wk in works:
+                    wk.wait()
+                self._optimize_mp()
+            else:
+                work = (self.reducer.all_reduce_async()
+                        if self.reducer is not None else None)
+                self.replay.update(idx, prio)
+                pending[0] = self.replay.sample(self.batch_size, self.beta)
+                if self.reducer is not None:
+                    self.reducer.finish(work)
+                self.model.clippingNorm(40.0)
+                self.optim.step()
+            self._cadence()
+            return stats
 
         return stepper
 
@@ -375,13 +423,27 @@ class R2D2Learner(LearnerBase):
                 raise TimeoutError("R2D2 replay warmup stalled")
             time.sleep(0.01)
         self.publish_weights(include_target=True)
+        stepper = None  # hipGraph-captured once the replay ring is full
         while self.step_count < max_steps:
             self.ingest()
-            stats = self.step()
+            if stepper is None and self.device.type == "cuda" \
+                    and len(self.replay) >= self.replay.capacity:
+                # n_valid is baked into the captured sample kernel; once the
+                # ring is full it stays at capacity, so capture is safe now
+                try:
+                    stepper = self.make_graphed_step()
+                except Exception as e:  # pragma: no cover
+                    print(f"[R2D2] graph capture failed ({e}); staying eager",
+                          flush=True)
+                    stepper = self.step
+            stats = (stepper or self.step)()
             if self.step_count % self.LOG_EVERY == 0:
                 rewards = self.transport.drain_rewards() if self.transport else []
-                mean_r = float(np.mean(rewards)) if rewards else -21.0
-                self.log_scalar("Reward", mean_r)
+                # skip the Reward scalar when nothing was drained (no -21
+                # Pong placeholder — see ApexLearner._log_block)
+                mean_r = float(np.mean(rewards)) if rewards else float("nan")
+                if rewards:
+                    self.log_scalar("Reward", mean_r)
                 self.log_scalar("value", float(stats["value"]))
                 self.log_scalar("norm", float(self.model.calculateNorm()))
                 if self.rank == 0:
@@ -434,7 +496,10 @@ class R2D2Player:
         self.cfg = cfg
         self.idx = idx
         self.transport = transport
-        self.env = env or make_env(env_kind, seed=seed if seed is not None else idx)
+        self.env = env or make_env(
+            env_kind, seed=seed if seed is not None else idx,
+            reward_clip=cfg.use_reward_clip,
+        )
         self.model = BaseAgent(cfg.model_info).to(cfg.actor_device).eval()
         self.target = BaseAgent(cfg.model_info).to(cfg.actor_device).eval()
         n_actors = max(cfg.num_actors, 2)

@@ -111,10 +111,17 @@ def vtrace(
 ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Returns (vs, pg_advantage, rho_clipped).
 
-    Matches the reference's clipping order and terminal handling:
-    the bootstrap value is masked by not_done (IMPALA/Player.py:183-186 emits
-    not_done; Learner applies it to the bootstrap), and the scan runs reversed
-    in time with c_i = lam * min(c_bar, rho_i) (IMPALA/Learner.py:176-200).
+    Terminal handling matches the reference: the bootstrap value is masked
+    by not_done (IMPALA/Player.py:183-186 emits not_done; Learner applies it
+    to the bootstrap), and the scan runs reversed in time with
+    c_i = lam * min(c_bar, rho_i) (IMPALA/Learner.py:176-200).
+
+    DELIBERATE DIVERGENCE (also recorded in docs/PARITY.md): the reference
+    omits the clipped-rho factor on the delta of the LAST unrolled step
+    (IMPALA/Learner.py:176-185 applies rho only inside the loop body, not on
+    the step that seeds the reversed scan); we apply rho_clipped to EVERY
+    step's delta — the V-trace paper's form. Values differ from the
+    reference whenever pi != mu on the final step of an unroll.
     """
     T, B = rewards.shape
     rho = (target_log_prob - behavior_log_prob).exp()

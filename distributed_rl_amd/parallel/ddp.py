@@ -64,6 +64,18 @@ class FlatGradReducer:
         dist.all_reduce(self.flat, group=self.group)
         self.flat.mul_(self._inv_world)
 
+    def all_reduce_async(self):
+        """Launch the all-reduce without blocking; returns a work handle or
+        None. Caller must ``finish(work)`` before reading grads."""
+        if self.world <= 1:
+            return None
+        return dist.all_reduce(self.flat, group=self.group, async_op=True)
+
+    def finish(self, work) -> None:
+        if work is not None:
+            work.wait()
+            self.flat.mul_(self._inv_world)
+
     def zero_(self) -> None:
         self.flat.zero_()
 

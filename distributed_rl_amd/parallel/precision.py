@@ -170,6 +170,18 @@ class MixedPrecisionTrainer:
             for g in self.groups:
                 dist.all_reduce(g.flat_cgrad, group=self.pg)
 
+    def allreduce_grads_async(self):
+        """Launch the grad all-reduce without blocking the caller; returns
+        the work handles (call ``wait()`` on each before reading grads).
+        With the RCCL backend the collective is enqueued on the comm stream
+        and overlaps whatever the compute stream runs next (the north-star
+        overlap: next replay sample + priority update run during the
+        all-reduce — SURVEY.md §2.9 C1)."""
+        if self.world > 1:
+            return [dist.all_reduce(g.flat_cgrad, group=self.pg, async_op=True)
+                    for g in self.groups]
+        return []
+
     def upcast_grads(self):
         """compute-dtype grad flats -> fp32 master grads (+ 1/world)."""
         for g in self.groups:
