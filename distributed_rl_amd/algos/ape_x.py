@@ -504,8 +504,7 @@ class ApexLearner(LearnerBase):
     def publish_weights(self, include_target: bool = False):
         if self.transport is None or self.rank != 0:
             return
-        cpu_sd = {k: v.detach().to("cpu", torch.float32) for k, v in
-                  self.model.state_dict().items()}
+        cpu_sd = self.snapshot_state_dict()
         payload: Dict[str, Any] = {"count": self.step_count, "state_dict": cpu_sd}
         if include_target:
             payload["target_state_dict"] = {

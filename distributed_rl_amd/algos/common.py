@@ -111,6 +111,45 @@ class LearnerBase:
         self._t_block = time.perf_counter()
         return out
 
+    # -- weight snapshot ---------------------------------------------------
+    def snapshot_state_dict(self):
+        """CPU fp32 state_dict of the online model for transport publish.
+
+        On GPU with the mixed-precision trainer this is ONE pinned D2H copy
+        per dtype group off the flat fp32 master buffer + host-side clones,
+        instead of ~30 per-tensor synchronous ``.to("cpu")`` copies (the
+        reference pays the per-tensor cost every 50 steps,
+        APE_X/Learner.py:264-272; at 1k+ steps/s the publish cadence is on
+        the hot path)."""
+        mp = getattr(self, "mp", None)
+        if (mp is None or self.device.type != "cuda"
+                or len(list(self.model.buffers())) > 0):
+            return {k: v.detach().to("cpu", torch.float32)
+                    for k, v in self.model.state_dict().items()}
+        if getattr(self, "_pub_plan", None) is None:
+            named = dict(self.model.named_parameters())
+            ident = {id(p): n for n, p in named.items()}
+            pins, plan = [], []
+            for gi, g in enumerate(mp.groups):
+                pins.append(torch.empty(g.flat_mparam.numel(),
+                                        dtype=torch.float32).pin_memory())
+                off = 0
+                for cp, mparam in zip(g.c_params, g.m_params):
+                    plan.append((ident[id(mparam)], gi, off, mparam))
+                    off += cp.numel()
+            self._pub_plan = (pins, plan, torch.cuda.Event())
+        pins, plan, evt = self._pub_plan
+        for gi, g in enumerate(mp.groups):
+            pins[gi].copy_(g.flat_mparam, non_blocking=True)
+        evt.record()
+        evt.synchronize()
+        from ..parallel.precision import _view_like
+
+        return {
+            name: _view_like(pins[gi][off : off + p.numel()], p).clone()
+            for name, gi, off, p in plan
+        }
+
     # -- checkpoint / resume ---------------------------------------------
     def checkpoint_dir(self) -> str:
         d = self.cfg.weight_dir(self.run_root, self.run_name)

@@ -316,9 +316,8 @@ class ImpalaLearner(LearnerBase):
     def publish_weights(self, include_target: bool = False):
         if self.transport is None or self.rank != 0:
             return
-        cpu_sd = {k: v.detach().to("cpu", torch.float32)
-                  for k, v in self.model.state_dict().items()}
-        self.transport.publish({"count": self.step_count, "state_dict": cpu_sd})
+        self.transport.publish({"count": self.step_count,
+                                "state_dict": self.snapshot_state_dict()})
 
     # -- run loop (per-step TB scalars; SURVEY §5.5: 9 scalars per step) ---
     def run(self, max_steps: int = 1_000_000, warmup_items: Optional[int] = None):

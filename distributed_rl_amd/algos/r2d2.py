@@ -403,8 +403,7 @@ class R2D2Learner(LearnerBase):
             return
         payload = {
             "count": self.step_count,
-            "state_dict": {k: v.detach().to("cpu", torch.float32)
-                           for k, v in self.model.state_dict().items()},
+            "state_dict": self.snapshot_state_dict(),
         }
         if include_target:
             payload["target_state_dict"] = {
