@@ -227,6 +227,19 @@ def main():
     if fleet is not None:
         learner.publish_weights(include_target=True)
         fleet.start()
+        # wait for the fleet to warm up (64 actor processes each import
+        # torch before their first env step) so the timed region measures
+        # steady-state ingest, not an idle transport
+        warm_rows, t_warm = 0, time.perf_counter()
+        warm_target = max(4096, 8 * args.with_actors)
+        while warm_rows < warm_target and time.perf_counter() - t_warm < 180:
+            warm_rows += learner.ingest()
+            time.sleep(0.02)
+        dt_warm = time.perf_counter() - t_warm
+        print(f"# fleet warm: {warm_rows} rows in {dt_warm:.1f}s "
+              f"({warm_rows / max(dt_warm, 1e-9):.0f} rows/s), "
+              f"{fleet.alive_count()}/{args.with_actors} actors alive",
+              file=sys.stderr)
 
     use_graph = args.graph == "on" or (
         args.graph == "auto" and has_cuda and hasattr(learner, "make_graphed_step")

@@ -144,6 +144,11 @@ def make_env(kind: str = "auto", reward_clip: bool = True, **kw):
     real-env path; the synthetic env has no unclipped rewards to clip."""
     if kind == "synthetic":
         return SyntheticEnv(**kw)
+    if kind == "pong":
+        from .pong import PongEnv
+
+        kw.pop("game", None)
+        return PongEnv(**kw)
     if kind == "atari":
         return AtariEnv(reward_clip=reward_clip, **kw)
     # auto: atari when gym importable, else synthetic

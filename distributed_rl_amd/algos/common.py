@@ -146,7 +146,8 @@ class LearnerBase:
         from ..parallel.precision import _view_like
 
         return {
-            name: _view_like(pins[gi][off : off + p.numel()], p).clone()
+            name: _view_like(pins[gi][off : off + p.numel()], p)
+            .clone(memory_format=torch.contiguous_format)
             for name, gi, off, p in plan
         }
 
