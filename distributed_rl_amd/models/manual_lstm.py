@@ -61,19 +61,45 @@ class _ManualLSTMSeq(torch.autograd.Function):
         # barrier) next round.
         import os as _os
 
+        # K5 v2 (round 2, default): ONE kernel per timestep — bf16-MFMA hh
+        # GEMM fused with the cell. The recurrence state (c), gates and
+        # saves stay fp32; h additionally keeps a bf16 shadow that feeds the
+        # next step's MFMA. Halves the launch count of the hot loop.
+        use_v2 = (
+            x.is_cuda and H == 512 and B <= 32
+            and _os.environ.get("DRL_LSTM_BF16_HH", "1") == "1"
+            and hasattr(ext, "lstm_step_fwd_bf16")
+        )
+        h_bfs = None
         used = False
-        if _os.environ.get("DRL_LSTM_PERSISTENT", "0") == "1":
+        if use_v2:
+            w_bf = w_hh.detach().to(torch.bfloat16).contiguous()
+            h_bfs = torch.empty(T + 1, B, H, dtype=torch.bfloat16, device=dev)
+            h_bfs[0].copy_(h0)
+            used = True
+            for t in range(T):
+                if not ext.lstm_step_fwd_bf16(
+                        xp[t], h_bfs[t], cs[t], w_bf, hs[t + 1], cs[t + 1],
+                        h_bfs[t + 1], acts[t], tanhc[t]):
+                    used = False
+                    break
+        if not used and _os.environ.get("DRL_LSTM_PERSISTENT", "0") == "1":
             ctr = torch.zeros(1, dtype=torch.int32, device=dev)
             used = bool(ext.lstm_seq_persistent(
                 xp, hs, cs, w_hh.contiguous(), acts, tanhc, ctr))
         if not used:
+            h_bfs = None
             w_hh_t = w_hh.t()
             for t in range(T):
                 gates = xp[t].reshape(B, 4 * H)
                 gates.addmm_(hs[t], w_hh_t)
                 ext.lstm_cell_fwd(gates, cs[t], hs[t + 1], cs[t + 1], acts[t],
                                   tanhc[t])
-        ctx.save_for_backward(x, hs, cs, acts, tanhc, w_ih, w_hh)
+        ctx.v2 = h_bfs is not None
+        if ctx.v2:
+            ctx.save_for_backward(x, hs, cs, acts, tanhc, w_ih, w_hh, h_bfs)
+        else:
+            ctx.save_for_backward(x, hs, cs, acts, tanhc, w_ih, w_hh)
         ctx.dims = (T, B, IN, H)
         return hs[1:].clone(), hs[T].clone(), cs[T].clone()
 
@@ -82,28 +108,54 @@ class _ManualLSTMSeq(torch.autograd.Function):
         from ..ops import hip_ext
 
         ext = hip_ext()
-        x, hs, cs, acts, tanhc, w_ih, w_hh = ctx.saved_tensors
+        if ctx.v2:
+            x, hs, cs, acts, tanhc, w_ih, w_hh, h_bfs = ctx.saved_tensors
+        else:
+            x, hs, cs, acts, tanhc, w_ih, w_hh = ctx.saved_tensors
+            h_bfs = None
         T, B, IN, H = ctx.dims
         dev = x.device
         dgates_all = torch.empty(T, B, 4 * H, device=dev)
-        dh = gh_T.contiguous().clone()
-        dc = gc_T.contiguous().clone()
-        dc_next = torch.empty(B, H, device=dev)
         gout = gout.contiguous()
-        for t in range(T - 1, -1, -1):
-            # the per-step output grad is folded into the cell-bwd kernel
-            ext.lstm_cell_bwd(dh.contiguous(), gout[t], dc, acts[t], tanhc[t],
-                              cs[t], dgates_all[t], dc_next)
-            dc, dc_next = dc_next, dc
-            dh = dgates_all[t].mm(w_hh)
-        dg_flat = dgates_all.reshape(T * B, 4 * H)
-        dw_hh = dg_flat.t().mm(hs[:-1].reshape(T * B, H))
+        if ctx.v2:
+            # one fused kernel per step: dh GEMM (bf16 MFMA) + cell-bwd
+            dg_bf_all = torch.empty(T, B, 4 * H, dtype=torch.bfloat16,
+                                    device=dev)
+            w_t_bf = w_hh.t().contiguous().to(torch.bfloat16)
+            dh_init = gh_T.contiguous()
+            dc = gc_T.contiguous().clone()
+            dc_next = torch.empty(B, H, device=dev)
+            empty_bf = torch.empty(0, dtype=torch.bfloat16, device=dev)
+            for t in range(T - 1, -1, -1):
+                dg_prev = dg_bf_all[t + 1] if t < T - 1 else empty_bf
+                ext.lstm_step_bwd_bf16(dg_prev, dh_init, gout[t], dc,
+                                       w_t_bf, acts[t], tanhc[t], cs[t],
+                                       dgates_all[t], dg_bf_all[t], dc_next)
+                dc, dc_next = dc_next, dc
+            dh = dgates_all[0].mm(w_hh)  # grad wrt h0 (fp32, once per seq)
+            dg_flat = dgates_all.reshape(T * B, 4 * H)
+            dg_bf = dg_bf_all.reshape(T * B, 4 * H)
+            dw_hh = dg_bf.t().mm(h_bfs[:-1].reshape(T * B, H)).float()
+        else:
+            dh = gh_T.contiguous().clone()
+            dc = gc_T.contiguous().clone()
+            dc_next = torch.empty(B, H, device=dev)
+            for t in range(T - 1, -1, -1):
+                # the per-step output grad is folded into the cell-bwd kernel
+                ext.lstm_cell_bwd(dh.contiguous(), gout[t], dc, acts[t],
+                                  tanhc[t], cs[t], dgates_all[t], dc_next)
+                dc, dc_next = dc_next, dc
+                dh = dgates_all[t].mm(w_hh)
+            dg_flat = dgates_all.reshape(T * B, 4 * H)
+            dg_bf = None
+            dw_hh = dg_flat.t().mm(hs[:-1].reshape(T * B, H))
         db = dg_flat.sum(0)
         if x.dtype == torch.bfloat16:
             # bf16 weight/input grads (fp32-accumulated inside the GEMM,
             # rounded to bf16 on output — same precision class as the conv
             # trunk's bf16 grads that feed the same fp32 master upcast)
-            dg_bf = dg_flat.to(torch.bfloat16)
+            if dg_bf is None:
+                dg_bf = dg_flat.to(torch.bfloat16)
             dw_ih = dg_bf.t().mm(x.reshape(T * B, IN)).float()
             dx = dg_bf.mm(w_ih.to(torch.bfloat16)).view(T, B, IN)
         else:

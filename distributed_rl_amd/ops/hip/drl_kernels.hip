@@ -687,6 +687,17 @@ bool lstm_step_fused(torch::Tensor xp_t, torch::Tensor h_in, torch::Tensor c_in,
 bool lstm_seq_persistent(torch::Tensor xp, torch::Tensor hs, torch::Tensor cs,
                          torch::Tensor w_hh, torch::Tensor acts,
                          torch::Tensor tanhc, torch::Tensor ctr);
+bool lstm_step_fwd_bf16(torch::Tensor xp_t, torch::Tensor h_bf,
+                        torch::Tensor c_in, torch::Tensor w_bf,
+                        torch::Tensor h_out, torch::Tensor c_out,
+                        torch::Tensor h_bf_out, torch::Tensor acts_t,
+                        torch::Tensor tanhc_t);
+bool lstm_step_bwd_bf16(torch::Tensor dg_prev, torch::Tensor dh_init,
+                        torch::Tensor gout_t, torch::Tensor dc_in,
+                        torch::Tensor w_t_bf, torch::Tensor acts_t,
+                        torch::Tensor tanhc_t, torch::Tensor c_prev,
+                        torch::Tensor dgates_t, torch::Tensor dg_bf_t,
+                        torch::Tensor dc_out);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -724,6 +735,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_persistent", &lstm_seq_persistent,
         "whole-sequence persistent LSTM: grid-resident, agent-scope step "
         "barriers (K5)");
+  m.def("lstm_step_fwd_bf16", &lstm_step_fwd_bf16,
+        "one-kernel LSTM timestep fwd: bf16-MFMA hh GEMM + cell (K5 v2)");
+  m.def("lstm_step_bwd_bf16", &lstm_step_bwd_bf16,
+        "one-kernel LSTM timestep bwd: bf16-MFMA dh GEMM + cell-bwd (K5 v2)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -1449,5 +1464,234 @@ bool lstm_seq_persistent(torch::Tensor xp, torch::Tensor hs, torch::Tensor cs,
                      w_hh.data_ptr<float>(), acts.data_ptr<float>(),
                      tanhc.data_ptr<float>(),
                      (unsigned int*)ctr.data_ptr(), B, T);
+  return true;
+}
+
+// ===========================================================================
+// K5 v2 (round 2): ONE kernel per LSTM timestep, forward AND backward, with
+// the hh GEMM on bf16 MFMA (v_mfma_f32_16x16x32_bf16, fp32 accumulate).
+//
+// Why: the round-1 per-step path is 2 kernels/step each direction
+// (hipBLASLt addmm + fused cell = 12.8 us fwd, 18.1 us bwd measured) and is
+// kernel-launch-floor-bound; the round-1 one-kernel fp32-MFMA step lost
+// because f32 MFMA issues at 1/16 the bf16 rate (256 MFMA x 32 cyc on a
+// 128-wave grid). bf16 operands cut the MFMA count 8x (K-step 32 vs 4), so
+// one kernel per step runs at the launch floor:
+//   fwd:  gates = xp_t + h_t @ W_hh^T  ->  cell  (h kept as a bf16 shadow)
+//   bwd:  dh_t = dgates_{t+1} @ W_hh   ->  cell-bwd -> dgates_t (fp32+bf16)
+// The backward fusion is only possible because the K=2048 GEMM fits one
+// block's wave budget in bf16. Gates/cell state/saves stay fp32; only the
+// GEMM operands (h, dgates, W) are bf16-rounded — the same precision class
+// as the bf16 conv trunk feeding the same fp32 master (R2D2/Learner.py:94-121
+// is the replaced op chain).
+//
+// MFMA fragment maps (guide §3, same as conv_mfma.hip):
+//   A: lane l holds A[row=l&15][k=(l>>4)*8+j], j=0..7
+//   B: lane l holds B[k=(l>>4)*8+j][col=l&15]
+//   C/D: lane l holds C[row=(l>>4)*4+r][col=l&15], r=0..3
+// ===========================================================================
+namespace {
+
+using bf16x8_k5 = __attribute__((ext_vector_type(8))) __bf16;
+
+__device__ __forceinline__ float sigm_f(float x) {
+  return 1.0f / (1.0f + __expf(-x));
+}
+
+// fwd: grid = H/16 blocks, 4 waves = 4 gates (torch order i,f,g,o), each
+// wave computes a (32 x 16) C tile over K=H.
+template <int H>
+__global__ __launch_bounds__(256) void lstm_step_fwd_bf16_kernel(
+    const float* __restrict__ xp_t,    // (B, 4H)
+    const __bf16* __restrict__ h_bf,   // (B, H) bf16 shadow of h_t
+    const float* __restrict__ c_in,    // (B, H)
+    const __bf16* __restrict__ w_bf,   // (4H, H) bf16 copy of W_hh
+    float* __restrict__ h_out,         // (B, H)
+    float* __restrict__ c_out,         // (B, H)
+    __bf16* __restrict__ h_bf_out,     // (B, H)
+    float* __restrict__ acts,          // (B, 4H)
+    float* __restrict__ tanhc,         // (B, H)
+    int B) {
+  __shared__ float gbuf[4][32][16];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;              // gate index
+  const int s16 = blockIdx.x * 16;        // h-column slice base
+  const int col = wave * H + s16 + (lane & 15);
+  const __bf16* wrow = w_bf + (int64_t)col * H;
+  const int kbase = (lane >> 4) * 8;
+  const int r0 = lane & 15, r1 = 16 + (lane & 15);
+  const int r0c = r0 < B ? r0 : 0, r1c = r1 < B ? r1 : 0;
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 4
+  for (int kb = 0; kb < H; kb += 32) {
+    bf16x8_k5 bfrag = *reinterpret_cast<const bf16x8_k5*>(wrow + kb + kbase);
+    bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
+        h_bf + (int64_t)r0c * H + kb + kbase);
+    bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
+        h_bf + (int64_t)r1c * H + kb + kbase);
+    acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
+    acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
+  }
+  const int crow = (lane >> 4) * 4;
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row0 = crow + r, row1 = 16 + crow + r;
+    int row0c = row0 < B ? row0 : 0, row1c = row1 < B ? row1 : 0;
+    gbuf[wave][row0][lane & 15] =
+        acc0[r] + xp_t[(int64_t)row0c * 4 * H + col];
+    gbuf[wave][row1][lane & 15] =
+        acc1[r] + xp_t[(int64_t)row1c * 4 * H + col];
+  }
+  __syncthreads();
+  for (int e = tid; e < 32 * 16; e += 256) {
+    int row = e >> 4;
+    int hc = e & 15;
+    if (row >= B) continue;
+    float i_ = sigm_f(gbuf[0][row][hc]);
+    float f_ = sigm_f(gbuf[1][row][hc]);
+    float g_ = tanhf(gbuf[2][row][hc]);
+    float o_ = sigm_f(gbuf[3][row][hc]);
+    int64_t hidx = (int64_t)row * H + s16 + hc;
+    float c = f_ * c_in[hidx] + i_ * g_;
+    float tc = tanhf(c);
+    float h = o_ * tc;
+    h_out[hidx] = h;
+    c_out[hidx] = c;
+    h_bf_out[hidx] = (__bf16)h;
+    tanhc[hidx] = tc;
+    float* a4 = acts + (int64_t)row * 4 * H + s16 + hc;
+    a4[0] = i_;
+    a4[H] = f_;
+    a4[2 * H] = g_;
+    a4[3 * H] = o_;
+  }
+}
+
+// bwd: grid = H/16 blocks; the 4 waves split K=4H four ways for the
+// dh = dgates_{t+1} @ W_hh GEMM (partials reduced through LDS), then the
+// block finishes cell-bwd for its 16 h-columns. At t=T-1 (dg_prev null)
+// dh comes from dh_init instead.
+template <int H>
+__global__ __launch_bounds__(256) void lstm_step_bwd_bf16_kernel(
+    const __bf16* __restrict__ dg_prev,  // (B, 4H) bf16 or null
+    const float* __restrict__ dh_init,   // (B, H), used when dg_prev null
+    const float* __restrict__ gout_t,    // (B, H) or null
+    const float* __restrict__ dc_in,     // (B, H)
+    const __bf16* __restrict__ w_t_bf,   // (H, 4H) bf16 = W_hh^T
+    const float* __restrict__ acts,      // (B, 4H) saved at t
+    const float* __restrict__ tanhc,     // (B, H)
+    const float* __restrict__ c_prev,    // (B, H)
+    float* __restrict__ dgates,          // (B, 4H) fp32 out
+    __bf16* __restrict__ dgates_bf,      // (B, 4H) bf16 out
+    float* __restrict__ dc_out,          // (B, H)
+    int B) {
+  __shared__ float partial[4][32][16];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int n0 = blockIdx.x * 16;  // h-column slice base
+  if (dg_prev != nullptr) {
+    const int kbase = (lane >> 4) * 8;
+    const int r0 = lane & 15, r1 = 16 + (lane & 15);
+    const int r0c = r0 < B ? r0 : 0, r1c = r1 < B ? r1 : 0;
+    const __bf16* wcol = w_t_bf + (int64_t)(n0 + (lane & 15)) * 4 * H;
+    f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+    const int k0 = wave * H;  // this wave's K chunk: [k0, k0+H)
+#pragma unroll 4
+    for (int kb = 0; kb < H; kb += 32) {
+      bf16x8_k5 bfrag =
+          *reinterpret_cast<const bf16x8_k5*>(wcol + k0 + kb + kbase);
+      bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
+          dg_prev + (int64_t)r0c * 4 * H + k0 + kb + kbase);
+      bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
+          dg_prev + (int64_t)r1c * 4 * H + k0 + kb + kbase);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
+    }
+    const int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      partial[wave][crow + r][lane & 15] = acc0[r];
+      partial[wave][16 + crow + r][lane & 15] = acc1[r];
+    }
+  }
+  __syncthreads();
+  for (int e = tid; e < 32 * 16; e += 256) {
+    int row = e >> 4;
+    int hc = e & 15;
+    if (row >= B) continue;
+    int64_t hidx = (int64_t)row * H + n0 + hc;
+    float dhv;
+    if (dg_prev != nullptr) {
+      dhv = partial[0][row][hc] + partial[1][row][hc] + partial[2][row][hc] +
+            partial[3][row][hc];
+    } else {
+      dhv = dh_init[hidx];
+    }
+    if (gout_t != nullptr) dhv += gout_t[hidx];
+    const float* a4 = acts + (int64_t)row * 4 * H + n0 + hc;
+    float i = a4[0], f = a4[H], g = a4[2 * H], o = a4[3 * H];
+    float tc = tanhc[hidx];
+    float do_ = dhv * tc;
+    float dct = dc_in[hidx] + dhv * o * (1.0f - tc * tc);
+    float di = dct * g;
+    float df = dct * c_prev[hidx];
+    float dg = dct * i;
+    dc_out[hidx] = dct * f;
+    float v0 = di * i * (1.0f - i);
+    float v1 = df * f * (1.0f - f);
+    float v2 = dg * (1.0f - g * g);
+    float v3 = do_ * o * (1.0f - o);
+    float* d4 = dgates + (int64_t)row * 4 * H + n0 + hc;
+    d4[0] = v0;
+    d4[H] = v1;
+    d4[2 * H] = v2;
+    d4[3 * H] = v3;
+    __bf16* b4 = dgates_bf + (int64_t)row * 4 * H + n0 + hc;
+    b4[0] = (__bf16)v0;
+    b4[H] = (__bf16)v1;
+    b4[2 * H] = (__bf16)v2;
+    b4[3 * H] = (__bf16)v3;
+  }
+}
+}  // namespace
+
+bool lstm_step_fwd_bf16(torch::Tensor xp_t, torch::Tensor h_bf,
+                        torch::Tensor c_in, torch::Tensor w_bf,
+                        torch::Tensor h_out, torch::Tensor c_out,
+                        torch::Tensor h_bf_out, torch::Tensor acts_t,
+                        torch::Tensor tanhc_t) {
+  int B = (int)c_in.size(0), H = (int)c_in.size(1);
+  if (B > kLstmMaxB || H != 512) return false;
+  constexpr int HH = 512;
+  hipLaunchKernelGGL(lstm_step_fwd_bf16_kernel<HH>, dim3(HH / 16), dim3(256),
+                     0, cur_stream(), xp_t.data_ptr<float>(),
+                     (const __bf16*)h_bf.data_ptr(), c_in.data_ptr<float>(),
+                     (const __bf16*)w_bf.data_ptr(), h_out.data_ptr<float>(),
+                     c_out.data_ptr<float>(), (__bf16*)h_bf_out.data_ptr(),
+                     acts_t.data_ptr<float>(), tanhc_t.data_ptr<float>(), B);
+  return true;
+}
+
+bool lstm_step_bwd_bf16(torch::Tensor dg_prev, torch::Tensor dh_init,
+                        torch::Tensor gout_t, torch::Tensor dc_in,
+                        torch::Tensor w_t_bf, torch::Tensor acts_t,
+                        torch::Tensor tanhc_t, torch::Tensor c_prev,
+                        torch::Tensor dgates_t, torch::Tensor dg_bf_t,
+                        torch::Tensor dc_out) {
+  int B = (int)dc_in.size(0), H = (int)dc_in.size(1);
+  if (B > kLstmMaxB || H != 512) return false;
+  constexpr int HH = 512;
+  const __bf16* dgp =
+      dg_prev.numel() ? (const __bf16*)dg_prev.data_ptr() : nullptr;
+  const float* gp = gout_t.numel() ? gout_t.data_ptr<float>() : nullptr;
+  hipLaunchKernelGGL(lstm_step_bwd_bf16_kernel<HH>, dim3(HH / 16), dim3(256),
+                     0, cur_stream(), dgp, dh_init.data_ptr<float>(), gp,
+                     dc_in.data_ptr<float>(),
+                     (const __bf16*)w_t_bf.data_ptr(),
+                     acts_t.data_ptr<float>(), tanhc_t.data_ptr<float>(),
+                     c_prev.data_ptr<float>(), dgates_t.data_ptr<float>(),
+                     (__bf16*)dg_bf_t.data_ptr(), dc_out.data_ptr<float>(), B);
   return true;
 }

@@ -183,3 +183,106 @@ def test_r2d2_gpu_graphed_step():
     torch.cuda.synchronize()
     assert np.isfinite(float(out["loss"]))
     assert learner.step_count == 3
+
+
+def test_manual_lstm_bf16_hh_close_to_fp32():
+    """K5 v2 (one kernel per step, bf16-MFMA hh GEMM): only the GEMM
+    operands (h, dgates, W) are bf16-rounded; gates/cell/saves stay fp32.
+    Requires H=512 (the R2D2 production width). Values and grads must track
+    the fp32 2-kernel path within bf16 rounding."""
+    import os
+
+    from distributed_rl_amd.models.manual_lstm import manual_lstm_seq
+
+    torch.manual_seed(17)
+    T, B, IN, H = 16, 8, 96, 512
+    lstm = torch.nn.LSTM(IN, H).to(DEV)
+    h0 = torch.randn(1, B, H, device=DEV) * 0.5
+    c0 = torch.randn(1, B, H, device=DEV) * 0.5
+    x_a = torch.randn(T, B, IN, device=DEV, requires_grad=True)
+    x_b = x_a.detach().clone().requires_grad_(True)
+    g = torch.randn(T, B, H, device=DEV)
+
+    os.environ["DRL_LSTM_BF16_HH"] = "0"
+    try:
+        out_ref, (hT_ref, cT_ref) = manual_lstm_seq(x_a, (h0, c0), lstm)
+        out_ref.backward(g)
+        gw_hh = lstm.weight_hh_l0.grad.clone()
+        gb = lstm.bias_ih_l0.grad.clone()
+        gx = x_a.grad.clone()
+        for p in lstm.parameters():
+            p.grad = None
+    finally:
+        os.environ["DRL_LSTM_BF16_HH"] = "1"
+
+    out, (hT, cT) = manual_lstm_seq(x_b, (h0, c0), lstm)
+    assert out.dtype == torch.float32
+    out.backward(g)
+
+    assert torch.allclose(out, out_ref, atol=0.06), (out - out_ref).abs().max()
+    assert torch.allclose(hT, hT_ref, atol=0.06)
+    assert torch.allclose(cT, cT_ref, atol=0.10)
+    assert torch.allclose(x_b.grad, gx, atol=0.10, rtol=0.05), \
+        (x_b.grad - gx).abs().max()
+    rel_w = (lstm.weight_hh_l0.grad - gw_hh).abs().max() / gw_hh.abs().max()
+    assert rel_w < 0.06, rel_w
+    rel_b = (lstm.bias_ih_l0.grad - gb).abs().max() / gb.abs().max()
+    assert rel_b < 0.06, rel_b
+
+
+def test_lstm_step_bf16_kernels_match_composition():
+    """Exact-ish oracle for the fused kernels: one step's fwd/bwd vs the
+    same math composed from torch ops with identically bf16-rounded
+    operands (differences only from MFMA vs torch accumulation order)."""
+    from distributed_rl_amd.ops import hip_ext
+
+    ext = hip_ext(required=True)
+    torch.manual_seed(23)
+    B, H = 32, 512
+    xp = torch.randn(B, 4 * H, device=DEV)
+    h = torch.randn(B, H, device=DEV)
+    c = torch.randn(B, H, device=DEV)
+    w_hh = torch.randn(4 * H, H, device=DEV) * 0.05
+    h_bf = h.to(torch.bfloat16)
+    w_bf = w_hh.to(torch.bfloat16)
+
+    h_out = torch.empty(B, H, device=DEV)
+    c_out = torch.empty(B, H, device=DEV)
+    h_bf_out = torch.empty(B, H, dtype=torch.bfloat16, device=DEV)
+    acts = torch.empty(B, 4 * H, device=DEV)
+    tanhc = torch.empty(B, H, device=DEV)
+    assert ext.lstm_step_fwd_bf16(xp, h_bf, c, w_bf, h_out, c_out, h_bf_out,
+                                  acts, tanhc)
+    # composition oracle with the same bf16-rounded operands
+    gates = xp + h_bf.float().mm(w_bf.float().t())
+    i = torch.sigmoid(gates[:, :H])
+    f = torch.sigmoid(gates[:, H:2 * H])
+    gg = torch.tanh(gates[:, 2 * H:3 * H])
+    o = torch.sigmoid(gates[:, 3 * H:])
+    c_ref = f * c + i * gg
+    h_ref = o * torch.tanh(c_ref)
+    assert torch.allclose(c_out, c_ref, atol=2e-3), (c_out - c_ref).abs().max()
+    assert torch.allclose(h_out, h_ref, atol=2e-3)
+    assert torch.allclose(h_bf_out.float(), h_ref, atol=8e-3)
+
+    # backward kernel vs composition
+    dg_prev = torch.randn(B, 4 * H, device=DEV).to(torch.bfloat16)
+    gout = torch.randn(B, H, device=DEV)
+    dc_in = torch.randn(B, H, device=DEV)
+    w_t_bf = w_hh.t().contiguous().to(torch.bfloat16)
+    dgates = torch.empty(B, 4 * H, device=DEV)
+    dg_bf = torch.empty(B, 4 * H, dtype=torch.bfloat16, device=DEV)
+    dc_out = torch.empty(B, H, device=DEV)
+    assert ext.lstm_step_bwd_bf16(dg_prev, gout, gout, dc_in, w_t_bf, acts,
+                                  tanhc, c, dgates, dg_bf, dc_out)
+    dh_ref = dg_prev.float().mm(w_bf.float()) + gout
+    do_ = dh_ref * tanhc
+    dct = dc_in + dh_ref * o * (1 - tanhc * tanhc)
+    d_ref = torch.cat([
+        dct * gg * i * (1 - i),
+        dct * c * f * (1 - f),
+        dct * i * (1 - gg * gg),
+        do_ * o * (1 - o),
+    ], dim=1)
+    assert torch.allclose(dgates, d_ref, atol=2e-3), (dgates - d_ref).abs().max()
+    assert torch.allclose(dc_out, dct * f, atol=2e-3)
