@@ -166,9 +166,6 @@ def main():
                     help="spawn K synthetic-env CPU actor processes feeding "
                          "the replay DURING the timed region (BASELINE "
                          "config-2 whole-node mode); 0 = learner-only")
-    ap.add_argument("--ingest-every", type=int, default=8,
-                    help="learner.ingest() cadence inside the timed loop "
-                         "(matches the production run loop)")
     ap.add_argument("--burn-in", type=int, default=None,
                     help="R2D2 burn-in override (BASELINE config 4 = 40)")
     ap.add_argument("--transport-dir", default=None)
@@ -227,19 +224,22 @@ def main():
     if fleet is not None:
         learner.publish_weights(include_target=True)
         fleet.start()
-        # wait for the fleet to warm up (64 actor processes each import
+        # ingest runs on its own daemon thread (the production run-loop
+        # shape); wait for the fleet to warm up (actor processes import
         # torch before their first env step) so the timed region measures
         # steady-state ingest, not an idle transport
-        warm_rows, t_warm = 0, time.perf_counter()
-        warm_target = max(4096, 8 * args.with_actors)
-        while warm_rows < warm_target and time.perf_counter() - t_warm < 180:
-            warm_rows += learner.ingest()
-            time.sleep(0.02)
+        learner.start_ingest_thread()
+        t_warm = time.perf_counter()
+        warm_target = max(8192, 16 * args.with_actors)
+        while learner.ingested_total < warm_target \
+                and time.perf_counter() - t_warm < 180:
+            time.sleep(0.05)
         dt_warm = time.perf_counter() - t_warm
-        print(f"# fleet warm: {warm_rows} rows in {dt_warm:.1f}s "
-              f"({warm_rows / max(dt_warm, 1e-9):.0f} rows/s), "
+        print(f"# fleet warm: {learner.ingested_total} rows in {dt_warm:.1f}s "
+              f"({learner.ingested_total / max(dt_warm, 1e-9):.0f} rows/s), "
               f"{fleet.alive_count()}/{args.with_actors} actors alive",
               file=sys.stderr)
+        learner.stop_ingest_thread()  # capture below needs quiet streams
 
     use_graph = args.graph == "on" or (
         args.graph == "auto" and has_cuda and hasattr(learner, "make_graphed_step")
@@ -265,20 +265,20 @@ def main():
         if has_cuda:
             torch.cuda.synchronize()
 
-    ingest_on = args.with_actors > 0
-    ingested = 0
+    if args.with_actors > 0:
+        learner.start_ingest_thread()  # concurrent with the timed region
     for i in range(args.warmup):
-        if ingest_on and i % args.ingest_every == 0:
-            learner.ingest()
         stepper()
     barrier_sync()
+    ingested0 = learner.ingested_total if args.with_actors else 0
     t0 = time.perf_counter()
     for i in range(args.steps):
-        if ingest_on and i % args.ingest_every == 0:
-            ingested += learner.ingest()
         stepper()
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    ingested = (learner.ingested_total - ingested0) if args.with_actors else 0
+    if args.with_actors > 0:
+        learner.stop_ingest_thread()
     if fleet is not None:
         fleet.stop()
     if session is not None:

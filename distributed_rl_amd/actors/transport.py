@@ -139,6 +139,25 @@ class SpscRing:
         return n
 
     # consumer -----------------------------------------------------------
+    def peek_records(self, max_n: int = 1 << 30):
+        """Zero-copy consume: returns (views, n) where views are 1-2 uint8
+        slices of the ring storage covering n records. SPSC contract: the
+        consumer owns [tail, tail+n) until it calls advance(n) — copy out of
+        the views FIRST, then advance."""
+        head, tail = self.head, self.tail
+        n = min(head - tail, max_n)
+        if n <= 0:
+            return None, 0
+        start = tail % self.slots
+        first = min(n, self.slots - start)
+        views = [self._data[start : start + first]]
+        if n > first:
+            views.append(self._data[: n - first])
+        return views, n
+
+    def advance(self, n: int) -> None:
+        self._set_u64(8, self.tail + n)
+
     def pop_records(self, max_n: int = 1 << 30) -> Optional[np.ndarray]:
         head, tail = self.head, self.tail
         n = min(head - tail, max_n)
@@ -399,6 +418,21 @@ class LearnerEndpoint:
 
     def drain(self):
         return self.session.drain(ring_ids=self.ring_ids)
+
+    @property
+    def record_dtype(self) -> np.dtype:
+        return self.session.codec.np_dtype
+
+    def drain_views(self):
+        """Zero-copy drain: yields (views, n, advance) per non-empty ring.
+        The caller must copy out of the shm views and THEN call advance(n)
+        (the single host copy goes shm -> pinned staging directly; see
+        ApexLearner.ingest)."""
+        for i in self.ring_ids:
+            ring = self.session.ring(i)
+            views, n = ring.peek_records()
+            if n:
+                yield views, n, ring.advance
 
     def publish(self, obj):
         self.session.weight_bus.publish(obj)

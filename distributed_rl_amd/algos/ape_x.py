@@ -157,6 +157,7 @@ class ApexLearner(LearnerBase):
 
     def _build_fast_forward(self, order: List[str]):
         self._fast_fwd = None
+        self._fast_heads = None
         if not order:
             return
         import torch.nn.functional as F
@@ -189,6 +190,27 @@ class ApexLearner(LearnerBase):
 
         self._fast_fwd = fast_fwd
 
+        def fast_heads(x):
+            feat = cnn(x)
+            h = F.relu(F.linear(feat.to(w1.dtype), w1, b1))
+            return a_head(h[:, :hidden]), v_head(h[:, hidden:])
+
+        self._fast_heads = fast_heads
+        # target-net twin over the flat target buffer (same pinned-front
+        # ordering as the compute flat buffer, so the same offsets hold)
+        t_cnn = self.target.nodes[cnn_node]
+        ta_head = self.target.nodes[a_node].body[2]
+        tv_head = self.target.nodes[v_node].body[2]
+        w1t = self.flat_tparam[: 2 * hidden * in_f].view(2 * hidden, in_f)
+        b1t = self.flat_tparam[2 * hidden * in_f : 2 * hidden * (in_f + 1)]
+
+        def target_heads(x):
+            feat = t_cnn(x)
+            h = F.relu(F.linear(feat.to(w1t.dtype), w1t, b1t))
+            return ta_head(h[:, :hidden]), tv_head(h[:, hidden:])
+
+        self._target_heads = target_heads
+
     def _online_q(self, x):
         if getattr(self, "_fast_fwd", None) is not None:
             return self._fast_fwd(x)
@@ -204,6 +226,7 @@ class ApexLearner(LearnerBase):
             from ..replay import make_apex_schema
 
             wire = make_apex_schema()  # wire format is always NCHW u8
+            self._wire_names = list(wire)
             self._staging = {
                 name: torch.empty((self._STAGE_ROWS, *shape), dtype=dtype
                                   ).pin_memory()
@@ -218,6 +241,9 @@ class ApexLearner(LearnerBase):
     def ingest(self) -> int:
         if self.transport is None:
             return 0
+        if self._ingest_stream is not None and \
+                hasattr(self.transport, "drain_views"):
+            return self._ingest_views()
         got = self.transport.drain()
         if got is None:
             return 0
@@ -228,6 +254,7 @@ class ApexLearner(LearnerBase):
                     for k, v in cols_np.items()}
             self.replay.push(cols, torch.from_numpy(
                 np.ascontiguousarray(prio_np)))
+            self.ingested_total += n
             return n
         # pinned staging ring: one hipHostMalloc for the process lifetime,
         # chunked numpy->pinned memcpy + async H2D on the side stream
@@ -275,7 +302,75 @@ class ApexLearner(LearnerBase):
             torch.cuda.current_stream(self.device).wait_stream(
                 self._ingest_stream)
             done += k
+        self.ingested_total += n
         return n
+
+    def _ingest_views(self) -> int:
+        """Fast ingest: ONE host copy per byte — the actor rings' shm
+        records (AoS) are column-copied straight into the pinned staging
+        buffers (SoA), then hipMemcpyAsync to HBM on the side stream. The
+        reference's equivalent path (Redis lrange + unpickle + np.stack,
+        APE_X/ReplayMemory.py:118-146) copies every byte 4+ times."""
+        stage = self._staging_buffers()
+        np_dtype = self.transport.record_dtype
+        names = list(self._wire_names)
+        total = 0
+        cursor = 0
+
+        def flush():
+            nonlocal cursor
+            k = cursor
+            if k == 0:
+                return
+            with torch.cuda.stream(self._ingest_stream):
+                dev_cols = {
+                    name: stage[name][:k].to(self.device, non_blocking=True)
+                    for name in names
+                }
+                if self._nhwc:
+                    for kk in ("state", "next_state"):
+                        dev_cols[kk] = dev_cols[kk].permute(0, 2, 3, 1
+                                                            ).contiguous()
+                if self.state_dtype != torch.uint8:
+                    for kk in ("state", "next_state"):
+                        dev_cols[kk] = dev_cols[kk].to(self.state_dtype) / 255.0
+                prio_dev = stage["__prio__"][:k].to(self.device,
+                                                    non_blocking=True)
+            # tree/ring mutation ordered after queued compute (see ingest)
+            self._ingest_stream.wait_stream(
+                torch.cuda.current_stream(self.device))
+            with torch.cuda.stream(self._ingest_stream):
+                self.replay.push(dev_cols, prio_dev)
+            self._stage_evt.record(self._ingest_stream)
+            self._stage_busy = True
+            torch.cuda.current_stream(self.device).wait_stream(
+                self._ingest_stream)
+            cursor = 0
+
+        for views, n, advance in self.transport.drain_views():
+            for v in views:
+                rec = v.reshape(-1).view(np_dtype)
+                i = 0
+                while i < len(rec):
+                    if cursor == 0 and self._stage_busy:
+                        # previous async H2D may still read the pinned bufs
+                        self._stage_evt.synchronize()
+                        self._stage_busy = False
+                    k = min(self._STAGE_ROWS - cursor, len(rec) - i)
+                    sl = rec[i : i + k]
+                    for name in names:
+                        stage[name][cursor : cursor + k].numpy()[:] = sl[name]
+                    stage["__prio__"][cursor : cursor + k].numpy()[:] = \
+                        sl["priority"]
+                    cursor += k
+                    i += k
+                    total += k
+                    if cursor == self._STAGE_ROWS:
+                        flush()
+            advance(n)
+        flush()
+        self.ingested_total += total
+        return total
 
     def push_experience(self, cols: Dict[str, torch.Tensor], prio: torch.Tensor):
         """Direct (in-process) push; frames arrive NCHW uint8 (wire format)."""
@@ -336,16 +431,28 @@ class ApexLearner(LearnerBase):
         dones = data["done"].to(self.device)
         weights = weights.to(self.device)
 
-        q_s = self._online_q(s) if cuda else self.net.forward([s])[0]
-        with torch.no_grad():
-            q_sp_on = self._online_q(sp) if cuda else self.net.forward([sp])[0]
-            q_sp_tg = self.target.forward([sp])[0]
-
-        loss, prio, qmean = ops.nstep_dqn_loss(
-            q_s, q_sp_on, q_sp_tg, actions, rewards,
-            dones, weights, self.gamma, self.n_step, self.alpha,
-            with_value_stat=True,
-        )
+        if cuda and getattr(self, "_fast_heads", None) is not None:
+            # fused whole-head path: dueling epilogues live inside the loss
+            # kernel; backward writes (g_adv, g_val) closed-form
+            adv_s, val_s = self._fast_heads(s)
+            with torch.no_grad():
+                adv_on, val_on = self._fast_heads(sp)
+                adv_tg, val_tg = self._target_heads(sp)
+            loss, prio, qmean = ops.dueling_nstep_dqn_loss(
+                adv_s, val_s, adv_on, val_on, adv_tg, val_tg, actions,
+                rewards, dones, weights, self.gamma, self.n_step, self.alpha,
+            )
+        else:
+            q_s = self._online_q(s) if cuda else self.net.forward([s])[0]
+            with torch.no_grad():
+                q_sp_on = (self._online_q(sp) if cuda
+                           else self.net.forward([sp])[0])
+                q_sp_tg = self.target.forward([sp])[0]
+            loss, prio, qmean = ops.nstep_dqn_loss(
+                q_s, q_sp_on, q_sp_tg, actions, rewards,
+                dones, weights, self.gamma, self.n_step, self.alpha,
+                with_value_stat=True,
+            )
         if self.mp is not None:
             self.mp.zero_grads()
             loss.backward()
@@ -520,7 +627,8 @@ class ApexLearner(LearnerBase):
         need = min_items if min_items is not None else self.cfg.buffer_size
         t0 = time.time()
         while len(self.replay) <= need:
-            self.ingest()
+            if self._ingest_thread is None:
+                self.ingest()
             if time.time() - t0 > timeout:
                 raise TimeoutError(
                     f"replay warmup stalled at {len(self.replay)}/{need}"
@@ -528,27 +636,29 @@ class ApexLearner(LearnerBase):
             time.sleep(0.01)
 
     def run(self, max_steps: int = 1_000_000, warmup_items: Optional[int] = None):
+        # ingest runs on its own daemon thread (reference parity: the Replay
+        # drain thread) so drain+pin+H2D never stall the train loop
+        self.start_ingest_thread()
         self.wait_memory(warmup_items)
         self.publish_weights(include_target=True)
         last_loss = None
         stepper = None  # hipGraph-captured once the replay ring is full
-        INGEST_EVERY = 8  # amortize drain+pin+H2D over several train steps
         while self.step_count < max_steps:
-            t0 = time.perf_counter()
-            if self.step_count % INGEST_EVERY == 0:
-                self.ingest()
-            self.time_block("ingest", time.perf_counter() - t0)
             t0 = time.perf_counter()
             if stepper is None and self.device.type == "cuda" \
                     and len(self.replay) >= self.replay.capacity:
                 # n_valid is baked into the captured sample kernel; once the
-                # ring is full it stays at capacity, so capture is safe now
+                # ring is full it stays at capacity, so capture is safe now.
+                # Pause the ingest thread: global-mode stream capture forbids
+                # concurrent stream work from other threads.
+                self.stop_ingest_thread()
                 try:
                     stepper = self.make_graphed_step()
                 except Exception as e:  # pragma: no cover
                     print(f"[APE_X] graph capture failed ({e}); staying eager",
                           flush=True)
                     stepper = self.step
+                self.start_ingest_thread()
             stats = (stepper or self.step)()
             self.time_block("train", time.perf_counter() - t0)
             last_loss = stats["loss"]

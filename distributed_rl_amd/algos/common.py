@@ -46,6 +46,43 @@ class LearnerBase:
         # opt-in replay persistence (SURVEY §5.4 "PER state optional"):
         # off by default — a full Ape-X buffer is multi-GB on disk
         self.checkpoint_replay = os.environ.get("DRL_CKPT_REPLAY", "0") == "1"
+        self.ingested_total = 0
+        self._ingest_thread = None
+        self._ingest_stop = None
+
+    # -- background ingest -------------------------------------------------
+    def start_ingest_thread(self):
+        """Run ingest() on a daemon thread (the reference's Replay daemon
+        thread, APE_X/ReplayMemory.py:19-27, re-created): drain + host
+        copies + async H2D happen concurrently with the learner loop; the
+        numpy memcpys release the GIL, and stream-ordering events keep the
+        replay mutation ordered against the compute stream."""
+        import threading
+
+        if self._ingest_thread is not None or self.transport is None:
+            return
+        self._ingest_stop = threading.Event()
+
+        def loop():
+            while not self._ingest_stop.is_set():
+                try:
+                    n = self.ingest()
+                except Exception as e:  # pragma: no cover
+                    print(f"[ingest-thread] died: {e!r}", flush=True)
+                    return
+                if n == 0:
+                    time.sleep(0.002)
+
+        self._ingest_thread = threading.Thread(target=loop, daemon=True,
+                                               name="drl-ingest")
+        self._ingest_thread.start()
+
+    def stop_ingest_thread(self):
+        if self._ingest_thread is None:
+            return
+        self._ingest_stop.set()
+        self._ingest_thread.join(10)
+        self._ingest_thread = None
 
     # -- model helpers ----------------------------------------------------
     def build_model(self) -> BaseAgent:

@@ -101,6 +101,7 @@ class ImpalaLearner(LearnerBase):
             cols = {k: v.pin_memory().to(self.device, non_blocking=True)
                     for k, v in cols.items()}
         self.replay.push(cols)
+        self.ingested_total += n
         return n
 
     def push_trajectories(self, cols: Dict[str, torch.Tensor]):
@@ -321,27 +322,34 @@ class ImpalaLearner(LearnerBase):
 
     # -- run loop (per-step TB scalars; SURVEY §5.5: 9 scalars per step) ---
     def run(self, max_steps: int = 1_000_000, warmup_items: Optional[int] = None):
+        self.start_ingest_thread()
         need = warmup_items if warmup_items is not None else self.batch_size
         t0 = time.time()
         while len(self.replay) < need:
-            self.ingest()
+            if self._ingest_thread is None:
+                self.ingest()
             if time.time() - t0 > 600:
                 raise TimeoutError("IMPALA replay warmup stalled")
             time.sleep(0.01)
         self.publish_weights()
         stepper = None  # hipGraph-captured once the FIFO ring is full
         while self.step_count < max_steps:
-            self.ingest()
+            if self._ingest_thread is None:
+                self.ingest()
             if stepper is None and self.device.type == "cuda" \
                     and len(self.replay) >= self.replay.capacity:
                 # n_valid is baked into the captured sample kernel; once the
-                # ring is full it stays at capacity, so capture is safe now
+                # ring is full it stays at capacity, so capture is safe now.
+                # Pause the ingest thread: global-mode stream capture forbids
+                # concurrent stream work from other threads.
+                self.stop_ingest_thread()
                 try:
                     stepper = self.make_graphed_step()
                 except Exception as e:  # pragma: no cover
                     print(f"[IMPALA] graph capture failed ({e}); staying eager",
                           flush=True)
                     stepper = self.step
+                self.start_ingest_thread()
             t1 = time.perf_counter()
             stats = (stepper or self.step)()
             dt = time.perf_counter() - t1

@@ -170,6 +170,7 @@ class R2D2Learner(LearnerBase):
                     for k, v in cols.items()}
             prio = prio.pin_memory().to(self.device, non_blocking=True)
         self.replay.push(cols, prio)
+        self.ingested_total += len(prio)
         return len(prio)
 
     def push_sequences(self, cols, prio):
@@ -414,27 +415,34 @@ class R2D2Learner(LearnerBase):
 
     # -- run loop ------------------------------------------------------------
     def run(self, max_steps: int = 1_000_000, warmup_items: Optional[int] = None):
+        self.start_ingest_thread()
         need = warmup_items if warmup_items is not None else self.cfg.buffer_size
         t0 = time.time()
         while len(self.replay) <= need:
-            self.ingest()
+            if self._ingest_thread is None:
+                self.ingest()
             if time.time() - t0 > 1200:
                 raise TimeoutError("R2D2 replay warmup stalled")
             time.sleep(0.01)
         self.publish_weights(include_target=True)
         stepper = None  # hipGraph-captured once the replay ring is full
         while self.step_count < max_steps:
-            self.ingest()
+            if self._ingest_thread is None:
+                self.ingest()
             if stepper is None and self.device.type == "cuda" \
                     and len(self.replay) >= self.replay.capacity:
                 # n_valid is baked into the captured sample kernel; once the
-                # ring is full it stays at capacity, so capture is safe now
+                # ring is full it stays at capacity, so capture is safe now.
+                # Pause the ingest thread: global-mode stream capture forbids
+                # concurrent stream work from other threads.
+                self.stop_ingest_thread()
                 try:
                     stepper = self.make_graphed_step()
                 except Exception as e:  # pragma: no cover
                     print(f"[R2D2] graph capture failed ({e}); staying eager",
                           flush=True)
                     stepper = self.step
+                self.start_ingest_thread()
             stats = (stepper or self.step)()
             if self.step_count % self.LOG_EVERY == 0:
                 rewards = self.transport.drain_rewards() if self.transport else []

@@ -272,6 +272,69 @@ def nstep_dqn_loss(q_s, q_sp_on, q_sp_tg, actions, rewards, dones, weights,
 
 
 # ---------------------------------------------------------------------------
+# K3+K4 fused — whole-head dueling n-step double-DQN loss (round 2):
+# the dueling epilogue of all three forwards lives inside the loss kernel
+# and the backward writes (g_adv, g_val) in closed form — no dueling_fwd,
+# no casts, no dueling_bwd, no dqn_loss_bwd launches.
+# ---------------------------------------------------------------------------
+
+
+class _DuelingDQNLossFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, adv_s, val_s, adv_on, val_on, adv_tg, val_tg, actions,
+                rewards, dones, weights, gamma_n, alpha):
+        ext = hip_ext()
+        B, A = adv_s.shape
+        dev = adv_s.device
+        acc2 = torch.zeros(2, dtype=torch.float32, device=dev)
+        prio = torch.empty(B, dtype=torch.float32, device=dev)
+        coef = torch.empty(B, dtype=torch.float32, device=dev)
+        ext.dueling_dqn_loss_fwd(
+            adv_s.contiguous(), val_s.reshape(B).contiguous(),
+            adv_on.contiguous(), val_on.reshape(B).contiguous(),
+            adv_tg.contiguous(), val_tg.reshape(B).contiguous(),
+            actions.contiguous(), rewards.contiguous(), dones.contiguous(),
+            weights.contiguous(), float(gamma_n), float(alpha),
+            acc2[0:1], prio, coef, acc2[1:2],
+        )
+        ctx.save_for_backward(coef, actions)
+        ctx.shape = (B, A)
+        ctx.in_dtype = adv_s.dtype
+        ctx.mark_non_differentiable(prio)
+        return acc2[0], prio, acc2[1]
+
+    @staticmethod
+    def backward(ctx, gout, _gprio, _gqm):
+        coef, actions = ctx.saved_tensors
+        B, A = ctx.shape
+        g_adv = torch.empty(B, A, dtype=ctx.in_dtype, device=coef.device)
+        g_val = torch.empty(B, 1, dtype=ctx.in_dtype, device=coef.device)
+        hip_ext().dueling_dqn_loss_bwd(coef, actions,
+                                       gout.reshape(1).contiguous(), g_adv,
+                                       g_val)
+        return (g_adv, g_val) + (None,) * 10
+
+
+def dueling_nstep_dqn_loss(adv_s, val_s, adv_on, val_on, adv_tg, val_tg,
+                           actions, rewards, dones, weights, gamma: float,
+                           n_step: int, alpha: float):
+    """Fused dueling + n-step double-DQN loss; returns (loss, prio, qmean).
+    adv_*: (B, A); val_*: (B, 1). Only the (adv_s, val_s) pair is
+    differentiable. On CPU composes dueling_head + nstep_dqn_loss (oracle)."""
+    if _use_hip(adv_s):
+        return _DuelingDQNLossFn.apply(
+            adv_s, val_s, adv_on, val_on, adv_tg, val_tg, actions.long(),
+            rewards.float(), dones.float(), weights.float(),
+            gamma ** n_step, alpha,
+        )
+    q_s = dueling_head(adv_s.float(), val_s.float())
+    q_on = dueling_head(adv_on.float(), val_on.float())
+    q_tg = dueling_head(adv_tg.float(), val_tg.float())
+    return nstep_dqn_loss(q_s, q_on, q_tg, actions, rewards, dones, weights,
+                          gamma, n_step, alpha, with_value_stat=True)
+
+
+# ---------------------------------------------------------------------------
 # K6 / K7
 # ---------------------------------------------------------------------------
 

@@ -698,6 +698,17 @@ bool lstm_step_bwd_bf16(torch::Tensor dg_prev, torch::Tensor dh_init,
                         torch::Tensor tanhc_t, torch::Tensor c_prev,
                         torch::Tensor dgates_t, torch::Tensor dg_bf_t,
                         torch::Tensor dc_out);
+void dueling_dqn_loss_fwd(torch::Tensor adv_s, torch::Tensor val_s,
+                          torch::Tensor adv_on, torch::Tensor val_on,
+                          torch::Tensor adv_tg, torch::Tensor val_tg,
+                          torch::Tensor act, torch::Tensor rew,
+                          torch::Tensor done, torch::Tensor w, double gamma_n,
+                          double alpha, torch::Tensor loss_out,
+                          torch::Tensor prio_out, torch::Tensor grad_coef,
+                          torch::Tensor qmax_out);
+void dueling_dqn_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
+                          torch::Tensor gout, torch::Tensor g_adv,
+                          torch::Tensor g_val);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -739,6 +750,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "one-kernel LSTM timestep fwd: bf16-MFMA hh GEMM + cell (K5 v2)");
   m.def("lstm_step_bwd_bf16", &lstm_step_bwd_bf16,
         "one-kernel LSTM timestep bwd: bf16-MFMA dh GEMM + cell-bwd (K5 v2)");
+  m.def("dueling_dqn_loss_fwd", &dueling_dqn_loss_fwd,
+        "whole-head dueling + n-step double-DQN loss fwd (K3+K4 fused)");
+  m.def("dueling_dqn_loss_bwd", &dueling_dqn_loss_bwd,
+        "whole-head dueling loss bwd: closed-form (g_adv, g_val) (K3+K4)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -1694,4 +1709,128 @@ bool lstm_step_bwd_bf16(torch::Tensor dg_prev, torch::Tensor dh_init,
                      c_prev.data_ptr<float>(), dgates_t.data_ptr<float>(),
                      (__bf16*)dg_bf_t.data_ptr(), dc_out.data_ptr<float>(), B);
   return true;
+}
+
+// ===========================================================================
+// K3+K4 fused (round 2): whole-head dueling n-step double-DQN loss.
+// The dueling epilogue Q = A - mean(A) + V of all THREE forwards (online s,
+// online s', target s') is folded INTO the loss kernel (argmax over Q(s')
+// equals argmax over A(s') since mean/V are row constants), and the
+// backward writes the head gradients (g_adv, g_val) in closed form:
+//   g_adv[i,j] = -coef_i * gout * (1{j==a_i} - 1/A),  g_val[i] = -coef_i*gout
+// Replaces: 3x dueling_fwd + 6 bf16->f32 casts + dueling_bwd + dqn_loss_bwd
+// (~10 launches -> 2). Reference op chains: cfg/ape_x.json:72-88 dueling
+// graph + APE_X/Learner.py:83-114 loss.
+// ===========================================================================
+namespace {
+template <typename T>
+__global__ void dueling_dqn_loss_fwd_kernel(
+    const T* __restrict__ adv_s, const T* __restrict__ val_s,
+    const T* __restrict__ adv_on, const T* __restrict__ val_on,
+    const T* __restrict__ adv_tg, const T* __restrict__ val_tg,
+    const int64_t* __restrict__ act, const float* __restrict__ rew,
+    const float* __restrict__ done, const float* __restrict__ w, int B, int A,
+    float gamma_n, float alpha, float* __restrict__ loss_out /*pre-zeroed*/,
+    float* __restrict__ prio_out, float* __restrict__ grad_coef,
+    float* __restrict__ qmax_out /*pre-zeroed*/) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  float contrib = 0.0f, qm = 0.0f;
+  if (i < B) {
+    const int64_t ro = (int64_t)i * A;
+    // a* = argmax_a Q_on(s',a) = argmax_a A_on(s',a)
+    int a_star = 0;
+    float best = drl_ld(adv_on + ro, 0);
+    for (int a = 1; a < A; ++a) {
+      float v = drl_ld(adv_on + ro, a);
+      if (v > best) { best = v; a_star = a; }
+    }
+    float mean_tg = 0.0f;
+    for (int a = 0; a < A; ++a) mean_tg += drl_ld(adv_tg + ro, a);
+    mean_tg /= A;
+    float q_tg = drl_ld(adv_tg + ro, a_star) - mean_tg + drl_ld(val_tg, i);
+    float target = rew[i] + gamma_n * q_tg * (1.0f - done[i]);
+    float mean_s = 0.0f, smax = drl_ld(adv_s + ro, 0);
+    for (int a = 0; a < A; ++a) {
+      float v = drl_ld(adv_s + ro, a);
+      mean_s += v;
+      smax = fmaxf(smax, v);
+    }
+    mean_s /= A;
+    float vs = drl_ld(val_s, i);
+    float q = drl_ld(adv_s + ro, (int)act[i]) - mean_s + vs;
+    qm = (smax - mean_s + vs) / B;  // mean over rows of max_a Q(s,a)
+    float raw = target - q;
+    float td = fminf(1.0f, fmaxf(-1.0f, raw));
+    prio_out[i] = __powf(fabsf(td) + 1e-7f, alpha);
+    float in_range = (raw > -1.0f && raw < 1.0f) ? 1.0f : 0.0f;
+    float invB = 1.0f / B;
+    grad_coef[i] = w[i] * td * in_range * invB;  // dL/dQ(s,a) = -grad_coef
+    contrib = 0.5f * w[i] * td * td * invB;
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    contrib += __shfl_down(contrib, off, 64);
+    qm += __shfl_down(qm, off, 64);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    if (contrib != 0.0f) atomicAdd(loss_out, contrib);
+    if (qm != 0.0f) atomicAdd(qmax_out, qm);
+  }
+}
+
+template <typename T>
+__global__ void dueling_dqn_loss_bwd_kernel(
+    const float* __restrict__ grad_coef, const int64_t* __restrict__ act,
+    const float* __restrict__ gout, int B, int A, T* __restrict__ g_adv,
+    T* __restrict__ g_val) {
+  int j = blockIdx.x * blockDim.x + threadIdx.x;
+  if (j >= B * A) return;
+  int i = j / A;
+  int a = j - i * A;
+  float c = -grad_coef[i] * gout[0];
+  float ind = (a == (int)act[i]) ? 1.0f : 0.0f;
+  g_adv[j] = (T)(c * (ind - 1.0f / A));
+  if (a == 0) g_val[i] = (T)c;
+}
+}  // namespace
+
+void dueling_dqn_loss_fwd(torch::Tensor adv_s, torch::Tensor val_s,
+                          torch::Tensor adv_on, torch::Tensor val_on,
+                          torch::Tensor adv_tg, torch::Tensor val_tg,
+                          torch::Tensor act, torch::Tensor rew,
+                          torch::Tensor done, torch::Tensor w, double gamma_n,
+                          double alpha, torch::Tensor loss_out,
+                          torch::Tensor prio_out, torch::Tensor grad_coef,
+                          torch::Tensor qmax_out) {
+  int B = (int)adv_s.size(0), A = (int)adv_s.size(1);
+  dim3 grid(ceil_div(B, kBlock)), blk(kBlock);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, adv_s.scalar_type(),
+      "dueling_dqn_loss_fwd", [&] {
+        hipLaunchKernelGGL(
+            dueling_dqn_loss_fwd_kernel<scalar_t>, grid, blk, 0, cur_stream(),
+            adv_s.data_ptr<scalar_t>(), val_s.data_ptr<scalar_t>(),
+            adv_on.data_ptr<scalar_t>(), val_on.data_ptr<scalar_t>(),
+            adv_tg.data_ptr<scalar_t>(), val_tg.data_ptr<scalar_t>(),
+            act.data_ptr<int64_t>(), rew.data_ptr<float>(),
+            done.data_ptr<float>(), w.data_ptr<float>(), B, A, (float)gamma_n,
+            (float)alpha, loss_out.data_ptr<float>(),
+            prio_out.data_ptr<float>(), grad_coef.data_ptr<float>(),
+            qmax_out.data_ptr<float>());
+      });
+}
+
+void dueling_dqn_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
+                          torch::Tensor gout, torch::Tensor g_adv,
+                          torch::Tensor g_val) {
+  int B = (int)g_adv.size(0), A = (int)g_adv.size(1);
+  dim3 grid(ceil_div((int64_t)B * A, kBlock)), blk(kBlock);
+  AT_DISPATCH_FLOATING_TYPES_AND2(
+      at::ScalarType::BFloat16, at::ScalarType::Half, g_adv.scalar_type(),
+      "dueling_dqn_loss_bwd", [&] {
+        hipLaunchKernelGGL(dueling_dqn_loss_bwd_kernel<scalar_t>, grid, blk, 0,
+                           cur_stream(), grad_coef.data_ptr<float>(),
+                           act.data_ptr<int64_t>(), gout.data_ptr<float>(), B,
+                           A, g_adv.data_ptr<scalar_t>(),
+                           g_val.data_ptr<scalar_t>());
+      });
 }

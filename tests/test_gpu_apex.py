@@ -78,3 +78,45 @@ def test_fast_forward_matches_dag():
     stats = learner.step()
     torch.cuda.synchronize()
     assert float(stats["loss"]) == float(stats["loss"])
+
+
+def test_fused_dueling_dqn_loss_matches_composition():
+    """K3+K4 fused op vs the composed dueling_head + nstep_dqn_loss path on
+    identical bf16 inputs: loss/prio/value-stat and the head grads must
+    match (the fused backward is the closed form of the composition)."""
+    from distributed_rl_amd import ops
+
+    dev = "cuda:0"
+    torch.manual_seed(31)
+    B, A = 64, 6
+    mk = lambda *s: (torch.randn(*s, device=dev) * 2).to(torch.bfloat16)
+    adv_s = mk(B, A).requires_grad_(True)
+    val_s = mk(B, 1).requires_grad_(True)
+    adv_s2 = adv_s.detach().clone().requires_grad_(True)
+    val_s2 = val_s.detach().clone().requires_grad_(True)
+    adv_on, val_on, adv_tg, val_tg = mk(B, A), mk(B, 1), mk(B, A), mk(B, 1)
+    actions = torch.randint(0, A, (B,), device=dev)
+    rewards = torch.randn(B, device=dev)
+    dones = (torch.rand(B, device=dev) < 0.1).float()
+    weights = torch.rand(B, device=dev) + 0.5
+
+    loss, prio, qm = ops.dueling_nstep_dqn_loss(
+        adv_s, val_s, adv_on, val_on, adv_tg, val_tg, actions, rewards,
+        dones, weights, 0.99, 3, 0.6)
+    loss.backward()
+
+    q_s = ops.dueling_head(adv_s2.float(), val_s2.float())
+    with torch.no_grad():
+        q_on = ops.dueling_head(adv_on.float(), val_on.float())
+        q_tg = ops.dueling_head(adv_tg.float(), val_tg.float())
+    loss2, prio2, qm2 = ops.nstep_dqn_loss(
+        q_s, q_on, q_tg, actions, rewards, dones, weights, 0.99, 3, 0.6,
+        with_value_stat=True)
+    loss2.backward()
+
+    torch.cuda.synchronize()
+    assert torch.allclose(loss, loss2, atol=1e-5), (float(loss), float(loss2))
+    assert torch.allclose(prio, prio2, atol=1e-5)
+    assert torch.allclose(qm, qm2, atol=1e-4)
+    assert torch.allclose(adv_s.grad.float(), adv_s2.grad.float(), atol=2e-3)
+    assert torch.allclose(val_s.grad.float(), val_s2.grad.float(), atol=2e-3)
