@@ -80,6 +80,13 @@ class ImpalaLearner(LearnerBase):
         self.transport = transport
         self.publish_every = publish_every
         self.gamma = cfg.gamma
+        # Replay-reuse cap (cfg MAX_REPLAY_REUSE, default off = reference
+        # behavior): bound consumed/ingested trajectories so a fast learner
+        # cannot spin on stale FIFO contents — the round-1 live run showed
+        # entropy collapse after ~2k steps when the learner outran the fleet
+        # (profiles/r01_learning_sanity.md); the reference is implicitly
+        # paced by its slow Redis pipe, this makes the ratio explicit.
+        self.max_replay_reuse = float(cfg.get("MAX_REPLAY_REUSE", 0) or 0)
         self.reducer = None
         self._sqsum_buf = (
             torch.zeros(1, device=self.device) if self.device.type == "cuda" else None
@@ -336,6 +343,17 @@ class ImpalaLearner(LearnerBase):
         while self.step_count < max_steps:
             if self._ingest_thread is None:
                 self.ingest()
+            if self.max_replay_reuse > 0 and self.transport is not None:
+                # block until the fleet has produced enough fresh unrolls
+                t_gate = time.time()
+                while ((self.step_count + 1) * self.batch_size
+                       > self.max_replay_reuse * max(self.ingested_total, 1)):
+                    if self._ingest_thread is None:
+                        self.ingest()
+                    time.sleep(0.002)
+                    if time.time() - t_gate > 600:
+                        raise TimeoutError("IMPALA reuse gate starved "
+                                           "(actors dead?)")
             if stepper is None and self.device.type == "cuda" \
                     and len(self.replay) >= self.replay.capacity:
                 # n_valid is baked into the captured sample kernel; once the

@@ -100,3 +100,51 @@ def test_spsc_ring_threaded_stress():
         assert ring.drops == 0 or ring.drops > 0  # drops counted, none lost
     finally:
         ring.close(unlink=True)
+
+
+def test_impala_replay_reuse_cap():
+    """MAX_REPLAY_REUSE bounds consumed/ingested: with the cap set, run()
+    must gate learner steps on fleet production rather than spin on stale
+    FIFO contents."""
+    import copy
+    import threading
+    import time
+
+    import torch
+
+    from distributed_rl_amd.actors.transport import InprocPipe
+    from distributed_rl_amd.algos.impala import ImpalaLearner
+    from distributed_rl_amd.config import Config, load_config
+
+    raw = copy.deepcopy(load_config("impala").raw)
+    raw.update({"REPLAY_MEMORY_LEN": 64, "BATCHSIZE": 2,
+                "MAX_REPLAY_REUSE": 4})
+    cfg = Config(raw=raw)
+    pipe = InprocPipe()
+    learner = ImpalaLearner(cfg, device="cpu", transport=pipe, enable_tb=False)
+    T = cfg.unroll_step
+
+    def feed(n):
+        for _ in range(n):
+            pipe.push({
+                "states": torch.randint(0, 255, (1, T + 1, 4, 84, 84),
+                                        dtype=torch.uint8).numpy(),
+                "actions": torch.randint(0, 6, (1, T),
+                                         dtype=torch.int32).numpy(),
+                "mu": torch.full((1, T), 1 / 6).numpy(),
+                "rewards": torch.randn(1, T).numpy(),
+                "not_done": torch.ones(1).numpy(),
+            }, None)
+
+    feed(8)  # 8 trajectories -> cap allows 4*8/2 = 16 learner steps
+    th = threading.Thread(
+        target=lambda: learner.run(max_steps=30, warmup_items=2), daemon=True)
+    th.start()
+    time.sleep(8)
+    # gated: cannot exceed reuse*ingested/batch
+    assert learner.step_count <= 16, learner.step_count
+    assert learner.step_count >= 10  # but it did train up to the gate
+    feed(8)  # fresh data lifts the gate
+    th.join(30)
+    assert not th.is_alive()
+    assert learner.step_count == 30
