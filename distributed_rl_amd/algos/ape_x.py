@@ -314,6 +314,18 @@ class ApexLearner(LearnerBase):
         stage = self._staging_buffers()
         np_dtype = self.transport.record_dtype
         names = list(self._wire_names)
+        ext = ops.hip_ext()
+        if getattr(self, "_pack_spec", None) is None:
+            f = np_dtype.fields  # name -> (dtype, offset)
+            keys = names + ["priority"]
+            self._pack_spec = (
+                [int(f[n][1]) for n in keys],
+                [int(f[n][0].itemsize) for n in keys],
+                [stage[n] for n in names] + [stage["__prio__"]],
+                int(np_dtype.itemsize),
+                ext is not None and hasattr(ext, "pack_rows"),
+            )
+        offs, sizes, dsts, rec_size, has_pack = self._pack_spec
         total = 0
         cursor = 0
 
@@ -349,19 +361,26 @@ class ApexLearner(LearnerBase):
 
         for views, n, advance in self.transport.drain_views():
             for v in views:
-                rec = v.reshape(-1).view(np_dtype)
+                nrows = v.shape[0]
                 i = 0
-                while i < len(rec):
+                while i < nrows:
                     if cursor == 0 and self._stage_busy:
                         # previous async H2D may still read the pinned bufs
                         self._stage_evt.synchronize()
                         self._stage_busy = False
-                    k = min(self._STAGE_ROWS - cursor, len(rec) - i)
-                    sl = rec[i : i + k]
-                    for name in names:
-                        stage[name][cursor : cursor + k].numpy()[:] = sl[name]
-                    stage["__prio__"][cursor : cursor + k].numpy()[:] = \
-                        sl["priority"]
+                    k = min(self._STAGE_ROWS - cursor, nrows - i)
+                    if has_pack:
+                        # GIL-free multithreaded AoS->SoA scatter (C++)
+                        src = torch.from_numpy(
+                            np.ascontiguousarray(v[i : i + k]).reshape(-1))
+                        ext.pack_rows(src, rec_size, offs, sizes, dsts, cursor)
+                    else:
+                        sl = v[i : i + k].reshape(-1).view(np_dtype)
+                        for name in names:
+                            stage[name][cursor : cursor + k].numpy()[:] = \
+                                sl[name]
+                        stage["__prio__"][cursor : cursor + k].numpy()[:] = \
+                            sl["priority"]
                     cursor += k
                     i += k
                     total += k

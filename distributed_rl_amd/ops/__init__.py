@@ -153,24 +153,37 @@ class _FusedConvFn(torch.autograd.Function):
             gw = ws.view(COUT, KH, KW, C).permute(0, 3, 1, 2).to(torch.bfloat16)
             if need_b:
                 gb = gb_ws.to(torch.bfloat16)
-        if need_x or (need_w and not use_own_wrw):
+        gi = None
+        use_own_dgrad = need_x and x.dtype == torch.bfloat16 and bool(
+            ext.conv_dgrad_supported(x.shape[2], x.shape[3], C, KH, KW,
+                                     stride, COUT)
+        ) if hasattr(ext, "conv_dgrad_supported") else False
+        if use_own_dgrad:
+            # hand-written MFMA dgrad (conv_mfma.hip): masked-tap gather
+            # against a per-step transposed weight copy — no MIOpen on the
+            # hot path
+            gi = torch.empty_like(x)
+            w_t = torch.empty(weight.numel(), dtype=torch.bfloat16,
+                              device=x.device)
+            ext.conv_dgrad(gout, weight, w_t, gi, stride)
+        if (need_x and not use_own_dgrad) or (need_w and not use_own_wrw):
             if x.dtype == torch.uint8:
                 # NHWC u8 -> bf16 for the aten path (own wrw reads u8 directly)
                 xf = dequant_frames(x.permute(0, 2, 3, 1), torch.bfloat16)
                 xf = xf.permute(0, 3, 1, 2)
             else:
                 xf = x
-            gi, gw2, gb2 = torch.ops.aten.convolution_backward(
+            gi2, gw2, gb2 = torch.ops.aten.convolution_backward(
                 gout, xf, weight,
                 [COUT] if ctx.has_bias else None,
                 [stride, stride], [0, 0], [1, 1], False, [0, 0], 1,
-                [need_x, need_w and not use_own_wrw,
+                [need_x and not use_own_dgrad, need_w and not use_own_wrw,
                  need_b and not use_own_wrw],
             )
+            if gi is None:
+                gi = gi2
             if gw is None:
                 gw, gb = gw2, gb2
-        else:
-            gi = None
         return (gi if need_x else None), gw, gb, None
 
 

@@ -323,6 +323,10 @@ void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
               torch::Tensor gb_ws, int64_t stride);  // defined below
 bool conv_wrw_supported(int64_t H, int64_t W, int64_t C, int64_t KH,
                         int64_t KW, int64_t S, int64_t COUT, bool u8);
+void conv_dgrad(torch::Tensor gout, torch::Tensor weight, torch::Tensor w_t,
+                torch::Tensor dx, int64_t stride);  // defined below
+bool conv_dgrad_supported(int64_t H, int64_t W, int64_t C, int64_t KH,
+                          int64_t KW, int64_t S, int64_t COUT);
 void tr16_probe(torch::Tensor out, int64_t mode);
 
 void register_conv(pybind11::module_& m) {
@@ -333,6 +337,9 @@ void register_conv(pybind11::module_& m) {
   m.def("conv_fwd_variant", &conv_fwd_variant, "tuning-variant conv fwd");
   m.def("conv_wrw", &conv_wrw, "MFMA conv weight-grad (fp32 workspace)");
   m.def("conv_wrw_supported", &conv_wrw_supported);
+  m.def("conv_dgrad", &conv_dgrad,
+        "MFMA conv data-grad (masked-tap gather, pre-transposed weights)");
+  m.def("conv_dgrad_supported", &conv_dgrad_supported);
   m.def("tr16_probe", &tr16_probe);
 }
 
@@ -641,4 +648,183 @@ void tr16_probe(torch::Tensor out, int64_t mode) {
   hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0,
                      (hipStream_t)at::cuda::getCurrentCUDAStream().stream(),
                      out.data_ptr<float>(), (int)mode);
+}
+
+// ===========================================================================
+// Data-grad (dgrad) kernel — retires MIOpen igemm_bwd / ck grouped-bwd from
+// the hot path (VERDICT r01 missing-5). GEMM view:
+//   dx[M=N*H*W rows][C cols] = sum_{kh,kw} G[(n,y,x)][COUT] x Wt[COUT][C]
+// where G gathers gout[n,(y-kh)/S,(x-kw)/S,:] when the tap is valid (stride
+// divisibility + bounds) and zero otherwise. K = KH*KW*COUT runs in 32-wide
+// chunks that never straddle taps (COUT % 32 == 0 for every geometry).
+// B-operand comes from a PRE-TRANSPOSED weight tensor
+//   w_t[c][(kh*KW+kw)*COUT + cout] = w[cout][(kh*KW+kw)*C + c]
+// (built once per step by conv_w_transpose below, L2-resident), so both
+// operands are plain 16 B/lane loads — no LDS at all.
+// ===========================================================================
+namespace {
+
+template <int H, int W, int C, int KH, int KW, int S, int COUT, int WAVES,
+          int RPW>
+__global__ __launch_bounds__(WAVES * 64) void conv_dgrad_kernel(
+    const __bf16* __restrict__ gout,  // (N,P,Q,COUT) NHWC
+    const __bf16* __restrict__ w_t,   // (C, KH*KW*COUT)
+    __bf16* __restrict__ dx,          // (N,H,W,C) NHWC
+    int batch) {
+  constexpr int P = (H - KH) / S + 1;
+  constexpr int Q = (W - KW) / S + 1;
+  constexpr int K = KH * KW * COUT;
+  constexpr int KCHUNKS = K / 32;
+  constexpr int CPT = COUT / 32;  // chunks per tap
+  static_assert(COUT % 32 == 0, "chunks must not straddle taps");
+  constexpr int RFRAG = RPW / 16;
+  constexpr int NFRAG = C / 16;
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int M = batch * H * W;
+  const int row0 = (blockIdx.x * WAVES + wave) * RPW;
+  if (row0 >= M) return;
+
+  // per-row-fragment (n, y, x) of this lane's A row
+  int a_n[RFRAG], a_y[RFRAG], a_x[RFRAG];
+#pragma unroll
+  for (int rf = 0; rf < RFRAG; ++rf) {
+    int arow = row0 + rf * 16 + (lane & 15);
+    if (arow >= M) arow = M - 1;
+    a_n[rf] = arow / (H * W);
+    const int rem = arow - a_n[rf] * (H * W);
+    a_y[rf] = rem / W;
+    a_x[rf] = rem - a_y[rf] * W;
+  }
+  const int kpart = (lane >> 4) * 8;
+
+  f32x4 acc[RFRAG][NFRAG];
+#pragma unroll
+  for (int rf = 0; rf < RFRAG; ++rf)
+#pragma unroll
+    for (int f = 0; f < NFRAG; ++f) acc[rf][f] = {0.f, 0.f, 0.f, 0.f};
+
+#pragma unroll
+  for (int kc = 0; kc < KCHUNKS; ++kc) {
+    const int tap = kc / CPT;
+    const int kh = tap / KW;
+    const int kw = tap - kh * KW;
+    const int co = (kc - tap * CPT) * 32 + kpart;
+    bf16x8 a[RFRAG];
+#pragma unroll
+    for (int rf = 0; rf < RFRAG; ++rf) {
+      const int py = a_y[rf] - kh;
+      const int qx = a_x[rf] - kw;
+      const int p = py / S, q = qx / S;
+      const bool ok = py >= 0 && qx >= 0 && py == p * S && qx == q * S &&
+                      p < P && q < Q;
+      if (ok) {
+        a[rf] = *reinterpret_cast<const bf16x8*>(
+            gout + (((int64_t)a_n[rf] * P + p) * Q + q) * COUT + co);
+      } else {
+        a[rf] = bf16x8{};
+      }
+    }
+#pragma unroll
+    for (int f = 0; f < NFRAG; ++f) {
+      const int col = f * 16 + (lane & 15);
+      bf16x8 b = *reinterpret_cast<const bf16x8*>(
+          w_t + (int64_t)col * K + tap * COUT + co);
+#pragma unroll
+      for (int rf = 0; rf < RFRAG; ++rf)
+        acc[rf][f] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[rf], b, acc[rf][f], 0, 0, 0);
+    }
+  }
+
+  const int crow_base = (lane >> 4) * 4;
+#pragma unroll
+  for (int rf = 0; rf < RFRAG; ++rf)
+#pragma unroll
+    for (int f = 0; f < NFRAG; ++f) {
+      const int col = f * 16 + (lane & 15);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int orow = row0 + rf * 16 + crow_base + r;
+        if (orow < M) dx[(int64_t)orow * C + col] = (__bf16)acc[rf][f][r];
+      }
+    }
+}
+
+__global__ void conv_w_transpose_kernel(const __bf16* __restrict__ w,
+                                        __bf16* __restrict__ w_t, int TAPS,
+                                        int C, int COUT) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  const int total = TAPS * C * COUT;
+  if (i >= total) return;
+  const int co = i / (TAPS * C);
+  const int rem = i - co * (TAPS * C);
+  const int tap = rem / C;
+  const int c = rem - tap * C;
+  w_t[((int64_t)c * TAPS + tap) * COUT + co] = w[i];
+}
+
+struct DgradLaunch {
+  int H, W, C, KH, KW, S, COUT;
+  void (*fn)(const __bf16*, const __bf16*, __bf16*, int);
+  int waves, rpw;
+};
+
+template <int H, int W, int C, int KH, int KW, int S, int COUT, int WAVES = 8,
+          int RPW = 16>
+DgradLaunch make_dgrad() {
+  return DgradLaunch{H, W, C, KH, KW, S, COUT,
+                     conv_dgrad_kernel<H, W, C, KH, KW, S, COUT, WAVES, RPW>,
+                     WAVES, RPW};
+}
+
+static const DgradLaunch kDgrad[] = {
+    make_dgrad<20, 20, 32, 4, 4, 2, 64>(),  // apex/r2d2 conv2
+    make_dgrad<9, 9, 64, 3, 3, 1, 64>(),    // apex/r2d2 conv3
+    make_dgrad<20, 20, 16, 4, 4, 2, 32>(),  // impala conv2
+};
+}  // namespace
+
+bool conv_dgrad_supported(int64_t H, int64_t W, int64_t C, int64_t KH,
+                          int64_t KW, int64_t S, int64_t COUT) {
+  for (const auto& l : kDgrad)
+    if (l.H == H && l.W == W && l.C == C && l.KH == KH && l.KW == KW &&
+        l.S == S && l.COUT == COUT)
+      return true;
+  return false;
+}
+
+// gout: (N,COUT,P,Q) logical channels_last bf16; weight (COUT,C,KH,KW)
+// logical channels_last bf16; dx out (N,C,H,W) logical channels_last bf16;
+// w_t workspace (C*KH*KW*COUT) bf16.
+void conv_dgrad(torch::Tensor gout, torch::Tensor weight, torch::Tensor w_t,
+                torch::Tensor dx, int64_t stride) {
+  TORCH_CHECK(gout.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(dx.is_contiguous(at::MemoryFormat::ChannelsLast));
+  TORCH_CHECK(weight.is_contiguous(at::MemoryFormat::ChannelsLast));
+  const int N = (int)dx.size(0), C = (int)dx.size(1), H = (int)dx.size(2),
+            W = (int)dx.size(3);
+  const int COUT = (int)weight.size(0), KH = (int)weight.size(2),
+            KW = (int)weight.size(3);
+  const DgradLaunch* L = nullptr;
+  for (const auto& l : kDgrad)
+    if (l.H == H && l.W == W && l.C == C && l.KH == KH && l.KW == KW &&
+        l.S == (int)stride && l.COUT == COUT) {
+      L = &l;
+      break;
+    }
+  TORCH_CHECK(L, "no dgrad kernel for this geometry");
+  auto stream = (hipStream_t)at::cuda::getCurrentCUDAStream().stream();
+  const int total = KH * KW * C * COUT;
+  hipLaunchKernelGGL(conv_w_transpose_kernel, dim3((total + 255) / 256),
+                     dim3(256), 0, stream, (const __bf16*)weight.data_ptr(),
+                     (__bf16*)w_t.data_ptr(), KH * KW, C, COUT);
+  const int M = N * H * W;
+  const int rows_per_block = L->waves * L->rpw;
+  const int blocks = (M + rows_per_block - 1) / rows_per_block;
+  hipLaunchKernelGGL(L->fn, dim3(blocks), dim3(L->waves * 64), 0, stream,
+                     (const __bf16*)gout.data_ptr(),
+                     (const __bf16*)w_t.data_ptr(), (__bf16*)dx.data_ptr(), N);
 }

@@ -709,6 +709,9 @@ void dueling_dqn_loss_fwd(torch::Tensor adv_s, torch::Tensor val_s,
 void dueling_dqn_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
                           torch::Tensor gout, torch::Tensor g_adv,
                           torch::Tensor g_val);
+void pack_rows(torch::Tensor src, int64_t rec_size, std::vector<int64_t> offs,
+               std::vector<int64_t> sizes, std::vector<torch::Tensor> dsts,
+               int64_t dst_row);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -754,6 +757,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "whole-head dueling + n-step double-DQN loss fwd (K3+K4 fused)");
   m.def("dueling_dqn_loss_bwd", &dueling_dqn_loss_bwd,
         "whole-head dueling loss bwd: closed-form (g_adv, g_val) (K3+K4)");
+  m.def("pack_rows", &pack_rows,
+        "AoS records -> SoA pinned staging, GIL-free multithreaded (C2)",
+        pybind11::call_guard<pybind11::gil_scoped_release>());
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -1833,4 +1839,34 @@ void dueling_dqn_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
                            A, g_adv.data_ptr<scalar_t>(),
                            g_val.data_ptr<scalar_t>());
       });
+}
+
+// ===========================================================================
+// Host-side ingest pack (C2): AoS shm records -> SoA pinned staging in one
+// GIL-free multithreaded pass. The Python per-field numpy copies held the
+// GIL and starved the learner loop (measured 2.9 ms/step whole-node with
+// the ingest thread vs 0.67 learner-only); this is one call per ring chunk.
+// ===========================================================================
+void pack_rows(torch::Tensor src, int64_t rec_size,
+               std::vector<int64_t> offs, std::vector<int64_t> sizes,
+               std::vector<torch::Tensor> dsts, int64_t dst_row) {
+  TORCH_CHECK(src.device().is_cpu() && src.dtype() == torch::kUInt8);
+  const uint8_t* s = src.data_ptr<uint8_t>();
+  int64_t nrows = src.numel() / rec_size;
+  int nf = (int)offs.size();
+  std::vector<uint8_t*> dp(nf);
+  std::vector<int64_t> stride(nf);
+  for (int f = 0; f < nf; ++f) {
+    dp[f] = (uint8_t*)dsts[f].data_ptr();
+    stride[f] = dsts[f].numel() * dsts[f].element_size() / dsts[f].size(0);
+    dp[f] += dst_row * stride[f];
+    TORCH_CHECK(stride[f] == sizes[f], "dst row stride != field size");
+  }
+  at::parallel_for(0, nrows, 16, [&](int64_t b, int64_t e) {
+    for (int64_t r = b; r < e; ++r) {
+      const uint8_t* row = s + r * rec_size;
+      for (int f = 0; f < nf; ++f)
+        memcpy(dp[f] + r * sizes[f], row + offs[f], (size_t)sizes[f]);
+    }
+  });
 }
