@@ -72,17 +72,27 @@ class _ManualLSTMSeq(torch.autograd.Function):
         )
         h_bfs = None
         used = False
+        ctx.persist = False
         if use_v2:
             w_bf = w_hh.detach().to(torch.bfloat16).contiguous()
             h_bfs = torch.empty(T + 1, B, H, dtype=torch.bfloat16, device=dev)
             h_bfs[0].copy_(h0)
-            used = True
-            for t in range(T):
-                if not ext.lstm_step_fwd_bf16(
-                        xp[t], h_bfs[t], cs[t], w_bf, hs[t + 1], cs[t + 1],
-                        h_bfs[t + 1], acts[t], tanhc[t]):
-                    used = False
-                    break
+            # K5 v3 (persistent whole-sequence, grid barriers): one launch
+            # for the entire recurrence instead of T
+            if _os.environ.get("DRL_LSTM_PERSISTENT_BF16", "1") == "1" and \
+                    hasattr(ext, "lstm_seq_fwd_bf16"):
+                ctr = torch.zeros(1, dtype=torch.int32, device=dev)
+                used = bool(ext.lstm_seq_fwd_bf16(
+                    xp, cs[0], w_bf, hs, cs, h_bfs, acts, tanhc, ctr))
+                ctx.persist = used
+            if not used:
+                used = True
+                for t in range(T):
+                    if not ext.lstm_step_fwd_bf16(
+                            xp[t], h_bfs[t], cs[t], w_bf, hs[t + 1], cs[t + 1],
+                            h_bfs[t + 1], acts[t], tanhc[t]):
+                        used = False
+                        break
         if not used and _os.environ.get("DRL_LSTM_PERSISTENT", "0") == "1":
             ctr = torch.zeros(1, dtype=torch.int32, device=dev)
             used = bool(ext.lstm_seq_persistent(
@@ -118,20 +128,31 @@ class _ManualLSTMSeq(torch.autograd.Function):
         dgates_all = torch.empty(T, B, 4 * H, device=dev)
         gout = gout.contiguous()
         if ctx.v2:
-            # one fused kernel per step: dh GEMM (bf16 MFMA) + cell-bwd
+            # one fused kernel per step: dh GEMM (bf16 MFMA) + cell-bwd —
+            # or, when the forward used the persistent kernel, ONE launch
+            # for the whole reversed scan
             dg_bf_all = torch.empty(T, B, 4 * H, dtype=torch.bfloat16,
                                     device=dev)
             w_t_bf = w_hh.t().contiguous().to(torch.bfloat16)
             dh_init = gh_T.contiguous()
-            dc = gc_T.contiguous().clone()
-            dc_next = torch.empty(B, H, device=dev)
-            empty_bf = torch.empty(0, dtype=torch.bfloat16, device=dev)
-            for t in range(T - 1, -1, -1):
-                dg_prev = dg_bf_all[t + 1] if t < T - 1 else empty_bf
-                ext.lstm_step_bwd_bf16(dg_prev, dh_init, gout[t], dc,
-                                       w_t_bf, acts[t], tanhc[t], cs[t],
-                                       dgates_all[t], dg_bf_all[t], dc_next)
-                dc, dc_next = dc_next, dc
+            done = False
+            if ctx.persist and hasattr(ext, "lstm_seq_bwd_bf16"):
+                ctr = torch.zeros(1, dtype=torch.int32, device=dev)
+                dc = torch.empty(B, H, device=dev)
+                done = bool(ext.lstm_seq_bwd_bf16(
+                    dh_init, gout, gc_T.contiguous(), w_t_bf, acts, tanhc,
+                    cs, dgates_all, dg_bf_all, dc, ctr))
+            if not done:
+                dc = gc_T.contiguous().clone()
+                dc_next = torch.empty(B, H, device=dev)
+                empty_bf = torch.empty(0, dtype=torch.bfloat16, device=dev)
+                for t in range(T - 1, -1, -1):
+                    dg_prev = dg_bf_all[t + 1] if t < T - 1 else empty_bf
+                    ext.lstm_step_bwd_bf16(dg_prev, dh_init, gout[t], dc,
+                                           w_t_bf, acts[t], tanhc[t], cs[t],
+                                           dgates_all[t], dg_bf_all[t],
+                                           dc_next)
+                    dc, dc_next = dc_next, dc
             dh = dgates_all[0].mm(w_hh)  # grad wrt h0 (fp32, once per seq)
             dg_flat = dgates_all.reshape(T * B, 4 * H)
             dg_bf = dg_bf_all.reshape(T * B, 4 * H)

@@ -712,6 +712,16 @@ void dueling_dqn_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
 void pack_rows(torch::Tensor src, int64_t rec_size, std::vector<int64_t> offs,
                std::vector<int64_t> sizes, std::vector<torch::Tensor> dsts,
                int64_t dst_row);
+bool lstm_seq_fwd_bf16(torch::Tensor xp, torch::Tensor c0, torch::Tensor w_bf,
+                       torch::Tensor hs, torch::Tensor cs, torch::Tensor h_bfs,
+                       torch::Tensor acts, torch::Tensor tanhc,
+                       torch::Tensor ctr);
+bool lstm_seq_bwd_bf16(torch::Tensor dh_init, torch::Tensor gout,
+                       torch::Tensor dc_T, torch::Tensor w_t_bf,
+                       torch::Tensor acts, torch::Tensor tanhc,
+                       torch::Tensor cs, torch::Tensor dgates,
+                       torch::Tensor dgates_bf, torch::Tensor dc0_out,
+                       torch::Tensor ctr);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -760,6 +770,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("pack_rows", &pack_rows,
         "AoS records -> SoA pinned staging, GIL-free multithreaded (C2)",
         pybind11::call_guard<pybind11::gil_scoped_release>());
+  m.def("lstm_seq_fwd_bf16", &lstm_seq_fwd_bf16,
+        "persistent whole-sequence LSTM fwd, bf16-MFMA hh + grid barriers "
+        "(K5 v3)");
+  m.def("lstm_seq_bwd_bf16", &lstm_seq_bwd_bf16,
+        "persistent whole-sequence LSTM bwd, bf16-MFMA dh + grid barriers "
+        "(K5 v3)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -1869,4 +1885,278 @@ void pack_rows(torch::Tensor src, int64_t rec_size,
         memcpy(dp[f] + r * sizes[f], row + offs[f], (size_t)sizes[f]);
     }
   });
+}
+
+// ===========================================================================
+// K5 v3: persistent whole-sequence LSTM on bf16 MFMA — the entire T-step
+// recurrence (fwd) / reversed scan (bwd) in ONE launch. The v2 per-step
+// kernels measure 9.6 us fwd / 10.8 us bwd, dominated by launch+tail on a
+// ~1-2 us workload; here 32 blocks stay resident and synchronize with the
+// agent-scope release/acquire grid barrier between timesteps (G16 recipe,
+// same as lstm_seq_persistent — which lost only because its fp32 MFMA
+// inner loop was 8x the instruction count). Cell/carry state (c fwd,
+// dc bwd) lives in REGISTERS: the thread->element map is fixed across
+// timesteps, so only h (fwd) / dgates (bwd) cross block boundaries.
+// ===========================================================================
+namespace {
+
+template <int H>
+__global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
+    const float* __restrict__ xp,      // (T, B, 4H)
+    const float* __restrict__ c0,      // (B, H)
+    const __bf16* __restrict__ w_bf,   // (4H, H)
+    float* __restrict__ hs,            // (T+1, B, H); hs[0] ignored here
+    float* __restrict__ cs,            // (T+1, B, H); cs[0] = c0 pre-filled
+    __bf16* __restrict__ h_bfs,        // (T+1, B, H); [0] pre-filled
+    float* __restrict__ acts,          // (T, B, 4H)
+    float* __restrict__ tanhc,         // (T, B, H)
+    unsigned int* __restrict__ ctr,    // zeroed before launch
+    int B, int T) {
+  __shared__ float gbuf[4][32][16];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int s16 = blockIdx.x * 16;
+  const int col = wave * H + s16 + (lane & 15);
+  const __bf16* wrow = w_bf + (int64_t)col * H;
+  const int kbase = (lane >> 4) * 8;
+  const int r0 = lane & 15, r1 = 16 + (lane & 15);
+  const int r0c = r0 < B ? r0 : 0, r1c = r1 < B ? r1 : 0;
+  const unsigned nblocks = gridDim.x;
+
+  // register-resident cell state: thread e owns elements e, e+256 of the
+  // block's (32 x 16) tile
+  float c_reg[2];
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    int e = tid + u * 256;
+    int row = e >> 4, hc = e & 15;
+    c_reg[u] = (row < B) ? c0[(int64_t)row * H + s16 + hc] : 0.0f;
+  }
+
+  for (int t = 0; t < T; ++t) {
+    const __bf16* h_bf = h_bfs + (int64_t)t * B * H;
+    const float* xp_t = xp + (int64_t)t * B * 4 * H;
+    f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+#pragma unroll 4
+    for (int kb = 0; kb < H; kb += 32) {
+      bf16x8_k5 bfrag = *reinterpret_cast<const bf16x8_k5*>(wrow + kb + kbase);
+      bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
+          h_bf + (int64_t)r0c * H + kb + kbase);
+      bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
+          h_bf + (int64_t)r1c * H + kb + kbase);
+      acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
+      acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
+    }
+    const int crow = (lane >> 4) * 4;
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row0 = crow + r, row1 = 16 + crow + r;
+      int row0c = row0 < B ? row0 : 0, row1c = row1 < B ? row1 : 0;
+      gbuf[wave][row0][lane & 15] =
+          acc0[r] + xp_t[(int64_t)row0c * 4 * H + col];
+      gbuf[wave][row1][lane & 15] =
+          acc1[r] + xp_t[(int64_t)row1c * 4 * H + col];
+    }
+    __syncthreads();
+    float* h_out = hs + (int64_t)(t + 1) * B * H;
+    float* c_out = cs + (int64_t)(t + 1) * B * H;
+    __bf16* h_bf_out = h_bfs + (int64_t)(t + 1) * B * H;
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int e = tid + u * 256;
+      int row = e >> 4, hc = e & 15;
+      if (row >= B) continue;
+      float i_ = sigm_f(gbuf[0][row][hc]);
+      float f_ = sigm_f(gbuf[1][row][hc]);
+      float g_ = tanhf(gbuf[2][row][hc]);
+      float o_ = sigm_f(gbuf[3][row][hc]);
+      int64_t hidx = (int64_t)row * H + s16 + hc;
+      float c = f_ * c_reg[u] + i_ * g_;
+      c_reg[u] = c;
+      float tc = tanhf(c);
+      float h = o_ * tc;
+      h_out[hidx] = h;
+      c_out[hidx] = c;
+      h_bf_out[hidx] = (__bf16)h;
+      tanhc[(int64_t)t * B * H + hidx] = tc;
+      float* a4 = acts + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + s16 + hc;
+      a4[0] = i_;
+      a4[H] = f_;
+      a4[2 * H] = g_;
+      a4[3 * H] = o_;
+    }
+    // agent-scope grid barrier: h_bf_out must be visible to every block
+    __syncthreads();
+    if (tid == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __hip_atomic_fetch_add((gu32_t*)ctr, 1u, DRL_RLX_AGENT);
+      const unsigned target = nblocks * (unsigned)(t + 1);
+      unsigned spins = 0;
+      while (__hip_atomic_load((gu32_t*)ctr, DRL_RLX_AGENT) < target) {
+        __builtin_amdgcn_s_sleep(4);
+        if (++spins > 5000000u) break;  // bounded: exit instead of wedging
+      }
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+  }
+}
+
+template <int H>
+__global__ __launch_bounds__(256) void lstm_seq_bwd_bf16_kernel(
+    const float* __restrict__ dh_init,  // (B, H)
+    const float* __restrict__ gout,     // (T, B, H)
+    const float* __restrict__ dc_T,     // (B, H)
+    const __bf16* __restrict__ w_t_bf,  // (H, 4H)
+    const float* __restrict__ acts,     // (T, B, 4H)
+    const float* __restrict__ tanhc,    // (T, B, H)
+    const float* __restrict__ cs,       // (T+1, B, H)
+    float* __restrict__ dgates,         // (T, B, 4H)
+    __bf16* __restrict__ dgates_bf,     // (T, B, 4H)
+    float* __restrict__ dc0_out,        // (B, H)
+    unsigned int* __restrict__ ctr,     // zeroed before launch
+    int B, int T) {
+  __shared__ float partial[4][32][16];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int n0 = blockIdx.x * 16;
+  const int kbase = (lane >> 4) * 8;
+  const int r0 = lane & 15, r1 = 16 + (lane & 15);
+  const int r0c = r0 < B ? r0 : 0, r1c = r1 < B ? r1 : 0;
+  const __bf16* wcol = w_t_bf + (int64_t)(n0 + (lane & 15)) * 4 * H;
+  const unsigned nblocks = gridDim.x;
+
+  float dc_reg[2];
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    int e = tid + u * 256;
+    int row = e >> 4, hc = e & 15;
+    dc_reg[u] = (row < B) ? dc_T[(int64_t)row * H + n0 + hc] : 0.0f;
+  }
+
+  for (int it = 0; it < T; ++it) {
+    const int t = T - 1 - it;
+    if (it > 0) {
+      const __bf16* dg_prev = dgates_bf + (int64_t)(t + 1) * B * 4 * H;
+      f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+      const int k0 = wave * H;
+#pragma unroll 4
+      for (int kb = 0; kb < H; kb += 32) {
+        bf16x8_k5 bfrag =
+            *reinterpret_cast<const bf16x8_k5*>(wcol + k0 + kb + kbase);
+        bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
+            dg_prev + (int64_t)r0c * 4 * H + k0 + kb + kbase);
+        bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
+            dg_prev + (int64_t)r1c * 4 * H + k0 + kb + kbase);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
+      }
+      const int crow = (lane >> 4) * 4;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        partial[wave][crow + r][lane & 15] = acc0[r];
+        partial[wave][16 + crow + r][lane & 15] = acc1[r];
+      }
+    }
+    __syncthreads();
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      int e = tid + u * 256;
+      int row = e >> 4, hc = e & 15;
+      if (row >= B) continue;
+      int64_t hidx = (int64_t)row * H + n0 + hc;
+      float dhv = (it > 0)
+          ? partial[0][row][hc] + partial[1][row][hc] + partial[2][row][hc] +
+                partial[3][row][hc]
+          : dh_init[hidx];
+      dhv += gout[(int64_t)t * B * H + hidx];
+      const float* a4 =
+          acts + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + n0 + hc;
+      float i = a4[0], f = a4[H], g = a4[2 * H], o = a4[3 * H];
+      float tc = tanhc[(int64_t)t * B * H + hidx];
+      float do_ = dhv * tc;
+      float dct = dc_reg[u] + dhv * o * (1.0f - tc * tc);
+      float di = dct * g;
+      float df = dct * cs[(int64_t)t * B * H + hidx];
+      float dg = dct * i;
+      dc_reg[u] = dct * f;
+      float v0 = di * i * (1.0f - i);
+      float v1 = df * f * (1.0f - f);
+      float v2 = dg * (1.0f - g * g);
+      float v3 = do_ * o * (1.0f - o);
+      float* d4 =
+          dgates + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + n0 + hc;
+      d4[0] = v0;
+      d4[H] = v1;
+      d4[2 * H] = v2;
+      d4[3 * H] = v3;
+      __bf16* b4 =
+          dgates_bf + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + n0 + hc;
+      b4[0] = (__bf16)v0;
+      b4[H] = (__bf16)v1;
+      b4[2 * H] = (__bf16)v2;
+      b4[3 * H] = (__bf16)v3;
+    }
+    __syncthreads();
+    if (tid == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __hip_atomic_fetch_add((gu32_t*)ctr, 1u, DRL_RLX_AGENT);
+      const unsigned target = nblocks * (unsigned)(it + 1);
+      unsigned spins = 0;
+      while (__hip_atomic_load((gu32_t*)ctr, DRL_RLX_AGENT) < target) {
+        __builtin_amdgcn_s_sleep(4);
+        if (++spins > 5000000u) break;
+      }
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+  }
+  // final dc (dL/dc0) back to global
+#pragma unroll
+  for (int u = 0; u < 2; ++u) {
+    int e = tid + u * 256;
+    int row = e >> 4, hc = e & 15;
+    if (row < B) dc0_out[(int64_t)row * H + n0 + hc] = dc_reg[u];
+  }
+}
+}  // namespace
+
+bool lstm_seq_fwd_bf16(torch::Tensor xp, torch::Tensor c0, torch::Tensor w_bf,
+                       torch::Tensor hs, torch::Tensor cs, torch::Tensor h_bfs,
+                       torch::Tensor acts, torch::Tensor tanhc,
+                       torch::Tensor ctr) {
+  int T = (int)xp.size(0), B = (int)xp.size(1), H = (int)c0.size(1);
+  if (B > kLstmMaxB || H != 512) return false;
+  constexpr int HH = 512;
+  hipLaunchKernelGGL(lstm_seq_fwd_bf16_kernel<HH>, dim3(HH / 16), dim3(256), 0,
+                     cur_stream(), xp.data_ptr<float>(), c0.data_ptr<float>(),
+                     (const __bf16*)w_bf.data_ptr(), hs.data_ptr<float>(),
+                     cs.data_ptr<float>(), (__bf16*)h_bfs.data_ptr(),
+                     acts.data_ptr<float>(), tanhc.data_ptr<float>(),
+                     (unsigned int*)ctr.data_ptr(), B, T);
+  return true;
+}
+
+bool lstm_seq_bwd_bf16(torch::Tensor dh_init, torch::Tensor gout,
+                       torch::Tensor dc_T, torch::Tensor w_t_bf,
+                       torch::Tensor acts, torch::Tensor tanhc,
+                       torch::Tensor cs, torch::Tensor dgates,
+                       torch::Tensor dgates_bf, torch::Tensor dc0_out,
+                       torch::Tensor ctr) {
+  int T = (int)gout.size(0), B = (int)gout.size(1), H = (int)gout.size(2);
+  if (B > kLstmMaxB || H != 512) return false;
+  constexpr int HH = 512;
+  hipLaunchKernelGGL(lstm_seq_bwd_bf16_kernel<HH>, dim3(HH / 16), dim3(256), 0,
+                     cur_stream(), dh_init.data_ptr<float>(),
+                     gout.data_ptr<float>(), dc_T.data_ptr<float>(),
+                     (const __bf16*)w_t_bf.data_ptr(), acts.data_ptr<float>(),
+                     tanhc.data_ptr<float>(), cs.data_ptr<float>(),
+                     dgates.data_ptr<float>(), (__bf16*)dgates_bf.data_ptr(),
+                     dc0_out.data_ptr<float>(),
+                     (unsigned int*)ctr.data_ptr(), B, T);
+  return true;
 }
