@@ -196,6 +196,15 @@ class ApexLearner(LearnerBase):
             return a_head(h[:, :hidden]), v_head(h[:, hidden:])
 
         self._fast_heads = fast_heads
+
+        def fast_hidden(x):
+            feat = cnn(x)
+            return F.relu(F.linear(feat.to(w1.dtype), w1, b1))
+
+        self._fast_hidden = fast_hidden
+        self._head_params = (a_head.weight, a_head.bias, v_head.weight,
+                             v_head.bias)
+        self._use_q_loss = ops.has_dueling_q_loss(hidden, self.cfg.action_size)
         # target-net twin over the flat target buffer (same pinned-front
         # ordering as the compute flat buffer, so the same offsets hold)
         t_cnn = self.target.nodes[cnn_node]
@@ -210,6 +219,14 @@ class ApexLearner(LearnerBase):
             return ta_head(h[:, :hidden]), tv_head(h[:, hidden:])
 
         self._target_heads = target_heads
+
+        def target_hidden(x):
+            feat = t_cnn(x)
+            return F.relu(F.linear(feat.to(w1t.dtype), w1t, b1t))
+
+        self._target_hidden = target_hidden
+        self._thead_params = (ta_head.weight, ta_head.bias, tv_head.weight,
+                              tv_head.bias)
 
     def _online_q(self, x):
         if getattr(self, "_fast_fwd", None) is not None:
@@ -450,7 +467,21 @@ class ApexLearner(LearnerBase):
         dones = data["done"].to(self.device)
         weights = weights.to(self.device)
 
-        if cuda and getattr(self, "_fast_heads", None) is not None:
+        if cuda and getattr(self, "_use_q_loss", False):
+            # deepest fusion: head projections + dueling + TD loss in ONE
+            # kernel; backward = closed-form dh + one head-grad reduction
+            h_s = self._fast_hidden(s)
+            with torch.no_grad():
+                h_on = self._fast_hidden(sp)
+                h_tg = self._target_hidden(sp)
+            wa, ba, wv, bv = self._head_params
+            wat, bat, wvt, bvt = self._thead_params
+            loss, prio, qmean = ops.dueling_q_head_loss(
+                h_s, wa, ba, wv, bv, h_on, h_tg, wat, bat, wvt, bvt,
+                actions, rewards, dones, weights, self.gamma, self.n_step,
+                self.alpha,
+            )
+        elif cuda and getattr(self, "_fast_heads", None) is not None:
             # fused whole-head path: dueling epilogues live inside the loss
             # kernel; backward writes (g_adv, g_val) closed-form
             adv_s, val_s = self._fast_heads(s)

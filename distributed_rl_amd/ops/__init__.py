@@ -348,6 +348,85 @@ def dueling_nstep_dqn_loss(adv_s, val_s, adv_on, val_on, adv_tg, val_tg,
 
 
 # ---------------------------------------------------------------------------
+# K3+K4+heads fused (round 2b): the dueling head projections join the loss
+# kernel; backward is closed-form dh + one reduction kernel for the head
+# weight/bias grads. Geometry-locked to hidden=512/A=6 (the shipped cfg).
+# ---------------------------------------------------------------------------
+
+
+def has_dueling_q_loss(hidden: int, actions: int) -> bool:
+    ext = hip_ext(required=False)
+    return (ext is not None and hasattr(ext, "dueling_q_loss_fwd")
+            and hidden == 512 and actions == 6)
+
+
+class _DuelingQLossFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, h_s, wa, ba, wv, bv, h_on, h_tg, wa_t, ba_t, wv_t, bv_t,
+                actions, rewards, dones, weights, gamma_n, alpha):
+        ext = hip_ext()
+        B = h_s.shape[0]
+        dev = h_s.device
+        acc2 = torch.zeros(2, dtype=torch.float32, device=dev)
+        prio = torch.empty(B, dtype=torch.float32, device=dev)
+        coef = torch.empty(B, dtype=torch.float32, device=dev)
+        ext.dueling_q_loss_fwd(
+            h_s.contiguous(), h_on.contiguous(), h_tg.contiguous(),
+            wa.contiguous(), ba.contiguous(), wv.reshape(-1).contiguous(),
+            bv.contiguous(), wa_t.contiguous(), ba_t.contiguous(),
+            wv_t.reshape(-1).contiguous(), bv_t.contiguous(),
+            actions.contiguous(), rewards.contiguous(), dones.contiguous(),
+            weights.contiguous(), float(gamma_n), float(alpha),
+            acc2[0:1], prio, coef, acc2[1:2],
+        )
+        ctx.save_for_backward(coef, actions, wa, wv, h_s)
+        ctx.mark_non_differentiable(prio)
+        return acc2[0], prio, acc2[1]
+
+    @staticmethod
+    def backward(ctx, gout, _gprio, _gqm):
+        coef, actions, wa, wv, h_s = ctx.saved_tensors
+        ext = hip_ext()
+        B = h_s.shape[0]
+        dev = h_s.device
+        dh = torch.empty(B, 1024, dtype=torch.bfloat16, device=dev)
+        dwa = torch.empty(6, 512, dtype=torch.bfloat16, device=dev)
+        dba = torch.empty(6, dtype=torch.bfloat16, device=dev)
+        dwv = torch.empty(1, 512, dtype=torch.bfloat16, device=dev)
+        dbv = torch.empty(1, dtype=torch.bfloat16, device=dev)
+        ext.dueling_q_loss_bwd(coef, actions, gout.reshape(1).contiguous(),
+                               wa.contiguous(), wv.reshape(-1).contiguous(),
+                               h_s.contiguous(), dh, dwa, dba, dwv, dbv)
+        return (dh, dwa, dba, dwv, dbv) + (None,) * 12
+
+
+def dueling_q_head_loss(h_s, wa, ba, wv, bv, h_on, h_tg, wa_t, ba_t, wv_t,
+                        bv_t, actions, rewards, dones, weights, gamma: float,
+                        n_step: int, alpha: float):
+    """Heads + dueling + n-step double-DQN loss fully fused.
+    h_*: (B, 1024) = [adv-stream | val-stream] post-ReLU hidden.
+    Only (h_s, wa, ba, wv, bv) are differentiable."""
+    if _use_hip(h_s):
+        return _DuelingQLossFn.apply(
+            h_s, wa, ba, wv, bv, h_on, h_tg, wa_t, ba_t, wv_t, bv_t,
+            actions.long(), rewards.float(), dones.float(), weights.float(),
+            gamma ** n_step, alpha,
+        )
+    import torch.nn.functional as F
+
+    def heads(h, wa_, ba_, wv_, bv_):
+        return (F.linear(h[:, :512].float(), wa_.float(), ba_.float()),
+                F.linear(h[:, 512:].float(), wv_.float(), bv_.float()))
+
+    adv_s, val_s = heads(h_s, wa, ba, wv, bv)
+    adv_on, val_on = heads(h_on, wa, ba, wv, bv)
+    adv_tg, val_tg = heads(h_tg, wa_t, ba_t, wv_t, bv_t)
+    return dueling_nstep_dqn_loss(adv_s, val_s, adv_on, val_on, adv_tg,
+                                  val_tg, actions, rewards, dones, weights,
+                                  gamma, n_step, alpha)
+
+
+# ---------------------------------------------------------------------------
 # K6 / K7
 # ---------------------------------------------------------------------------
 

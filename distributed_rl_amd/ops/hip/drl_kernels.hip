@@ -722,6 +722,20 @@ bool lstm_seq_bwd_bf16(torch::Tensor dh_init, torch::Tensor gout,
                        torch::Tensor cs, torch::Tensor dgates,
                        torch::Tensor dgates_bf, torch::Tensor dc0_out,
                        torch::Tensor ctr);
+void dueling_q_loss_fwd(torch::Tensor h_s, torch::Tensor h_on,
+                        torch::Tensor h_tg, torch::Tensor wa, torch::Tensor ba,
+                        torch::Tensor wv, torch::Tensor bv, torch::Tensor wa_t,
+                        torch::Tensor ba_t, torch::Tensor wv_t,
+                        torch::Tensor bv_t, torch::Tensor act,
+                        torch::Tensor rew, torch::Tensor done, torch::Tensor w,
+                        double gamma_n, double alpha, torch::Tensor loss_out,
+                        torch::Tensor prio_out, torch::Tensor grad_coef,
+                        torch::Tensor qmax_out);
+void dueling_q_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
+                        torch::Tensor gout, torch::Tensor wa, torch::Tensor wv,
+                        torch::Tensor h_s, torch::Tensor dh, torch::Tensor dwa,
+                        torch::Tensor dba, torch::Tensor dwv,
+                        torch::Tensor dbv);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -776,6 +790,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("lstm_seq_bwd_bf16", &lstm_seq_bwd_bf16,
         "persistent whole-sequence LSTM bwd, bf16-MFMA dh + grid barriers "
         "(K5 v3)");
+  m.def("dueling_q_loss_fwd", &dueling_q_loss_fwd,
+        "heads+dueling+n-step-DQN loss in one kernel (K3+K4+heads)");
+  m.def("dueling_q_loss_bwd", &dueling_q_loss_bwd,
+        "closed-form dh + head weight/bias grads (2 kernels)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -2159,4 +2177,238 @@ bool lstm_seq_bwd_bf16(torch::Tensor dh_init, torch::Tensor gout,
                      dc0_out.data_ptr<float>(),
                      (unsigned int*)ctr.data_ptr(), B, T);
   return true;
+}
+
+// ===========================================================================
+// K3+K4+heads fused (round 2b): the dueling HEAD projections join the loss.
+//   fwd: one wave per batch row computes adv_j = <h[:512], Wa_j> + ba_j and
+//        val = <h[512:], Wv> + bv for all THREE forwards (online s, online
+//        s', target s'), then dueling + n-step double-DQN TD inline.
+//   bwd: dh closed form  dh[b,k<512] = -coef_b*gout*(Wa[a_b,k] - mean_j Wa[j,k])
+//                        dh[b,512+k] = -coef_b*gout*Wv[k]
+//        and a reduction kernel for (dWa, dba, dWv, dbv).
+// Replaces the 6 head GEMM launches per step + the head-backward GEMM/
+// reduce chain (~11 launches -> 3). h layout: the fused (B, 1024) hidden
+// [adv-stream | val-stream] produced by the single w1 GEMM
+// (ApexLearner._build_fast_forward).
+// ===========================================================================
+namespace {
+
+// wave-level sum over 64 lanes
+__device__ __forceinline__ float wave_sum(float v) {
+  for (int off = 32; off > 0; off >>= 1) v += __shfl_down(v, off, 64);
+  return v;
+}
+
+template <int HH, int A>  // HH = per-stream hidden (512), A = actions (6)
+__global__ __launch_bounds__(256) void dueling_q_loss_fwd_kernel(
+    const __bf16* __restrict__ h_s,   // (B, 2*HH)
+    const __bf16* __restrict__ h_on,  // (B, 2*HH)
+    const __bf16* __restrict__ h_tg,  // (B, 2*HH)
+    const __bf16* __restrict__ wa, const __bf16* __restrict__ ba,   // online
+    const __bf16* __restrict__ wv, const __bf16* __restrict__ bv,
+    const __bf16* __restrict__ wa_t, const __bf16* __restrict__ ba_t,  // target
+    const __bf16* __restrict__ wv_t, const __bf16* __restrict__ bv_t,
+    const int64_t* __restrict__ act, const float* __restrict__ rew,
+    const float* __restrict__ done, const float* __restrict__ w, int B,
+    float gamma_n, float alpha, float* __restrict__ loss_out,
+    float* __restrict__ prio_out, float* __restrict__ grad_coef,
+    float* __restrict__ qmax_out) {
+  const int lane = threadIdx.x & 63;
+  const int b = blockIdx.x * 4 + (threadIdx.x >> 6);
+  if (b >= B) return;
+  const int k8 = lane * 8;  // this lane's 8-elem slice of each stream
+
+  // per-lane fragments of the three hidden rows
+  float ha_s[8], hv_s[8], ha_on[8], hv_on[8], ha_tg[8], hv_tg[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    ha_s[j] = (float)h_s[(int64_t)b * 2 * HH + k8 + j];
+    hv_s[j] = (float)h_s[(int64_t)b * 2 * HH + HH + k8 + j];
+    ha_on[j] = (float)h_on[(int64_t)b * 2 * HH + k8 + j];
+    hv_on[j] = (float)h_on[(int64_t)b * 2 * HH + HH + k8 + j];
+    ha_tg[j] = (float)h_tg[(int64_t)b * 2 * HH + k8 + j];
+    hv_tg[j] = (float)h_tg[(int64_t)b * 2 * HH + HH + k8 + j];
+  }
+  float q_s[A], q_on[A], q_tg[A];  // adv dots (bias added on lane 0 later)
+#pragma unroll
+  for (int j = 0; j < A; ++j) {
+    float pa_s = 0.f, pa_on = 0.f, pa_tg = 0.f;
+#pragma unroll
+    for (int u = 0; u < 8; ++u) {
+      const float wav = (float)wa[(int64_t)j * HH + k8 + u];
+      const float wat = (float)wa_t[(int64_t)j * HH + k8 + u];
+      pa_s += ha_s[u] * wav;
+      pa_on += ha_on[u] * wav;
+      pa_tg += ha_tg[u] * wat;
+    }
+    q_s[j] = wave_sum(pa_s);
+    q_on[j] = wave_sum(pa_on);
+    q_tg[j] = wave_sum(pa_tg);
+  }
+  float pv_s = 0.f, pv_on = 0.f, pv_tg = 0.f;
+#pragma unroll
+  for (int u = 0; u < 8; ++u) {
+    const float wvv = (float)wv[k8 + u];
+    const float wvt = (float)wv_t[k8 + u];
+    pv_s += hv_s[u] * wvv;
+    pv_on += hv_on[u] * wvv;
+    pv_tg += hv_tg[u] * wvt;
+  }
+  const float v_s = wave_sum(pv_s);
+  const float v_tg = wave_sum(pv_tg);
+  wave_sum(pv_on);  // unused (argmax needs adv only) — keep lanes converged
+
+  if (lane == 0) {
+    float mean_s = 0.f, mean_tg = 0.f, best = -1e30f;
+    int a_star = 0;
+#pragma unroll
+    for (int j = 0; j < A; ++j) {
+      const float adv_on = q_on[j] + (float)ba[j];
+      if (adv_on > best) { best = adv_on; a_star = j; }
+      q_s[j] += (float)ba[j];
+      q_tg[j] += (float)ba_t[j];
+      mean_s += q_s[j];
+      mean_tg += q_tg[j];
+    }
+    mean_s /= A;
+    mean_tg /= A;
+    const float vs_full = v_s + (float)bv[0];
+    const float vt_full = v_tg + (float)bv_t[0];
+    const float q_tgt = q_tg[a_star] - mean_tg + vt_full;
+    const float target = rew[b] + gamma_n * q_tgt * (1.0f - done[b]);
+    float smax = q_s[0];
+#pragma unroll
+    for (int j = 1; j < A; ++j) smax = fmaxf(smax, q_s[j]);
+    const float q = q_s[act[b]] - mean_s + vs_full;
+    const float raw = target - q;
+    const float td = fminf(1.0f, fmaxf(-1.0f, raw));
+    prio_out[b] = __powf(fabsf(td) + 1e-7f, alpha);
+    const float in_range = (raw > -1.0f && raw < 1.0f) ? 1.0f : 0.0f;
+    const float invB = 1.0f / B;
+    grad_coef[b] = w[b] * td * in_range * invB;
+    atomicAdd(loss_out, 0.5f * w[b] * td * td * invB);
+    atomicAdd(qmax_out, (smax - mean_s + vs_full) * invB);
+  }
+}
+
+template <int HH, int A>
+__global__ void dueling_q_loss_bwd_dh_kernel(
+    const float* __restrict__ grad_coef, const int64_t* __restrict__ act,
+    const float* __restrict__ gout, const __bf16* __restrict__ wa,
+    const __bf16* __restrict__ wv, int B, __bf16* __restrict__ dh) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= B * 2 * HH) return;
+  const int b = i / (2 * HH);
+  const int k = i - b * 2 * HH;
+  const float c = -grad_coef[b] * gout[0];
+  float v;
+  if (k < HH) {
+    float m = 0.f;
+#pragma unroll
+    for (int j = 0; j < A; ++j) m += (float)wa[(int64_t)j * HH + k];
+    v = c * ((float)wa[(int64_t)act[b] * HH + k] - m / A);
+  } else {
+    v = c * (float)wv[k - HH];
+  }
+  dh[i] = (__bf16)v;
+}
+
+// block j in [0, A]: j < A -> dWa row j + dba[j]; j == A -> dWv + dbv.
+template <int HH, int A>
+__global__ __launch_bounds__(256) void dueling_q_loss_bwd_dw_kernel(
+    const float* __restrict__ grad_coef, const int64_t* __restrict__ act,
+    const float* __restrict__ gout, const __bf16* __restrict__ h_s, int B,
+    __bf16* __restrict__ dwa, __bf16* __restrict__ dba,
+    __bf16* __restrict__ dwv, __bf16* __restrict__ dbv) {
+  const int j = blockIdx.x;
+  const int tid = threadIdx.x;
+  const float g = gout[0];
+  float accw[HH / 256];  // each thread owns HH/256 k-columns
+#pragma unroll
+  for (int u = 0; u < HH / 256; ++u) accw[u] = 0.f;
+  float accb = 0.f;
+  for (int b = 0; b < B; ++b) {
+    const float c = -grad_coef[b] * g;
+    float alpha_jb;
+    int64_t hoff;
+    if (j < A) {
+      alpha_jb = c * (((int)act[b] == j ? 1.0f : 0.0f) - 1.0f / A);
+      hoff = (int64_t)b * 2 * HH;
+    } else {
+      alpha_jb = c;
+      hoff = (int64_t)b * 2 * HH + HH;
+    }
+    accb += alpha_jb;
+#pragma unroll
+    for (int u = 0; u < HH / 256; ++u) {
+      const int k = u * 256 + tid;
+      accw[u] += alpha_jb * (float)h_s[hoff + k];
+    }
+  }
+#pragma unroll
+  for (int u = 0; u < HH / 256; ++u) {
+    const int k = u * 256 + tid;
+    if (j < A)
+      dwa[(int64_t)j * HH + k] = (__bf16)accw[u];
+    else
+      dwv[k] = (__bf16)accw[u];
+  }
+  // bias: wave-reduce accb? accb is identical across threads (independent of
+  // tid), so thread 0 just writes it
+  if (tid == 0) {
+    if (j < A)
+      dba[j] = (__bf16)accb;
+    else
+      dbv[0] = (__bf16)accb;
+  }
+}
+}  // namespace
+
+void dueling_q_loss_fwd(torch::Tensor h_s, torch::Tensor h_on,
+                        torch::Tensor h_tg, torch::Tensor wa, torch::Tensor ba,
+                        torch::Tensor wv, torch::Tensor bv, torch::Tensor wa_t,
+                        torch::Tensor ba_t, torch::Tensor wv_t,
+                        torch::Tensor bv_t, torch::Tensor act,
+                        torch::Tensor rew, torch::Tensor done, torch::Tensor w,
+                        double gamma_n, double alpha, torch::Tensor loss_out,
+                        torch::Tensor prio_out, torch::Tensor grad_coef,
+                        torch::Tensor qmax_out) {
+  int B = (int)h_s.size(0);
+  TORCH_CHECK(h_s.size(1) == 1024 && wa.size(0) == 6, "geometry: HH=512 A=6");
+  hipLaunchKernelGGL((dueling_q_loss_fwd_kernel<512, 6>),
+                     dim3((B + 3) / 4), dim3(256), 0, cur_stream(),
+                     (const __bf16*)h_s.data_ptr(),
+                     (const __bf16*)h_on.data_ptr(),
+                     (const __bf16*)h_tg.data_ptr(),
+                     (const __bf16*)wa.data_ptr(), (const __bf16*)ba.data_ptr(),
+                     (const __bf16*)wv.data_ptr(), (const __bf16*)bv.data_ptr(),
+                     (const __bf16*)wa_t.data_ptr(),
+                     (const __bf16*)ba_t.data_ptr(),
+                     (const __bf16*)wv_t.data_ptr(),
+                     (const __bf16*)bv_t.data_ptr(), act.data_ptr<int64_t>(),
+                     rew.data_ptr<float>(), done.data_ptr<float>(),
+                     w.data_ptr<float>(), B, (float)gamma_n, (float)alpha,
+                     loss_out.data_ptr<float>(), prio_out.data_ptr<float>(),
+                     grad_coef.data_ptr<float>(), qmax_out.data_ptr<float>());
+}
+
+void dueling_q_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
+                        torch::Tensor gout, torch::Tensor wa, torch::Tensor wv,
+                        torch::Tensor h_s, torch::Tensor dh, torch::Tensor dwa,
+                        torch::Tensor dba, torch::Tensor dwv,
+                        torch::Tensor dbv) {
+  int B = (int)h_s.size(0);
+  hipLaunchKernelGGL((dueling_q_loss_bwd_dh_kernel<512, 6>),
+                     dim3(ceil_div((int64_t)B * 1024, kBlock)), dim3(kBlock),
+                     0, cur_stream(), grad_coef.data_ptr<float>(),
+                     act.data_ptr<int64_t>(), gout.data_ptr<float>(),
+                     (const __bf16*)wa.data_ptr(), (const __bf16*)wv.data_ptr(),
+                     B, (__bf16*)dh.data_ptr());
+  hipLaunchKernelGGL((dueling_q_loss_bwd_dw_kernel<512, 6>), dim3(7),
+                     dim3(256), 0, cur_stream(), grad_coef.data_ptr<float>(),
+                     act.data_ptr<int64_t>(), gout.data_ptr<float>(),
+                     (const __bf16*)h_s.data_ptr(), B,
+                     (__bf16*)dwa.data_ptr(), (__bf16*)dba.data_ptr(),
+                     (__bf16*)dwv.data_ptr(), (__bf16*)dbv.data_ptr());
 }
