@@ -409,53 +409,61 @@ __global__ __launch_bounds__(256) void conv_wrw_kernel(
 #pragma unroll
   for (int cb = 0; cb < CBLK; ++cb) acc[cb] = {0.f, 0.f, 0.f, 0.f};
 
-  // staging assignment: 8-elem pieces, plain row-major destination
-  const int sm = tid >> 3;            // sample row 0..31
-  const int sk8 = (tid & 7) * 8;      // kelem piece base 0..56
+  // staging assignment (round 2, ROUND2_PLAN item 3): the thread halves
+  // stage a_t and g_t CONCURRENTLY — threads 0-127 load 16-elem pieces of
+  // the im2col chunk (16 B/lane for u8 input, 2x16 B for bf16; was
+  // 8 B/lane with all 256 threads serialized behind g_t), threads 128-255
+  // stage gout. A 16-piece never crosses a patch row: ROWC % 16 == 0 for
+  // every geometry.
+  static_assert(G::ROWC % 16 == 0, "16-elem staging piece crosses a row");
+  const int sm = tid >> 2;             // sample row 0..31 (tid < 128)
+  const int sk16 = (tid & 3) * 16;     // kelem piece base 0..48
 
   for (int mc = mb; mc < n_chunks; mc += mblocks) {
     const int m0 = mc * 32;
-    // ---- stage im2col chunk -> a_t[m][kelem] (coalesced b128 writes)
-    {
+    if (tid < 128) {
+      // ---- stage im2col chunk -> a_t[m][kelem]
       const int m = m0 + sm;
       const int mcl = m < M ? m : M - 1;
       const int n = mcl / (G::P * G::Q);
       const int rem = mcl - n * (G::P * G::Q);
       const int p = rem / G::Q;
       const int q = rem - p * G::Q;
-      const int kelem = ktile * 64 + sk8;
+      const int kelem = ktile * 64 + sk16;
       const int dy = kelem / G::ROWC;
       const int dx = kelem - dy * G::ROWC;
       const int64_t goff =
           ((int64_t)n * H + p * S + dy) * (W * C) + q * S * C + dx;
-      bf16x8 v;
+      bf16x8 v0, v1;
       if constexpr (U8IN) {
         const uint8_t* src = reinterpret_cast<const uint8_t*>(in_v) + goff;
-        uint2 raw = *reinterpret_cast<const uint2*>(src);
+        uint4 raw = *reinterpret_cast<const uint4*>(src);
         const float inv255 = 1.0f / 255.0f;
+        const unsigned rw[4] = {raw.x, raw.y, raw.z, raw.w};
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          unsigned byte =
-              (j < 4 ? raw.x >> (8 * j) : raw.y >> (8 * (j - 4))) & 0xFF;
-          v[j] = (__bf16)(byte * inv255);
+          v0[j] = (__bf16)(((rw[j >> 2] >> (8 * (j & 3))) & 0xFF) * inv255);
+          v1[j] = (__bf16)(((rw[2 + (j >> 2)] >> (8 * (j & 3))) & 0xFF)
+                           * inv255);
         }
       } else {
-        v = *reinterpret_cast<const bf16x8*>(
-            reinterpret_cast<const __bf16*>(in_v) + goff);
+        const __bf16* src = reinterpret_cast<const __bf16*>(in_v) + goff;
+        v0 = *reinterpret_cast<const bf16x8*>(src);
+        v1 = *reinterpret_cast<const bf16x8*>(src + 8);
       }
       if (m >= M) {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) v[j] = (__bf16)0.0f;
+        for (int j = 0; j < 8; ++j) {
+          v0[j] = (__bf16)0.0f;
+          v1[j] = (__bf16)0.0f;
+        }
       }
-      *reinterpret_cast<bf16x8*>(a_t + sm * ASTRIDE + sk8) = v;
-    }
-    // ---- stage gout chunk -> g_t[m][cout] (coalesced, zero-padded);
-    // ktile-0 blocks also fold the bias grad into per-thread registers
-    // (one atomic per thread at kernel end — per-chunk atomics serialized
-    // badly on the small gb array, profiles/)
-    {
+      *reinterpret_cast<bf16x8*>(a_t + sm * ASTRIDE + sk16) = v0;
+      *reinterpret_cast<bf16x8*>(a_t + sm * ASTRIDE + sk16 + 8) = v1;
+    } else {
+      // ---- stage gout chunk -> g_t[m][cout] (coalesced, zero-padded)
       constexpr int PIECES = COUT / 8;
-      for (int piece = tid; piece < 32 * PIECES; piece += 256) {
+      for (int piece = tid - 128; piece < 32 * PIECES; piece += 128) {
         const int m = m0 + piece / PIECES;
         const int c0 = (piece % PIECES) * 8;
         bf16x8 v;
