@@ -111,7 +111,16 @@ class CNN2D(nn.Module):
             from .. import ops as _ops
 
             y = x
-            for conv in self._convs:
+            last = len(self._convs) - 1
+            for i, conv in enumerate(self._convs):
+                # last conv of a flattening stack writes the 2-D CHW-ordered
+                # output directly (flatten fused into the epilogue)
+                if i == last and self.flatten and _ops.conv_chw_supported(
+                        y.shape[2], y.shape[3], y.shape[1],
+                        conv.weight.shape[2], conv.weight.shape[3],
+                        conv.stride[0], conv.weight.shape[0]):
+                    return _ops.fused_conv_relu(y, conv.weight, conv.bias,
+                                                conv.stride[0], chw_out=True)
                 y = _ops.fused_conv_relu(y, conv.weight, conv.bias,
                                          conv.stride[0])
         else:

@@ -98,36 +98,54 @@ def conv_supported(H, W, C, KH, KW, S, COUT, u8: bool) -> bool:
 
 class _FusedConvFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, x, weight, bias, stride):
+    def forward(ctx, x, weight, bias, stride, chw_out=False):
         ext = hip_ext()
         N, C, H, W = x.shape
         COUT, _, KH, KW = weight.shape
         P = (H - KH) // stride + 1
         Q = (W - KW) // stride + 1
-        out = torch.empty(
-            N, COUT, P, Q, dtype=torch.bfloat16, device=x.device,
-            memory_format=torch.channels_last,
-        )
-        ext.conv_fwd(
-            x, weight,
-            bias if bias is not None else torch.empty(0, device=x.device),
-            out, stride,
-        )
+        b = bias if bias is not None else torch.empty(0, device=x.device)
+        if chw_out:
+            # flatten fused into the epilogue: 2-D (N, COUT*P*Q) out in
+            # logical-NCHW element order (= torch.flatten of the 4-D out)
+            out = torch.empty(N, COUT * P * Q, dtype=torch.bfloat16,
+                              device=x.device)
+            ext.conv_fwd_chw(x, weight, b, out, stride)
+        else:
+            out = torch.empty(
+                N, COUT, P, Q, dtype=torch.bfloat16, device=x.device,
+                memory_format=torch.channels_last,
+            )
+            ext.conv_fwd(x, weight, b, out, stride)
         ctx.save_for_backward(x, weight, out)
         ctx.stride = stride
         ctx.has_bias = bias is not None
+        ctx.chw = chw_out
+        ctx.pq = (P, Q)
         return out
 
     @staticmethod
     def backward(ctx, gout):
         x, weight, out = ctx.saved_tensors
         stride = ctx.stride
-        if not gout.is_contiguous(memory_format=torch.channels_last):
-            gout = gout.contiguous(memory_format=torch.channels_last)
-        # fused ReLU mask (one kernel instead of compare+mul)
-        masked = torch.empty_like(gout)
-        hip_ext().relu_mask_bwd(gout, out, masked)
-        gout = masked
+        COUT = weight.shape[0]
+        if ctx.chw:
+            # one kernel: relu-mask + CHW->NHWC transpose
+            P, Q = ctx.pq
+            N = gout.shape[0]
+            masked = torch.empty(N, COUT, P, Q, dtype=torch.bfloat16,
+                                 device=gout.device,
+                                 memory_format=torch.channels_last)
+            hip_ext().relu_mask_bwd_chw(gout.contiguous(), out, masked,
+                                        COUT, P * Q)
+            gout = masked
+        else:
+            if not gout.is_contiguous(memory_format=torch.channels_last):
+                gout = gout.contiguous(memory_format=torch.channels_last)
+            # fused ReLU mask (one kernel instead of compare+mul)
+            masked = torch.empty_like(gout)
+            hip_ext().relu_mask_bwd(gout, out, masked)
+            gout = masked
         need_x = ctx.needs_input_grad[0]
         need_w = ctx.needs_input_grad[1]
         need_b = ctx.has_bias and ctx.needs_input_grad[2]
@@ -184,12 +202,20 @@ class _FusedConvFn(torch.autograd.Function):
                 gi = gi2
             if gw is None:
                 gw, gb = gw2, gb2
-        return (gi if need_x else None), gw, gb, None
+        return (gi if need_x else None), gw, gb, None, None
 
 
-def fused_conv_relu(x, weight, bias, stride: int):
-    """One fused conv+bias+ReLU layer (u8 or bf16 channels_last input)."""
-    return _FusedConvFn.apply(x, weight, bias, stride)
+def fused_conv_relu(x, weight, bias, stride: int, chw_out: bool = False):
+    """One fused conv+bias+ReLU layer (u8 or bf16 channels_last input).
+    chw_out=True additionally fuses the trailing flatten into the epilogue
+    (returns 2-D (N, COUT*P*Q) in logical-NCHW order)."""
+    return _FusedConvFn.apply(x, weight, bias, stride, chw_out)
+
+
+def conv_chw_supported(H, W, C, KH, KW, S, COUT) -> bool:
+    ext = hip_ext(required=False)
+    return ext is not None and hasattr(ext, "conv_fwd_chw_supported") and \
+        bool(ext.conv_fwd_chw_supported(H, W, C, KH, KW, S, COUT))
 
 
 # ---------------------------------------------------------------------------

@@ -239,3 +239,35 @@ def test_conv_dgrad_in_fused_backward(ext):
     out2.backward(g.float())
     assert torch.allclose(gx.float(), xf.grad, atol=5e-2, rtol=5e-2), \
         (gx.float() - xf.grad).abs().max()
+
+
+def test_conv_chw_out_matches_flatten(ext):
+    """CHW-out epilogue (flatten fused) must equal conv_fwd + torch.flatten,
+    forward and backward (mask+transpose kernel)."""
+    from distributed_rl_amd.ops import fused_conv_relu
+
+    torch.manual_seed(9)
+    N, C, H, W, COUT, KH, S = 16, 64, 9, 9, 64, 3, 1
+    x = (torch.randn(N, C, H, W, device=DEV) * 0.5).to(torch.bfloat16
+        ).contiguous(memory_format=torch.channels_last).requires_grad_(True)
+    x2 = x.detach().clone().requires_grad_(True)
+    w = ((torch.randn(COUT, C, KH, KH, device=DEV) * 0.05).to(torch.bfloat16)
+         .contiguous(memory_format=torch.channels_last).requires_grad_(True))
+    w2 = w.detach().clone().requires_grad_(True)
+    b = torch.randn(COUT, device=DEV).to(torch.bfloat16).requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+
+    y_chw = fused_conv_relu(x, w, b, S, chw_out=True)
+    assert y_chw.dim() == 2
+    y_ref = torch.flatten(fused_conv_relu(x2, w2, b2, S), 1)
+    assert torch.equal(y_chw, y_ref)
+
+    g = torch.randn_like(y_ref)
+    y_chw.backward(g)
+    y_ref.backward(g)
+    torch.cuda.synchronize()
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=1e-2), \
+        (x.grad.float() - x2.grad.float()).abs().max()
+    assert torch.allclose(w.grad.float(), w2.grad.float(), atol=1e-2,
+                          rtol=1e-2)
+    assert torch.allclose(b.grad.float(), b2.grad.float(), atol=1e-2)
