@@ -171,6 +171,9 @@ def main():
     ap.add_argument("--transport-dir", default=None)
     args = ap.parse_args()
 
+    # cap intra-op threads: the default (ncores) oversubscribes the
+    # box against the actor fleet and the ingest pack
+    torch.set_num_threads(min(16, os.cpu_count() or 16))
     rank, local_rank, world = init_distributed()
     has_cuda = torch.cuda.is_available()
     device = f"cuda:{local_rank}" if has_cuda else "cpu"

@@ -236,19 +236,36 @@ class R2D2Learner(LearnerBase):
         with torch.no_grad():
             q_tgt_full = self._seq_forward(self.target, frames, h0,
                                            burn_in_split=False)
-        # Double-DQN argmax is only consulted at steps t+n >= burn_in + 1,
-        # all inside the training window, so the burn-in rows of the online
-        # view are never read — pad them with (detached) target rows instead
-        # of paying a third sequence pass.
-        q_online_full = torch.cat([q_tgt_full[: self.burn_in], q_train], dim=0)
-
-        td, q_taken, targets = nstep_recurrent_targets(
-            q_online_full, q_tgt_full, actions, rewards, done, self.burn_in,
-            self.n_step, self.gamma, self.use_rescaling,
-        )
-        # loss: IS-weighted 0.5 * mean_t(td^2) per sequence
-        loss = 0.5 * (weights * td.pow(2).mean(dim=0)).mean()
-        prio = ops.sequence_priority(td.detach().abs(), self.alpha, ETA)
+        if self.device.type == "cuda" and ops.has_r2d2_seq_loss():
+            # fused sequence loss: per-(t,b) truncated n-step targets +
+            # rescale + double-DQN argmax + IS loss + eta-mix priority in 3
+            # kernels (replaces the ~30-launch torch chain incl. an fp64
+            # prefix sum)
+            loss, prio, value, td_abs = ops.r2d2_sequence_loss(
+                q_train.float(), q_tgt_full.float(), actions, rewards, done,
+                weights, self.burn_in, self.n_step, self.gamma, self.alpha,
+                ETA, self.use_rescaling,
+            )
+            stats = {"loss": loss.detach(), "value": value, "td_abs": td_abs}
+        else:
+            # Double-DQN argmax is only consulted at steps t+n >= burn_in+1,
+            # all inside the training window, so the burn-in rows of the
+            # online view are never read — pad them with (detached) target
+            # rows instead of paying a third sequence pass.
+            q_online_full = torch.cat([q_tgt_full[: self.burn_in], q_train],
+                                      dim=0)
+            td, q_taken, targets = nstep_recurrent_targets(
+                q_online_full, q_tgt_full, actions, rewards, done,
+                self.burn_in, self.n_step, self.gamma, self.use_rescaling,
+            )
+            # loss: IS-weighted 0.5 * mean_t(td^2) per sequence
+            loss = 0.5 * (weights * td.pow(2).mean(dim=0)).mean()
+            prio = ops.sequence_priority(td.detach().abs(), self.alpha, ETA)
+            stats = {
+                "loss": loss.detach(),
+                "value": q_taken.detach().mean(),
+                "td_abs": td.detach().abs().mean(),
+            }
 
         if self.mp is not None:
             self.mp.zero_grads()
@@ -256,11 +273,6 @@ class R2D2Learner(LearnerBase):
         else:
             self.optim.zero_grad(set_to_none=False)
             loss.backward()
-        stats = {
-            "loss": loss.detach(),
-            "value": q_taken.detach().mean(),
-            "td_abs": td.detach().abs().mean(),
-        }
         return stats, prio
 
     def _optimize_mp(self):

@@ -426,6 +426,56 @@ class _DuelingQLossFn(torch.autograd.Function):
         return (dh, dwa, dba, dwv, dbv) + (None,) * 12
 
 
+def has_r2d2_seq_loss() -> bool:
+    ext = hip_ext(required=False)
+    return ext is not None and hasattr(ext, "r2d2_loss_fwd")
+
+
+class _R2D2SeqLossFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, q_train, q_tgt, actions, rewards, done, weights,
+                burn_in, n_step, gamma, alpha, eta, use_rescaling):
+        ext = hip_ext()
+        T, B, A = q_tgt.shape
+        W = T - 1 - burn_in
+        dev = q_tgt.device
+        td = torch.empty(W, B, dtype=torch.float32, device=dev)
+        stats = torch.zeros(3, dtype=torch.float32, device=dev)
+        ext.r2d2_loss_fwd(q_train, q_tgt, actions, rewards, done, weights,
+                          burn_in, n_step, float(gamma), bool(use_rescaling),
+                          td, stats)
+        prio = torch.empty(B, dtype=torch.float32, device=dev)
+        ext.r2d2_prio(td, float(alpha), float(eta), prio)
+        ctx.save_for_backward(td, actions, weights)
+        ctx.meta = (T, burn_in, tuple(q_train.shape))
+        ctx.mark_non_differentiable(prio)
+        return stats[0], prio, stats[1], stats[2]
+
+    @staticmethod
+    def backward(ctx, gout, _gp, _gv, _gt):
+        td, actions, weights = ctx.saved_tensors
+        T, m, qshape = ctx.meta
+        dq = torch.empty(qshape, dtype=torch.float32, device=td.device)
+        hip_ext().r2d2_loss_bwd(td, actions, weights,
+                                gout.reshape(1).contiguous(), T, m, dq)
+        return (dq,) + (None,) * 11
+
+
+def r2d2_sequence_loss(q_train, q_tgt, actions, rewards, done, weights,
+                       burn_in: int, n_step: int, gamma: float, alpha: float,
+                       eta: float, use_rescaling: bool):
+    """Fused R2D2 target construction + IS loss + eta-mix priority.
+    q_train: (T-burn_in, B, A) fp32 differentiable; q_tgt: (T, B, A) fp32;
+    actions: (T, B) int32; rewards (T, B); done/weights (B).
+    Returns (loss, prio, value_stat, td_abs_stat)."""
+    return _R2D2SeqLossFn.apply(
+        q_train.contiguous(), q_tgt.contiguous(),
+        actions.to(torch.int32).contiguous(), rewards.float().contiguous(),
+        done.float().contiguous(), weights.float().contiguous(),
+        burn_in, n_step, gamma, alpha, eta, use_rescaling,
+    )
+
+
 def dueling_q_head_loss(h_s, wa, ba, wv, bv, h_on, h_tg, wa_t, ba_t, wv_t,
                         bv_t, actions, rewards, dones, weights, gamma: float,
                         n_step: int, alpha: float):

@@ -736,6 +736,13 @@ void dueling_q_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
                         torch::Tensor h_s, torch::Tensor dh, torch::Tensor dwa,
                         torch::Tensor dba, torch::Tensor dwv,
                         torch::Tensor dbv);
+void r2d2_loss_fwd(torch::Tensor q_train, torch::Tensor q_tgt,
+                   torch::Tensor act, torch::Tensor rew, torch::Tensor done,
+                   torch::Tensor w, int64_t m, int64_t n_step, double gamma,
+                   bool rescale, torch::Tensor td_out, torch::Tensor stats);
+void r2d2_prio(torch::Tensor td, double alpha, double eta, torch::Tensor prio);
+void r2d2_loss_bwd(torch::Tensor td, torch::Tensor act, torch::Tensor w,
+                   torch::Tensor gout, int64_t T, int64_t m, torch::Tensor dq);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -794,6 +801,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "heads+dueling+n-step-DQN loss in one kernel (K3+K4+heads)");
   m.def("dueling_q_loss_bwd", &dueling_q_loss_bwd,
         "closed-form dh + head weight/bias grads (2 kernels)");
+  m.def("r2d2_loss_fwd", &r2d2_loss_fwd,
+        "R2D2 n-step targets + rescale + IS loss, one kernel (K4/K6 seq)");
+  m.def("r2d2_prio", &r2d2_prio, "eta-mix sequence priority (K7)");
+  m.def("r2d2_loss_bwd", &r2d2_loss_bwd,
+        "R2D2 loss backward: closed-form dq_train scatter");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv

@@ -58,7 +58,11 @@ class ReplayBase:
         if n == 0:
             return
         if n > self.capacity:
-            raise ValueError("push larger than capacity")
+            # oversized batch (e.g. a big ingest drain into a small ring):
+            # only the LAST `capacity` rows survive a ring write anyway
+            columns = {k: v[n - self.capacity :] for k, v in columns.items()}
+            priorities = priorities[n - self.capacity :]
+            n = self.capacity
         idx = self._ring_indices(n)
         self._write_columns(columns, idx)
         self._set_priorities(idx, priorities.to(self.device, torch.float32))

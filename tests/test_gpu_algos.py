@@ -286,3 +286,42 @@ def test_lstm_step_bf16_kernels_match_composition():
     ], dim=1)
     assert torch.allclose(dgates, d_ref, atol=2e-3), (dgates - d_ref).abs().max()
     assert torch.allclose(dc_out, dct * f, atol=2e-3)
+
+
+def test_r2d2_fused_seq_loss_matches_torch():
+    """Fused R2D2 target/loss/priority kernels vs the torch composition
+    (nstep_recurrent_targets + IS loss + sequence_priority) on identical
+    inputs — values, priorities and the q_train gradient."""
+    from distributed_rl_amd import ops
+    from distributed_rl_amd.algos.r2d2 import ETA, nstep_recurrent_targets
+
+    torch.manual_seed(3)
+    T, B, A, m, n_step, gamma = 16, 8, 6, 4, 5, 0.997
+    q_train = (torch.randn(T - m, B, A, device=DEV) * 2).requires_grad_(True)
+    q_train2 = q_train.detach().clone().requires_grad_(True)
+    q_tgt = torch.randn(T, B, A, device=DEV) * 2
+    actions = torch.randint(0, A, (T, B), device=DEV, dtype=torch.int32)
+    rewards = torch.randn(T, B, device=DEV)
+    done = (torch.rand(B, device=DEV) < 0.3).float()
+    weights = torch.rand(B, device=DEV) + 0.5
+
+    loss, prio, value, td_abs = ops.r2d2_sequence_loss(
+        q_train, q_tgt, actions, rewards, done, weights, m, n_step, gamma,
+        0.9, ETA, True)
+    loss.backward()
+
+    q_full = torch.cat([q_tgt[:m], q_train2], dim=0)
+    td, q_taken, _ = nstep_recurrent_targets(
+        q_full, q_tgt, actions.long(), rewards, done, m, n_step, gamma, True)
+    loss2 = 0.5 * (weights * td.pow(2).mean(dim=0)).mean()
+    prio2 = ops.sequence_priority(td.detach().abs(), 0.9, ETA)
+    loss2.backward()
+    torch.cuda.synchronize()
+
+    assert torch.allclose(loss, loss2, atol=1e-4, rtol=1e-4), \
+        (float(loss), float(loss2))
+    assert torch.allclose(prio, prio2, atol=1e-4, rtol=1e-3)
+    assert torch.allclose(value, q_taken.detach().mean(), atol=1e-4)
+    assert torch.allclose(td_abs, td.detach().abs().mean(), atol=1e-4)
+    assert torch.allclose(q_train.grad, q_train2.grad, atol=1e-5), \
+        (q_train.grad - q_train2.grad).abs().max()

@@ -45,6 +45,9 @@ def main():
     from distributed_rl_amd.algos.ape_x import ApexLearner
     from distributed_rl_amd.config import Config, load_config
 
+    # cap intra-op threads: the default (ncores) oversubscribes the
+    # box against the actor fleet and the ingest pack
+    torch.set_num_threads(min(16, os.cpu_count() or 16))
     torch.manual_seed(0)
     device = args.device or ("cuda:0" if torch.cuda.is_available() else "cpu")
     raw = copy.deepcopy(load_config("ape_x").raw)
