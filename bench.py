@@ -281,6 +281,10 @@ def main():
     elapsed = time.perf_counter() - t0
     ingested = (learner.ingested_total - ingested0) if args.with_actors else 0
     if args.with_actors > 0:
+        drops = (learner.transport.total_drops()
+                 if hasattr(learner.transport, "total_drops") else -1)
+        print(f"# transport: {ingested} rows ingested in timed region, "
+              f"{drops} dropped (ring backpressure) total", file=sys.stderr)
         learner.stop_ingest_thread()
     if fleet is not None:
         fleet.stop()

@@ -440,6 +440,11 @@ class LearnerEndpoint:
     def drain_rewards(self):
         return self.session.drain_rewards()
 
+    def total_drops(self) -> int:
+        """Backpressure telemetry: rows the actors pushed that the rings
+        had no space for (SPSC drop counters)."""
+        return sum(self.session.ring(i).drops for i in self.ring_ids)
+
 
 # ---------------------------------------------------------------------------
 # In-process pipe (tests / single-process integration)
