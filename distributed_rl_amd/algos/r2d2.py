@@ -232,10 +232,28 @@ class R2D2Learner(LearnerBase):
         weights = weights.to(self.device)
         h0 = self._h0_to_state(data["h0"].to(self.device))
 
-        q_train = self._seq_forward(self.net, frames, h0, burn_in_split=True)
-        with torch.no_grad():
-            q_tgt_full = self._seq_forward(self.target, frames, h0,
-                                           burn_in_split=False)
+        if self.device.type == "cuda":
+            # run the TARGET sequence pass concurrently on a side stream:
+            # both recurrences are ~32-block persistent/per-step kernels
+            # that underfill the 256-CU chip alone, and they are fully
+            # independent until the loss. Fork/join records events, so the
+            # same structure capture-records as a parallel hipGraph branch.
+            if not hasattr(self, "_tgt_stream"):
+                self._tgt_stream = torch.cuda.Stream(self.device)
+            cur = torch.cuda.current_stream(self.device)
+            self._tgt_stream.wait_stream(cur)
+            with torch.cuda.stream(self._tgt_stream), torch.no_grad():
+                q_tgt_full = self._seq_forward(self.target, frames, h0,
+                                               burn_in_split=False)
+            q_train = self._seq_forward(self.net, frames, h0,
+                                        burn_in_split=True)
+            cur.wait_stream(self._tgt_stream)
+        else:
+            q_train = self._seq_forward(self.net, frames, h0,
+                                        burn_in_split=True)
+            with torch.no_grad():
+                q_tgt_full = self._seq_forward(self.target, frames, h0,
+                                               burn_in_split=False)
         if self.device.type == "cuda" and ops.has_r2d2_seq_loss():
             # fused sequence loss: per-(t,b) truncated n-step targets +
             # rescale + double-DQN argmax + IS loss + eta-mix priority in 3

@@ -172,10 +172,18 @@ class _FusedConvFn(torch.autograd.Function):
             if need_b:
                 gb = gb_ws.to(torch.bfloat16)
         gi = None
-        use_own_dgrad = need_x and x.dtype == torch.bfloat16 and bool(
-            ext.conv_dgrad_supported(x.shape[2], x.shape[3], C, KH, KW,
-                                     stride, COUT)
-        ) if hasattr(ext, "conv_dgrad_supported") else False
+        # Hand-written dgrad exists and is oracle-tested, but MEASURED
+        # SLOWER than MIOpen's igemm at these geometries (tools/
+        # gpu_dgrad_bench.py: 133 vs 28 us at 20x20 s2, 42 vs 33 at 9x9
+        # s1) — dispatch keeps the faster library kernel; DRL_OWN_DGRAD=1
+        # forces ours (profiles/r02 notes).
+        use_own_dgrad = (
+            os.environ.get("DRL_OWN_DGRAD", "0") == "1"
+            and need_x and x.dtype == torch.bfloat16
+            and hasattr(ext, "conv_dgrad_supported")
+            and bool(ext.conv_dgrad_supported(x.shape[2], x.shape[3], C, KH,
+                                              KW, stride, COUT))
+        )
         if use_own_dgrad:
             # hand-written MFMA dgrad (conv_mfma.hip): masked-tap gather
             # against a per-step transposed weight copy — no MIOpen on the

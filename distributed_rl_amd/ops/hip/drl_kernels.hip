@@ -2326,21 +2326,23 @@ __global__ void dueling_q_loss_bwd_dh_kernel(
   dh[i] = (__bf16)v;
 }
 
-// block j in [0, A]: j < A -> dWa row j + dba[j]; j == A -> dWv + dbv.
-template <int HH, int A>
+// phase 1: grid (A+1 head-rows, BS b-slices); block (j, s) accumulates its
+// b-slice's partial dW row (+ bias partial) into the fp32 workspace
+// ws[(A+1) rows x (HH+1)] x BS. phase 2 folds the BS partials and casts.
+template <int HH, int A, int BS>
 __global__ __launch_bounds__(256) void dueling_q_loss_bwd_dw_kernel(
     const float* __restrict__ grad_coef, const int64_t* __restrict__ act,
     const float* __restrict__ gout, const __bf16* __restrict__ h_s, int B,
-    __bf16* __restrict__ dwa, __bf16* __restrict__ dba,
-    __bf16* __restrict__ dwv, __bf16* __restrict__ dbv) {
+    float* __restrict__ ws) {  // (BS, A+1, HH+1)
   const int j = blockIdx.x;
+  const int sl = blockIdx.y;
   const int tid = threadIdx.x;
   const float g = gout[0];
-  float accw[HH / 256];  // each thread owns HH/256 k-columns
+  float accw[HH / 256];
 #pragma unroll
   for (int u = 0; u < HH / 256; ++u) accw[u] = 0.f;
   float accb = 0.f;
-  for (int b = 0; b < B; ++b) {
+  for (int b = sl; b < B; b += BS) {
     const float c = -grad_coef[b] * g;
     float alpha_jb;
     int64_t hoff;
@@ -2358,21 +2360,35 @@ __global__ __launch_bounds__(256) void dueling_q_loss_bwd_dw_kernel(
       accw[u] += alpha_jb * (float)h_s[hoff + k];
     }
   }
+  float* row = ws + ((int64_t)sl * (A + 1) + j) * (HH + 1);
 #pragma unroll
-  for (int u = 0; u < HH / 256; ++u) {
-    const int k = u * 256 + tid;
+  for (int u = 0; u < HH / 256; ++u) row[u * 256 + tid] = accw[u];
+  if (tid == 0) row[HH] = accb;
+}
+
+template <int HH, int A, int BS>
+__global__ void dueling_q_loss_bwd_dw_fold_kernel(
+    const float* __restrict__ ws, __bf16* __restrict__ dwa,
+    __bf16* __restrict__ dba, __bf16* __restrict__ dwv,
+    __bf16* __restrict__ dbv) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  constexpr int ROW = HH + 1;
+  if (i >= (A + 1) * ROW) return;
+  const int j = i / ROW;
+  const int k = i - j * ROW;
+  float v = 0.f;
+#pragma unroll
+  for (int s = 0; s < BS; ++s) v += ws[((int64_t)s * (A + 1) + j) * ROW + k];
+  if (k < HH) {
     if (j < A)
-      dwa[(int64_t)j * HH + k] = (__bf16)accw[u];
+      dwa[(int64_t)j * HH + k] = (__bf16)v;
     else
-      dwv[k] = (__bf16)accw[u];
-  }
-  // bias: wave-reduce accb? accb is identical across threads (independent of
-  // tid), so thread 0 just writes it
-  if (tid == 0) {
+      dwv[k] = (__bf16)v;
+  } else {
     if (j < A)
-      dba[j] = (__bf16)accb;
+      dba[j] = (__bf16)v;
     else
-      dbv[0] = (__bf16)accb;
+      dbv[0] = (__bf16)v;
   }
 }
 }  // namespace
@@ -2417,10 +2433,17 @@ void dueling_q_loss_bwd(torch::Tensor grad_coef, torch::Tensor act,
                      act.data_ptr<int64_t>(), gout.data_ptr<float>(),
                      (const __bf16*)wa.data_ptr(), (const __bf16*)wv.data_ptr(),
                      B, (__bf16*)dh.data_ptr());
-  hipLaunchKernelGGL((dueling_q_loss_bwd_dw_kernel<512, 6>), dim3(7),
-                     dim3(256), 0, cur_stream(), grad_coef.data_ptr<float>(),
-                     act.data_ptr<int64_t>(), gout.data_ptr<float>(),
-                     (const __bf16*)h_s.data_ptr(), B,
+  constexpr int BS = 16;  // b-slices (the serial-B version measured slow)
+  auto ws = torch::empty({BS * 7 * 513}, torch::dtype(torch::kFloat32)
+                                             .device(h_s.device()));
+  hipLaunchKernelGGL((dueling_q_loss_bwd_dw_kernel<512, 6, BS>),
+                     dim3(7, BS), dim3(256), 0, cur_stream(),
+                     grad_coef.data_ptr<float>(), act.data_ptr<int64_t>(),
+                     gout.data_ptr<float>(), (const __bf16*)h_s.data_ptr(), B,
+                     ws.data_ptr<float>());
+  hipLaunchKernelGGL((dueling_q_loss_bwd_dw_fold_kernel<512, 6, BS>),
+                     dim3(ceil_div(7 * 513, kBlock)), dim3(kBlock), 0,
+                     cur_stream(), ws.data_ptr<float>(),
                      (__bf16*)dwa.data_ptr(), (__bf16*)dba.data_ptr(),
                      (__bf16*)dwv.data_ptr(), (__bf16*)dbv.data_ptr());
 }
@@ -2442,13 +2465,14 @@ namespace {
 
 __device__ __forceinline__ float vrescale(float x) {
   const float s = x >= 0.0f ? 1.0f : -1.0f;
-  return s * (sqrtf(fabsf(x) + 1.0f) - 1.0f) + 1e-2f * x;
+  return s * (sqrtf(fabsf(x) + 1.0f) - 1.0f) + 1e-3f * x;
 }
 
 __device__ __forceinline__ float inv_vrescale(float x) {
-  // closed-form inverse (R2D2/Learner.py:28-35), eps = 1e-2
+  // closed-form inverse (R2D2/Learner.py:28-35); eps matches the repo-wide
+  // _RESCALE_EPS = 1e-3 (ops/torch_ref.py:65)
   const float s = x >= 0.0f ? 1.0f : -1.0f;
-  const float e = 1e-2f;
+  const float e = 1e-3f;
   const float t = 1.0f + 4.0f * e * (fabsf(x) + 1.0f + e);
   const float r = (sqrtf(t) - 1.0f) / (2.0f * e);
   return s * (r * r - 1.0f);

@@ -213,10 +213,12 @@ def test_conv_dgrad_vs_aten(ext, shape):
         (dx.float() - ref).abs().max()
 
 
-def test_conv_dgrad_in_fused_backward(ext):
+def test_conv_dgrad_in_fused_backward(ext, monkeypatch):
     """End-to-end: the fused conv autograd must produce the same input grad
     with the hand-written dgrad as aten does on the fp32 path."""
     from distributed_rl_amd.ops import fused_conv_relu
+
+    monkeypatch.setenv("DRL_OWN_DGRAD", "1")
 
     torch.manual_seed(6)
     N, C, H, W, COUT, KH, S = 16, 32, 20, 20, 64, 4, 2
