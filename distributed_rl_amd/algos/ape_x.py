@@ -661,8 +661,13 @@ class ApexLearner(LearnerBase):
     def publish_weights(self, include_target: bool = False):
         if self.transport is None or self.rank != 0:
             return
+        if self.request_publish(include_target):
+            return  # async publisher thread will snapshot + publish
+        self._publish_sync(include_target, self.step_count)
+
+    def _publish_sync(self, include_target: bool, count: int):
         cpu_sd = self.snapshot_state_dict()
-        payload: Dict[str, Any] = {"count": self.step_count, "state_dict": cpu_sd}
+        payload: Dict[str, Any] = {"count": count, "state_dict": cpu_sd}
         if include_target:
             payload["target_state_dict"] = {
                 k: v.detach().to("cpu", torch.float32)

@@ -49,6 +49,7 @@ class LearnerBase:
         self.ingested_total = 0
         self._ingest_thread = None
         self._ingest_stop = None
+        self._pub_thread = None
 
     # -- background ingest -------------------------------------------------
     def start_ingest_thread(self):
@@ -76,13 +77,71 @@ class LearnerBase:
         self._ingest_thread = threading.Thread(target=loop, daemon=True,
                                                name="drl-ingest")
         self._ingest_thread.start()
+        self.start_publisher_thread()
 
     def stop_ingest_thread(self):
+        self.stop_publisher_thread()
         if self._ingest_thread is None:
             return
         self._ingest_stop.set()
         self._ingest_thread.join(10)
         self._ingest_thread = None
+
+    # -- async weight publisher -------------------------------------------
+    def start_publisher_thread(self):
+        """Publish weights from a daemon thread: the D2H snapshot copy is
+        stream-ordered on the compute stream (consistent between step
+        graphs) but the event wait + state-dict build + pickle + shm write
+        block only the publisher — the reference pays its pickle+Redis cost
+        inline in the train loop (APE_X/Learner.py:212-216; IMPALA
+        publishes EVERY step, Learner.py:286-287)."""
+        import threading
+
+        if self._pub_thread is not None or self.transport is None \
+                or self.rank != 0:
+            return
+        self._pub_req = None
+        self._pub_stop = threading.Event()
+        self._pub_wake = threading.Event()
+
+        def loop():
+            while not self._pub_stop.is_set():
+                self._pub_wake.wait(0.2)
+                self._pub_wake.clear()
+                req = self._pub_req
+                if req is None:
+                    continue
+                self._pub_req = None
+                try:
+                    self._publish_sync(include_target=req[1], count=req[0])
+                except Exception as e:  # pragma: no cover
+                    print(f"[publisher-thread] died: {e!r}", flush=True)
+                    return
+
+        self._pub_thread = threading.Thread(target=loop, daemon=True,
+                                            name="drl-publish")
+        self._pub_thread.start()
+
+    def stop_publisher_thread(self):
+        if getattr(self, "_pub_thread", None) is None:
+            return
+        self._pub_stop.set()
+        self._pub_wake.set()
+        self._pub_thread.join(10)
+        self._pub_thread = None
+
+    def request_publish(self, include_target: bool = False) -> bool:
+        """Queue an async publish (coalescing: only the latest request
+        survives). Returns False if the publisher thread is not running —
+        caller should publish synchronously."""
+        if getattr(self, "_pub_thread", None) is None:
+            return False
+        # never drop an include_target request in favor of a plain one
+        prev = self._pub_req
+        inc = include_target or (prev is not None and prev[1])
+        self._pub_req = (self.step_count, inc)
+        self._pub_wake.set()
+        return True
 
     # -- model helpers ----------------------------------------------------
     def build_model(self) -> BaseAgent:

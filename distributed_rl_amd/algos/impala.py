@@ -324,7 +324,12 @@ class ImpalaLearner(LearnerBase):
     def publish_weights(self, include_target: bool = False):
         if self.transport is None or self.rank != 0:
             return
-        self.transport.publish({"count": self.step_count,
+        if self.request_publish(include_target):
+            return  # async publisher thread (IMPALA publishes EVERY step)
+        self._publish_sync(include_target, self.step_count)
+
+    def _publish_sync(self, include_target: bool, count: int):
+        self.transport.publish({"count": count,
                                 "state_dict": self.snapshot_state_dict()})
 
     # -- run loop (per-step TB scalars; SURVEY §5.5: 9 scalars per step) ---

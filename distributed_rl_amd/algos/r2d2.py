@@ -432,8 +432,13 @@ class R2D2Learner(LearnerBase):
     def publish_weights(self, include_target: bool = False):
         if self.transport is None or self.rank != 0:
             return
+        if self.request_publish(include_target):
+            return  # async publisher thread
+        self._publish_sync(include_target, self.step_count)
+
+    def _publish_sync(self, include_target: bool, count: int):
         payload = {
-            "count": self.step_count,
+            "count": count,
             "state_dict": self.snapshot_state_dict(),
         }
         if include_target:
