@@ -181,9 +181,15 @@ class ApexLearner(LearnerBase):
         )
         cnn = self.net.nodes[cnn_node]
 
+        own_lin = ops.linear_relu_supported(in_f, 2 * hidden)
+
+        def hidden_of(feat):
+            if own_lin and feat.dtype == torch.bfloat16:
+                return ops.fused_linear_relu(feat, w1, b1)
+            return F.relu(F.linear(feat.to(w1.dtype), w1, b1))
+
         def fast_fwd(x):
-            feat = cnn(x)
-            h = F.relu(F.linear(feat.to(w1.dtype), w1, b1))
+            h = hidden_of(cnn(x))
             adv = a_head(h[:, :hidden])
             val = v_head(h[:, hidden:])
             return ops.dueling_head(adv.float(), val.float())
@@ -191,15 +197,13 @@ class ApexLearner(LearnerBase):
         self._fast_fwd = fast_fwd
 
         def fast_heads(x):
-            feat = cnn(x)
-            h = F.relu(F.linear(feat.to(w1.dtype), w1, b1))
+            h = hidden_of(cnn(x))
             return a_head(h[:, :hidden]), v_head(h[:, hidden:])
 
         self._fast_heads = fast_heads
 
         def fast_hidden(x):
-            feat = cnn(x)
-            return F.relu(F.linear(feat.to(w1.dtype), w1, b1))
+            return hidden_of(cnn(x))
 
         self._fast_hidden = fast_hidden
         self._head_params = (a_head.weight, a_head.bias, v_head.weight,
@@ -213,16 +217,19 @@ class ApexLearner(LearnerBase):
         w1t = self.flat_tparam[: 2 * hidden * in_f].view(2 * hidden, in_f)
         b1t = self.flat_tparam[2 * hidden * in_f : 2 * hidden * (in_f + 1)]
 
+        def t_hidden_of(feat):
+            if own_lin and feat.dtype == torch.bfloat16:
+                return ops.fused_linear_relu(feat, w1t, b1t)
+            return F.relu(F.linear(feat.to(w1t.dtype), w1t, b1t))
+
         def target_heads(x):
-            feat = t_cnn(x)
-            h = F.relu(F.linear(feat.to(w1t.dtype), w1t, b1t))
+            h = t_hidden_of(t_cnn(x))
             return ta_head(h[:, :hidden]), tv_head(h[:, hidden:])
 
         self._target_heads = target_heads
 
         def target_hidden(x):
-            feat = t_cnn(x)
-            return F.relu(F.linear(feat.to(w1t.dtype), w1t, b1t))
+            return t_hidden_of(t_cnn(x))
 
         self._target_hidden = target_hidden
         self._thead_params = (ta_head.weight, ta_head.bias, tv_head.weight,

@@ -398,3 +398,44 @@ def test_dqn_loss_bf16_inputs(ext):
     assert torch.allclose(prio.cpu(), prio_ref, atol=1e-3)
     qm_ref = q_s.detach().float().max(1).values.mean()
     assert abs(qmean.item() - qm_ref.item()) < 1e-3
+
+
+def test_linear_relu_matches_torch():
+    """Own MFMA Linear+bias+ReLU vs F.relu(F.linear) on identical bf16
+    operands, forward + backward."""
+    import torch.nn.functional as F
+
+    from distributed_rl_amd import ops
+
+    if not ops.linear_relu_supported(3136, 1024):
+        import pytest
+
+        pytest.skip("no instantiation")
+    torch.manual_seed(12)
+    dev = "cuda:0"
+    M, K, N = 512, 3136, 1024
+    x = (torch.randn(M, K, device=dev) * 0.5).to(torch.bfloat16
+                                                 ).requires_grad_(True)
+    x2 = x.detach().clone().requires_grad_(True)
+    w = (torch.randn(N, K, device=dev) * 0.02).to(torch.bfloat16
+                                                  ).requires_grad_(True)
+    w2 = w.detach().clone().requires_grad_(True)
+    b = torch.randn(N, device=dev).to(torch.bfloat16).requires_grad_(True)
+    b2 = b.detach().clone().requires_grad_(True)
+
+    out = ops.fused_linear_relu(x, w, b)
+    ref = F.relu(F.linear(x2, w2, b2))
+    torch.cuda.synchronize()
+    assert torch.allclose(out.float(), ref.float(), atol=2e-2, rtol=2e-2), \
+        (out.float() - ref.float()).abs().max()
+
+    g = torch.randn_like(ref)
+    out.backward(g)
+    ref.backward(g)
+    assert torch.allclose(x.grad.float(), x2.grad.float(), atol=2e-2,
+                          rtol=2e-2)
+    rel = (w.grad.float() - w2.grad.float()).abs().max() / \
+        w2.grad.float().abs().max()
+    assert rel < 0.05, rel
+    assert torch.allclose(b.grad.float(), b2.grad.float(), atol=2e-2,
+                          rtol=2e-2)
