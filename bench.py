@@ -209,8 +209,10 @@ def main():
         codec = RecordCodec(schema, with_priority=with_prio)
         tdir = args.transport_dir or tempfile.mkdtemp(prefix="drl_bench_")
         if rank == 0:
+            # rings sized to absorb the graph-capture pause (ingest thread
+            # stops for a few seconds; 256 slots measured 33k drops)
             session = TransportSession(tdir, codec, num_rings=args.with_actors,
-                                       ring_slots=256, create=True)
+                                       ring_slots=1024, create=True)
         if world > 1:
             torch.distributed.barrier()
         if rank != 0:

@@ -383,7 +383,10 @@ class ApexLearner(LearnerBase):
                 self._ingest_stream)
             cursor = 0
 
+        stop_evt = self._ingest_stop
         for views, n, advance in self.transport.drain_views():
+            if stop_evt is not None and stop_evt.is_set():
+                break  # prompt exit mid-sweep (thread shutdown)
             for v in views:
                 nrows = v.shape[0]
                 i = 0
@@ -690,7 +693,8 @@ class ApexLearner(LearnerBase):
         t0 = time.time()
         while len(self.replay) <= need:
             if self._ingest_thread is None:
-                self.ingest()
+                with self._ingest_lock:
+                    self.ingest()
             if time.time() - t0 > timeout:
                 raise TimeoutError(
                     f"replay warmup stalled at {len(self.replay)}/{need}"

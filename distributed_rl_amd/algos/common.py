@@ -50,6 +50,12 @@ class LearnerBase:
         self._ingest_thread = None
         self._ingest_stop = None
         self._pub_thread = None
+        import threading as _threading
+
+        # serializes ingest() across the daemon thread and any inline
+        # caller — a stop-join timeout must never leave two drains racing
+        # on the shared staging buffers
+        self._ingest_lock = _threading.Lock()
 
     # -- background ingest -------------------------------------------------
     def start_ingest_thread(self):
@@ -64,10 +70,13 @@ class LearnerBase:
             return
         self._ingest_stop = threading.Event()
 
+        stop = self._ingest_stop
+
         def loop():
-            while not self._ingest_stop.is_set():
+            while not stop.is_set():
                 try:
-                    n = self.ingest()
+                    with self._ingest_lock:
+                        n = self.ingest()
                 except Exception as e:  # pragma: no cover
                     print(f"[ingest-thread] died: {e!r}", flush=True)
                     return
@@ -84,7 +93,10 @@ class LearnerBase:
         if self._ingest_thread is None:
             return
         self._ingest_stop.set()
-        self._ingest_thread.join(10)
+        self._ingest_thread.join(60)
+        if self._ingest_thread.is_alive():  # pragma: no cover
+            print("[ingest-thread] did not stop within 60s (continuing; "
+                  "the ingest lock keeps a restart safe)", flush=True)
         self._ingest_thread = None
 
     # -- async weight publisher -------------------------------------------

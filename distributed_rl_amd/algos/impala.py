@@ -339,7 +339,8 @@ class ImpalaLearner(LearnerBase):
         t0 = time.time()
         while len(self.replay) < need:
             if self._ingest_thread is None:
-                self.ingest()
+                with self._ingest_lock:
+                    self.ingest()
             if time.time() - t0 > 600:
                 raise TimeoutError("IMPALA replay warmup stalled")
             time.sleep(0.01)
@@ -347,14 +348,16 @@ class ImpalaLearner(LearnerBase):
         stepper = None  # hipGraph-captured once the FIFO ring is full
         while self.step_count < max_steps:
             if self._ingest_thread is None:
-                self.ingest()
+                with self._ingest_lock:
+                    self.ingest()
             if self.max_replay_reuse > 0 and self.transport is not None:
                 # block until the fleet has produced enough fresh unrolls
                 t_gate = time.time()
                 while ((self.step_count + 1) * self.batch_size
                        > self.max_replay_reuse * max(self.ingested_total, 1)):
                     if self._ingest_thread is None:
-                        self.ingest()
+                        with self._ingest_lock:
+                            self.ingest()
                     time.sleep(0.002)
                     if time.time() - t_gate > 600:
                         raise TimeoutError("IMPALA reuse gate starved "

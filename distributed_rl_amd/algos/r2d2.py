@@ -455,7 +455,8 @@ class R2D2Learner(LearnerBase):
         t0 = time.time()
         while len(self.replay) <= need:
             if self._ingest_thread is None:
-                self.ingest()
+                with self._ingest_lock:
+                    self.ingest()
             if time.time() - t0 > 1200:
                 raise TimeoutError("R2D2 replay warmup stalled")
             time.sleep(0.01)
@@ -463,7 +464,8 @@ class R2D2Learner(LearnerBase):
         stepper = None  # hipGraph-captured once the replay ring is full
         while self.step_count < max_steps:
             if self._ingest_thread is None:
-                self.ingest()
+                with self._ingest_lock:
+                    self.ingest()
             if stepper is None and self.device.type == "cuda" \
                     and len(self.replay) >= self.replay.capacity:
                 # n_valid is baked into the captured sample kernel; once the

@@ -99,7 +99,8 @@ def _paced_run(learner, max_steps, train_per_ingest):
     ingested = len(learner.replay)
     budget = 0.0
     while learner.step_count < max_steps:
-        got = learner.ingest()
+        with learner._ingest_lock:
+            got = learner.ingest()
         ingested += got
         budget += got * train_per_ingest / learner.batch_size
         if budget < 1.0:
