@@ -1942,17 +1942,30 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
     float* __restrict__ tanhc,         // (T, B, H)
     unsigned int* __restrict__ ctr,    // zeroed before launch
     int B, int T) {
+  // Latency discipline (the non-LDS version measured ~8-12 us/step: 16
+  // serialized L2 round-trips per wave): the block's W slice (64 gate-rows
+  // x H) is burst-staged into LDS ONCE and reused for all T steps; h is
+  // burst-staged per step. Fragment reads then come from LDS.
+  constexpr int LROW = H + 32;  // (LROW*2) % 256 == 64: spreads b128 quads
+  __shared__ __bf16 wlds[64][LROW];
+  __shared__ __bf16 hlds[32][LROW];
   __shared__ float gbuf[4][32][16];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int s16 = blockIdx.x * 16;
-  const int col = wave * H + s16 + (lane & 15);
-  const __bf16* wrow = w_bf + (int64_t)col * H;
+  const int col = wave * H + s16 + (lane & 15);  // gate column in 4H
   const int kbase = (lane >> 4) * 8;
-  const int r0 = lane & 15, r1 = 16 + (lane & 15);
-  const int r0c = r0 < B ? r0 : 0, r1c = r1 < B ? r1 : 0;
   const unsigned nblocks = gridDim.x;
+
+  // stage W: LDS row g*16+c <- w_bf[g*H + s16 + c][:]
+  for (int base = tid * 8; base < 64 * H; base += 256 * 8) {
+    const int r = base / H;
+    const int k = base - r * H;
+    const int g = r >> 4, c = r & 15;
+    *reinterpret_cast<bf16x8_k5*>(&wlds[r][k]) =
+        *reinterpret_cast<const bf16x8_k5*>(w_bf + ((int64_t)g * H + s16 + c) * H + k);
+  }
 
   // register-resident cell state: thread e owns elements e, e+256 of the
   // block's (32 x 16) tile
@@ -1963,18 +1976,30 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
     int row = e >> 4, hc = e & 15;
     c_reg[u] = (row < B) ? c0[(int64_t)row * H + s16 + hc] : 0.0f;
   }
+  __syncthreads();
 
   for (int t = 0; t < T; ++t) {
     const __bf16* h_bf = h_bfs + (int64_t)t * B * H;
     const float* xp_t = xp + (int64_t)t * B * 4 * H;
+    // burst-stage h_t (zero-pad rows >= B)
+    for (int base = tid * 8; base < 32 * H; base += 256 * 8) {
+      const int r = base / H;
+      const int k = base - r * H;
+      bf16x8_k5 v{};
+      if (r < B)
+        v = *reinterpret_cast<const bf16x8_k5*>(h_bf + (int64_t)r * H + k);
+      *reinterpret_cast<bf16x8_k5*>(&hlds[r][k]) = v;
+    }
+    __syncthreads();
     f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
 #pragma unroll 4
     for (int kb = 0; kb < H; kb += 32) {
-      bf16x8_k5 bfrag = *reinterpret_cast<const bf16x8_k5*>(wrow + kb + kbase);
+      bf16x8_k5 bfrag = *reinterpret_cast<const bf16x8_k5*>(
+          &wlds[wave * 16 + (lane & 15)][kb + kbase]);
       bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
-          h_bf + (int64_t)r0c * H + kb + kbase);
+          &hlds[lane & 15][kb + kbase]);
       bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
-          h_bf + (int64_t)r1c * H + kb + kbase);
+          &hlds[16 + (lane & 15)][kb + kbase]);
       acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
       acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
     }
@@ -2048,16 +2073,31 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_bf16_kernel(
     float* __restrict__ dc0_out,        // (B, H)
     unsigned int* __restrict__ ctr,     // zeroed before launch
     int B, int T) {
+  // Latency discipline (see the fwd kernel): the block's W^T slice
+  // (16 h-cols x 4H) is burst-staged into LDS once; dg_{t+1} (32 x 4H,
+  // 128 KB bf16) is burst-staged per step in TWO 64 KB phases reusing one
+  // buffer, with the k-split re-drawn so each wave covers a quarter of
+  // each phase (the union over waves x phases is the full K; partials
+  // still fold 4-ways through LDS).
+  constexpr int LROWW = 4 * H + 32;
+  constexpr int LROWD = 2 * H + 32;
+  __shared__ __bf16 wtlds[16][LROWW];
+  __shared__ __bf16 dglds[32][LROWD];
   __shared__ float partial[4][32][16];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int n0 = blockIdx.x * 16;
   const int kbase = (lane >> 4) * 8;
-  const int r0 = lane & 15, r1 = 16 + (lane & 15);
-  const int r0c = r0 < B ? r0 : 0, r1c = r1 < B ? r1 : 0;
-  const __bf16* wcol = w_t_bf + (int64_t)(n0 + (lane & 15)) * 4 * H;
   const unsigned nblocks = gridDim.x;
+
+  for (int base = tid * 8; base < 16 * 4 * H; base += 256 * 8) {
+    const int c = base / (4 * H);
+    const int k = base - c * 4 * H;
+    *reinterpret_cast<bf16x8_k5*>(&wtlds[c][k]) =
+        *reinterpret_cast<const bf16x8_k5*>(
+            w_t_bf + (int64_t)(n0 + c) * 4 * H + k);
+  }
 
   float dc_reg[2];
 #pragma unroll
@@ -2072,17 +2112,31 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_bf16_kernel(
     if (it > 0) {
       const __bf16* dg_prev = dgates_bf + (int64_t)(t + 1) * B * 4 * H;
       f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
-      const int k0 = wave * H;
+#pragma unroll
+      for (int ph = 0; ph < 2; ++ph) {
+        __syncthreads();  // dglds reuse: prior phase reads must finish
+        for (int base = tid * 8; base < 32 * 2 * H; base += 256 * 8) {
+          const int r = base / (2 * H);
+          const int k = base - r * 2 * H;
+          bf16x8_k5 v{};
+          if (r < B)
+            v = *reinterpret_cast<const bf16x8_k5*>(
+                dg_prev + (int64_t)r * 4 * H + ph * 2 * H + k);
+          *reinterpret_cast<bf16x8_k5*>(&dglds[r][k]) = v;
+        }
+        __syncthreads();
+        const int kw0 = wave * (H / 2);  // this wave's quarter of the phase
 #pragma unroll 4
-      for (int kb = 0; kb < H; kb += 32) {
-        bf16x8_k5 bfrag =
-            *reinterpret_cast<const bf16x8_k5*>(wcol + k0 + kb + kbase);
-        bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
-            dg_prev + (int64_t)r0c * 4 * H + k0 + kb + kbase);
-        bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
-            dg_prev + (int64_t)r1c * 4 * H + k0 + kb + kbase);
-        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
-        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
+        for (int kb = kw0; kb < kw0 + H / 2; kb += 32) {
+          bf16x8_k5 bfrag = *reinterpret_cast<const bf16x8_k5*>(
+              &wtlds[lane & 15][ph * 2 * H + kb + kbase]);
+          bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
+              &dglds[lane & 15][kb + kbase]);
+          bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
+              &dglds[16 + (lane & 15)][kb + kbase]);
+          acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
+          acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
+        }
       }
       const int crow = (lane >> 4) * 4;
 #pragma unroll
