@@ -265,6 +265,17 @@ def fused_linear_relu(x, w, b):
     return _LinearReluFn.apply(x, w, b)
 
 
+def seq_transpose(src: torch.Tensor) -> torch.Tensor:
+    """(B, T, ...) -> (T, B, ...) whole-row block copy (the torch
+    permute+contiguous of R2D2's 72 MB u8 frame block ran at ~1.5 TB/s)."""
+    if _use_hip(src) and (src[0, 0].numel() * src.element_size()) % 16 == 0:
+        dst = torch.empty((src.shape[1], src.shape[0], *src.shape[2:]),
+                          dtype=src.dtype, device=src.device)
+        hip_ext().seq_transpose_rows(src.contiguous(), dst)
+        return dst
+    return src.transpose(0, 1).contiguous()
+
+
 # ---------------------------------------------------------------------------
 # K3 — fused dueling-head epilogue (A + V) - mean(A)
 # ---------------------------------------------------------------------------

@@ -980,16 +980,37 @@ __global__ __launch_bounds__(256) void linear_relu_kernel(
 #pragma unroll
   for (int f = 0; f < NFRAG; ++f) acc[f] = {0.f, 0.f, 0.f, 0.f};
 
+  // software pipeline: the next chunk's W pieces are prefetched into
+  // registers while the current chunk's MFMAs run (a single-buffered
+  // stage->sync->MFMA loop exposed the full global latency per chunk)
+  constexpr int PIECES = NT * KB / (256 * 8);
+  bf16x8 wpre[PIECES];
+#pragma unroll
+  for (int pc = 0; pc < PIECES; ++pc) {
+    const int base = (tid + pc * 256) * 8;
+    const int r = base / KB;
+    wpre[pc] = *reinterpret_cast<const bf16x8*>(
+        w + (int64_t)(blockIdx.y * NT + r) * K + (base - r * KB));
+  }
   for (int k0 = 0; k0 < K; k0 += KB) {
-    // stage W[NT rows][KB] (each thread 2x bf16x8 pieces)
     __syncthreads();
-    for (int base = tid * 8; base < NT * KB; base += 256 * 8) {
+#pragma unroll
+    for (int pc = 0; pc < PIECES; ++pc) {
+      const int base = (tid + pc * 256) * 8;
       const int r = base / KB;
-      const int k = base - r * KB;
-      *reinterpret_cast<bf16x8*>(&wlds[r][k]) = *reinterpret_cast<const bf16x8*>(
-          w + (int64_t)(blockIdx.y * NT + r) * K + k0 + k);
+      *reinterpret_cast<bf16x8*>(&wlds[r][base - r * KB]) = wpre[pc];
     }
     __syncthreads();
+    if (k0 + KB < K) {
+#pragma unroll
+      for (int pc = 0; pc < PIECES; ++pc) {
+        const int base = (tid + pc * 256) * 8;
+        const int r = base / KB;
+        wpre[pc] = *reinterpret_cast<const bf16x8*>(
+            w + (int64_t)(blockIdx.y * NT + r) * K + k0 + KB +
+            (base - r * KB));
+      }
+    }
 #pragma unroll
     for (int kc = 0; kc < KB / 32; ++kc) {
       bf16x8 a = *reinterpret_cast<const bf16x8*>(

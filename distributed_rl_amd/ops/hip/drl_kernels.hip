@@ -743,6 +743,7 @@ void r2d2_loss_fwd(torch::Tensor q_train, torch::Tensor q_tgt,
 void r2d2_prio(torch::Tensor td, double alpha, double eta, torch::Tensor prio);
 void r2d2_loss_bwd(torch::Tensor td, torch::Tensor act, torch::Tensor w,
                    torch::Tensor gout, int64_t T, int64_t m, torch::Tensor dq);
+void seq_transpose_rows(torch::Tensor src, torch::Tensor dst);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -806,6 +807,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("r2d2_prio", &r2d2_prio, "eta-mix sequence priority (K7)");
   m.def("r2d2_loss_bwd", &r2d2_loss_bwd,
         "R2D2 loss backward: closed-form dq_train scatter");
+  m.def("seq_transpose_rows", &seq_transpose_rows,
+        "(B,T,row) -> (T,B,row) whole-row block copy");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -2035,11 +2038,12 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
       c_out[hidx] = c;
       h_bf_out[hidx] = (__bf16)h;
       tanhc[(int64_t)t * B * H + hidx] = tc;
-      float* a4 = acts + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + s16 + hc;
-      a4[0] = i_;
-      a4[H] = f_;
-      a4[2 * H] = g_;
-      a4[3 * H] = o_;
+      // acts packed (T,B,H,4): ONE 16 B store instead of four 4 B stores
+      // 2 KB apart — the barrier's vmcnt(0) drain paid for every scattered
+      // store (layout is internal to the persistent fwd/bwd pair)
+      *reinterpret_cast<float4*>(
+          acts + (((int64_t)t * B + row) * H + s16 + hc) * 4) =
+          float4{i_, f_, g_, o_};
     }
     // agent-scope grid barrier: h_bf_out must be visible to every block
     __syncthreads();
@@ -2157,9 +2161,9 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_bf16_kernel(
                 partial[3][row][hc]
           : dh_init[hidx];
       dhv += gout[(int64_t)t * B * H + hidx];
-      const float* a4 =
-          acts + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + n0 + hc;
-      float i = a4[0], f = a4[H], g = a4[2 * H], o = a4[3 * H];
+      const float4 a4 = *reinterpret_cast<const float4*>(
+          acts + (((int64_t)t * B + row) * H + n0 + hc) * 4);
+      float i = a4.x, f = a4.y, g = a4.z, o = a4.w;
       float tc = tanhc[(int64_t)t * B * H + hidx];
       float do_ = dhv * tc;
       float dct = dc_reg[u] + dhv * o * (1.0f - tc * tc);
@@ -2167,22 +2171,38 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_bf16_kernel(
       float df = dct * cs[(int64_t)t * B * H + hidx];
       float dg = dct * i;
       dc_reg[u] = dct * f;
-      float v0 = di * i * (1.0f - i);
-      float v1 = df * f * (1.0f - f);
-      float v2 = dg * (1.0f - g * g);
-      float v3 = do_ * o * (1.0f - o);
-      float* d4 =
-          dgates + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + n0 + hc;
-      d4[0] = v0;
-      d4[H] = v1;
-      d4[2 * H] = v2;
-      d4[3 * H] = v3;
-      __bf16* b4 =
-          dgates_bf + (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + n0 + hc;
-      b4[0] = (__bf16)v0;
-      b4[H] = (__bf16)v1;
-      b4[2 * H] = (__bf16)v2;
-      b4[3 * H] = (__bf16)v3;
+      // park dgates in the (already-consumed) partial buffer: each thread
+      // reads/writes only its own (row, hc) entries, so no cross-thread
+      // hazard; a cooperative burst below emits coalesced 32 B runs
+      // instead of 8 scalar stores 2 KB apart (barrier drain cost)
+      partial[0][row][hc] = di * i * (1.0f - i);
+      partial[1][row][hc] = df * f * (1.0f - f);
+      partial[2][row][hc] = dg * (1.0f - g * g);
+      partial[3][row][hc] = do_ * o * (1.0f - o);
+    }
+    __syncthreads();
+    {
+      // burst dgates (fp32, gate-major) + dgates_bf (bf16) from LDS:
+      // thread pair covers one (gate, row): 16 contiguous cols
+      const int pr = tid >> 1;        // 0..127 = gate*32 + row
+      const int half = (tid & 1) * 8;
+      const int g_ = pr >> 5;
+      const int row = pr & 31;
+      if (row < B) {
+        const int64_t base =
+            (int64_t)t * B * 4 * H + (int64_t)row * 4 * H + g_ * H + n0;
+        float4 lo = *reinterpret_cast<const float4*>(&partial[g_][row][half]);
+        float4 hi =
+            *reinterpret_cast<const float4*>(&partial[g_][row][half + 4]);
+        *reinterpret_cast<float4*>(dgates + base + half) = lo;
+        *reinterpret_cast<float4*>(dgates + base + half + 4) = hi;
+        bf16x8_k5 bv;
+        bv[0] = (__bf16)lo.x; bv[1] = (__bf16)lo.y;
+        bv[2] = (__bf16)lo.z; bv[3] = (__bf16)lo.w;
+        bv[4] = (__bf16)hi.x; bv[5] = (__bf16)hi.y;
+        bv[6] = (__bf16)hi.z; bv[7] = (__bf16)hi.w;
+        *reinterpret_cast<bf16x8_k5*>(dgates_bf + base + half) = bv;
+      }
     }
     __syncthreads();
     if (tid == 0) {
@@ -2670,4 +2690,34 @@ void r2d2_loss_bwd(torch::Tensor td, torch::Tensor act, torch::Tensor w,
                      act.data_ptr<int>(), w.data_ptr<float>(),
                      gout.data_ptr<float>(), (int)T, B, A, (int)m,
                      dq.data_ptr<float>());
+}
+
+// ===========================================================================
+// Sequence transpose (R2D2): (B, T, R bytes) -> (T, B, R) row gather. The
+// torch permute().contiguous() of the 72 MB uint8 frame block ran at
+// ~1.5 TB/s (92.7 us); whole-row uint4 block copies stream at full rate.
+// ===========================================================================
+namespace {
+__global__ void seq_transpose_rows_kernel(const uint4* __restrict__ src,
+                                          uint4* __restrict__ dst, int B,
+                                          int T, int r16) {
+  const int bt = blockIdx.x;
+  const int b = bt / T;
+  const int t = bt - b * T;
+  const uint4* s = src + (int64_t)bt * r16;
+  uint4* d = dst + ((int64_t)t * B + b) * r16;
+  for (int i = threadIdx.x; i < r16; i += blockDim.x) d[i] = s[i];
+}
+}  // namespace
+
+void seq_transpose_rows(torch::Tensor src, torch::Tensor dst) {
+  // src: (B, T, ...) contiguous; dst: (T, B, ...) contiguous, same dtype
+  const int B = (int)src.size(0), T = (int)src.size(1);
+  const int64_t row_bytes =
+      src.numel() * src.element_size() / ((int64_t)B * T);
+  TORCH_CHECK(row_bytes % 16 == 0, "row bytes must be 16-aligned");
+  TORCH_CHECK(src.is_contiguous() && dst.is_contiguous());
+  hipLaunchKernelGGL(seq_transpose_rows_kernel, dim3(B * T), dim3(256), 0,
+                     cur_stream(), (const uint4*)src.data_ptr(),
+                     (uint4*)dst.data_ptr(), B, T, (int)(row_bytes / 16));
 }
