@@ -181,7 +181,10 @@ class ApexLearner(LearnerBase):
         )
         cnn = self.net.nodes[cnn_node]
 
-        own_lin = ops.linear_relu_supported(in_f, 2 * hidden)
+        import os as _os
+
+        own_lin = (ops.linear_relu_supported(in_f, 2 * hidden)
+                   and _os.environ.get("DRL_OWN_LINEAR", "1") == "1")
 
         def hidden_of(feat):
             if own_lin and feat.dtype == torch.bfloat16:

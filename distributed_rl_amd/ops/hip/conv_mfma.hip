@@ -963,7 +963,7 @@ __global__ __launch_bounds__(256) void linear_relu_kernel(
     __bf16* __restrict__ out,       // (M, N)
     int M) {
   constexpr int KB = 64;            // K chunk staged per iteration
-  constexpr int LROW = KB + 8;      // LDS pad
+  constexpr int LROW = KB + 32;     // (LROW*2)%256 == 192: bank-spread pad
   __shared__ __bf16 wlds[NT][LROW];
   const int tid = threadIdx.x;
   const int lane = tid & 63;
@@ -992,6 +992,13 @@ __global__ __launch_bounds__(256) void linear_relu_kernel(
     wpre[pc] = *reinterpret_cast<const bf16x8*>(
         w + (int64_t)(blockIdx.y * NT + r) * K + (base - r * KB));
   }
+  // A fragments also 1-deep prefetched (they were the remaining serial
+  // latency chain: one L2 round-trip per 16 MFMAs)
+  bf16x8 a_cur[2], a_nxt[2];
+#pragma unroll
+  for (int kc = 0; kc < 2; ++kc)
+    a_cur[kc] = *reinterpret_cast<const bf16x8*>(
+        x + (int64_t)arow_c * K + kc * 32 + kpart);
   for (int k0 = 0; k0 < K; k0 += KB) {
     __syncthreads();
 #pragma unroll
@@ -1001,6 +1008,12 @@ __global__ __launch_bounds__(256) void linear_relu_kernel(
       *reinterpret_cast<bf16x8*>(&wlds[r][base - r * KB]) = wpre[pc];
     }
     __syncthreads();
+    if (k0 + KB < K) {
+#pragma unroll
+      for (int kc = 0; kc < 2; ++kc)
+        a_nxt[kc] = *reinterpret_cast<const bf16x8*>(
+            x + (int64_t)arow_c * K + k0 + KB + kc * 32 + kpart);
+    }
     if (k0 + KB < K) {
 #pragma unroll
       for (int pc = 0; pc < PIECES; ++pc) {
@@ -1013,8 +1026,7 @@ __global__ __launch_bounds__(256) void linear_relu_kernel(
     }
 #pragma unroll
     for (int kc = 0; kc < KB / 32; ++kc) {
-      bf16x8 a = *reinterpret_cast<const bf16x8*>(
-          x + (int64_t)arow_c * K + k0 + kc * 32 + kpart);
+      bf16x8 a = a_cur[kc];
 #pragma unroll
       for (int f = 0; f < NFRAG; ++f) {
         const int wl = (wave & 1) * (NT / 2) + f * 16 + (lane & 15);
@@ -1023,6 +1035,8 @@ __global__ __launch_bounds__(256) void linear_relu_kernel(
         acc[f] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bf, acc[f], 0, 0, 0);
       }
     }
+    a_cur[0] = a_nxt[0];
+    a_cur[1] = a_nxt[1];
   }
   const int crow = (lane >> 4) * 4;
 #pragma unroll
