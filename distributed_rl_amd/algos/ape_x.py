@@ -183,8 +183,12 @@ class ApexLearner(LearnerBase):
 
         import os as _os
 
+        # own linear_relu kernel measured SLOWER than hipBLASLt at this
+        # shape even after prefetch pipelining (A/B: 0.741 vs 0.576 ms
+        # step) — the 49-chunk stage/sync cadence dominates; dispatch keeps
+        # the library GEMM, DRL_OWN_LINEAR=1 forces ours
         own_lin = (ops.linear_relu_supported(in_f, 2 * hidden)
-                   and _os.environ.get("DRL_OWN_LINEAR", "1") == "1")
+                   and _os.environ.get("DRL_OWN_LINEAR", "0") == "1")
 
         def hidden_of(feat):
             if own_lin and feat.dtype == torch.bfloat16:

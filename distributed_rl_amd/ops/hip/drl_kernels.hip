@@ -744,6 +744,8 @@ void r2d2_prio(torch::Tensor td, double alpha, double eta, torch::Tensor prio);
 void r2d2_loss_bwd(torch::Tensor td, torch::Tensor act, torch::Tensor w,
                    torch::Tensor gout, int64_t T, int64_t m, torch::Tensor dq);
 void seq_transpose_rows(torch::Tensor src, torch::Tensor dst);
+void gather_rows(torch::Tensor idx, std::vector<torch::Tensor> srcs,
+                 std::vector<torch::Tensor> dsts);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -809,6 +811,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "R2D2 loss backward: closed-form dq_train scatter");
   m.def("seq_transpose_rows", &seq_transpose_rows,
         "(B,T,row) -> (T,B,row) whole-row block copy");
+  m.def("gather_rows", &gather_rows,
+        "fused multi-column replay row gather (one launch per sample)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -2720,4 +2724,54 @@ void seq_transpose_rows(torch::Tensor src, torch::Tensor dst) {
   hipLaunchKernelGGL(seq_transpose_rows_kernel, dim3(B * T), dim3(256), 0,
                      cur_stream(), (const uint4*)src.data_ptr(),
                      (uint4*)dst.data_ptr(), B, T, (int)(row_bytes / 16));
+}
+
+// ===========================================================================
+// Fused replay gather (K10): one launch copies every column's sampled rows
+// (the generic per-column index_select chain was 8-10 kernels per sample:
+// 2 big frame gathers + small-column gathers/scatters, ~50 us/step).
+// Row bytes that are 16-aligned stream as uint4; ragged columns fall back
+// to byte copies. Block (b, c) copies row idx[b] of column c.
+// ===========================================================================
+namespace {
+struct GatherCols {
+  const uint8_t* src[8];
+  uint8_t* dst[8];
+  int64_t row_bytes[8];
+  int ncols;
+};
+
+__global__ void gather_rows_kernel(GatherCols a,
+                                   const int64_t* __restrict__ idx, int k) {
+  const int b = blockIdx.x;
+  const int c = blockIdx.y;
+  if (b >= k || c >= a.ncols) return;
+  const int64_t rb = a.row_bytes[c];
+  const uint8_t* s = a.src[c] + idx[b] * rb;
+  uint8_t* d = a.dst[c] + (int64_t)b * rb;
+  if ((rb & 15) == 0) {
+    const uint4* s4 = reinterpret_cast<const uint4*>(s);
+    uint4* d4 = reinterpret_cast<uint4*>(d);
+    for (int64_t i = threadIdx.x; i < rb / 16; i += blockDim.x) d4[i] = s4[i];
+  } else {
+    for (int64_t i = threadIdx.x; i < rb; i += blockDim.x) d[i] = s[i];
+  }
+}
+}  // namespace
+
+void gather_rows(torch::Tensor idx, std::vector<torch::Tensor> srcs,
+                 std::vector<torch::Tensor> dsts) {
+  TORCH_CHECK(srcs.size() == dsts.size() && srcs.size() <= 8);
+  GatherCols a{};
+  a.ncols = (int)srcs.size();
+  for (int c = 0; c < a.ncols; ++c) {
+    TORCH_CHECK(srcs[c].is_contiguous() && dsts[c].is_contiguous());
+    a.src[c] = (const uint8_t*)srcs[c].data_ptr();
+    a.dst[c] = (uint8_t*)dsts[c].data_ptr();
+    a.row_bytes[c] =
+        srcs[c].numel() * srcs[c].element_size() / srcs[c].size(0);
+  }
+  const int k = (int)idx.numel();
+  hipLaunchKernelGGL(gather_rows_kernel, dim3(k, a.ncols), dim3(256), 0,
+                     cur_stream(), a, idx.data_ptr<int64_t>(), k);
 }

@@ -70,6 +70,19 @@ class ReplayBase:
         self.count += n
 
     def gather(self, idx: torch.Tensor) -> Dict[str, torch.Tensor]:
+        if self.device.type == "cuda" and len(self.data) <= 8:
+            from ..ops import hip_ext
+
+            ext = hip_ext(required=False)
+            if ext is not None and hasattr(ext, "gather_rows"):
+                # fused multi-column row gather: ONE launch instead of the
+                # per-column index_select chain (8-10 kernels per sample)
+                srcs = list(self.data.values())
+                dsts = [torch.empty((idx.numel(), *c.shape[1:]),
+                                    dtype=c.dtype, device=c.device)
+                        for c in srcs]
+                ext.gather_rows(idx.to(torch.int64), srcs, dsts)
+                return dict(zip(self.data.keys(), dsts))
         return {name: col.index_select(0, idx) for name, col in self.data.items()}
 
     # -- checkpoint (optional; SURVEY §5.4 "PER state optional") ----------
