@@ -235,9 +235,12 @@ def main():
         # steady-state ingest, not an idle transport
         learner.start_ingest_thread()
         t_warm = time.perf_counter()
-        warm_target = max(8192, 16 * args.with_actors)
+        warm_target = max(8192, 32 * args.with_actors)
+        # big fleets take minutes to finish 200+ concurrent torch imports —
+        # wait longer so the timed region sees the steady state
+        warm_deadline = 180 if args.with_actors <= 128 else 420
         while learner.ingested_total < warm_target \
-                and time.perf_counter() - t_warm < 180:
+                and time.perf_counter() - t_warm < warm_deadline:
             time.sleep(0.05)
         dt_warm = time.perf_counter() - t_warm
         print(f"# fleet warm: {learner.ingested_total} rows in {dt_warm:.1f}s "
@@ -285,8 +288,13 @@ def main():
     if args.with_actors > 0:
         drops = (learner.transport.total_drops()
                  if hasattr(learner.transport, "total_drops") else -1)
+        pushed = (learner.transport.total_pushed()
+                  if hasattr(learner.transport, "total_pushed") else -1)
+        alive = (learner._ingest_thread is not None
+                 and learner._ingest_thread.is_alive())
         print(f"# transport: {ingested} rows ingested in timed region, "
-              f"{drops} dropped (ring backpressure) total", file=sys.stderr)
+              f"{pushed} pushed / {drops} dropped total, "
+              f"ingest_thread_alive={alive}", file=sys.stderr)
         learner.stop_ingest_thread()
     if fleet is not None:
         fleet.stop()

@@ -445,6 +445,11 @@ class LearnerEndpoint:
         had no space for (SPSC drop counters)."""
         return sum(self.session.ring(i).drops for i in self.ring_ids)
 
+    def total_pushed(self) -> int:
+        """Rows the fleet has ever written (ring head counters) —
+        distinguishes idle actors from a stalled drain."""
+        return sum(self.session.ring(i).head for i in self.ring_ids)
+
 
 # ---------------------------------------------------------------------------
 # In-process pipe (tests / single-process integration)

@@ -2742,18 +2742,26 @@ struct GatherCols {
 };
 
 __global__ void gather_rows_kernel(GatherCols a,
-                                   const int64_t* __restrict__ idx, int k) {
+                                   const int64_t* __restrict__ idx, int k,
+                                   int chunks) {
   const int b = blockIdx.x;
   const int c = blockIdx.y;
+  const int ch = blockIdx.z;  // big rows split across chunk blocks (a
+                              // 564 KB R2D2 row on ONE block starved the
+                              // chip: 32 blocks for an 18 MB gather)
   if (b >= k || c >= a.ncols) return;
   const int64_t rb = a.row_bytes[c];
   const uint8_t* s = a.src[c] + idx[b] * rb;
   uint8_t* d = a.dst[c] + (int64_t)b * rb;
   if ((rb & 15) == 0) {
+    const int64_t n16 = rb / 16;
+    const int64_t per = (n16 + chunks - 1) / chunks;
+    const int64_t lo = ch * per;
+    const int64_t hi = lo + per < n16 ? lo + per : n16;
     const uint4* s4 = reinterpret_cast<const uint4*>(s);
     uint4* d4 = reinterpret_cast<uint4*>(d);
-    for (int64_t i = threadIdx.x; i < rb / 16; i += blockDim.x) d4[i] = s4[i];
-  } else {
+    for (int64_t i = lo + threadIdx.x; i < hi; i += blockDim.x) d4[i] = s4[i];
+  } else if (ch == 0) {
     for (int64_t i = threadIdx.x; i < rb; i += blockDim.x) d[i] = s[i];
   }
 }
@@ -2772,6 +2780,13 @@ void gather_rows(torch::Tensor idx, std::vector<torch::Tensor> srcs,
         srcs[c].numel() * srcs[c].element_size() / srcs[c].size(0);
   }
   const int k = (int)idx.numel();
-  hipLaunchKernelGGL(gather_rows_kernel, dim3(k, a.ncols), dim3(256), 0,
-                     cur_stream(), a, idx.data_ptr<int64_t>(), k);
+  int64_t max_rb = 0;
+  for (int c = 0; c < a.ncols; ++c)
+    max_rb = a.row_bytes[c] > max_rb ? a.row_bytes[c] : max_rb;
+  // target >= ~1024 blocks for the biggest column
+  int chunks = 1;
+  while ((int64_t)k * chunks < 1024 && (max_rb / 16) / chunks > 256)
+    chunks *= 2;
+  hipLaunchKernelGGL(gather_rows_kernel, dim3(k, a.ncols, chunks), dim3(256),
+                     0, cur_stream(), a, idx.data_ptr<int64_t>(), k, chunks);
 }
