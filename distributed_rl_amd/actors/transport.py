@@ -423,14 +423,16 @@ class LearnerEndpoint:
     def record_dtype(self) -> np.dtype:
         return self.session.codec.np_dtype
 
-    def drain_views(self):
+    def drain_views(self, max_per_ring: int = 1 << 30):
         """Zero-copy drain: yields (views, n, advance) per non-empty ring.
         The caller must copy out of the shm views and THEN call advance(n)
         (the single host copy goes shm -> pinned staging directly; see
-        ApexLearner.ingest)."""
+        ApexLearner.ingest). max_per_ring bounds a sweep so a large backlog
+        (e.g. after the graph-capture pause) cannot turn one drain call
+        into a multi-second monolith."""
         for i in self.ring_ids:
             ring = self.session.ring(i)
-            views, n = ring.peek_records()
+            views, n = ring.peek_records(max_per_ring)
             if n:
                 yield views, n, ring.advance
 

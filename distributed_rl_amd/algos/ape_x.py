@@ -391,7 +391,8 @@ class ApexLearner(LearnerBase):
             cursor = 0
 
         stop_evt = self._ingest_stop
-        for views, n, advance in self.transport.drain_views():
+        for views, n, advance in self.transport.drain_views(
+                max_per_ring=self._STAGE_ROWS):
             if stop_evt is not None and stop_evt.is_set():
                 break  # prompt exit mid-sweep (thread shutdown)
             for v in views:
@@ -418,11 +419,13 @@ class ApexLearner(LearnerBase):
                     cursor += k
                     i += k
                     total += k
+                    self.ingested_total += k  # incremental: telemetry must
+                    # not wait for the whole sweep (a post-capture backlog
+                    # once made one call span seconds, reading as "0 rows")
                     if cursor == self._STAGE_ROWS:
                         flush()
             advance(n)
         flush()
-        self.ingested_total += total
         return total
 
     def push_experience(self, cols: Dict[str, torch.Tensor], prio: torch.Tensor):
