@@ -213,6 +213,10 @@ class ApexLearner(LearnerBase):
             return hidden_of(cnn(x))
 
         self._fast_hidden = fast_hidden
+        # direct-grad leaf list: the forward uses the FUSED (w1, b1) views
+        # instead of the four pinned params, and w1/b1's flattened grads
+        # cover exactly the flat buffer's pinned prefix
+        self._direct_list = [w1, b1] + list(self.mp.groups[0].c_params[4:])
         self._head_params = (a_head.weight, a_head.bias, v_head.weight,
                              v_head.bias)
         self._use_q_loss = ops.has_dueling_q_loss(hidden, self.cfg.action_size)
@@ -524,8 +528,7 @@ class ApexLearner(LearnerBase):
                 with_value_stat=True,
             )
         if self.mp is not None:
-            self.mp.zero_grads()
-            loss.backward()
+            self.mp.direct_grads(loss, getattr(self, "_direct_list", None))
         else:
             self.optim.zero_grad(set_to_none=False)
             loss.backward()

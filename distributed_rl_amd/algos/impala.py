@@ -181,8 +181,7 @@ class ImpalaLearner(LearnerBase):
             loss = -obj_actor + critic_loss
 
         if self.mp is not None:
-            self.mp.zero_grads()
-            loss.backward()
+            self.mp.direct_grads(loss)
         else:
             self.optim.zero_grad(set_to_none=False)
             loss.backward()

@@ -306,8 +306,7 @@ class R2D2Learner(LearnerBase):
             }
 
         if self.mp is not None:
-            self.mp.zero_grads()
-            loss.backward()
+            self.mp.direct_grads(loss)
         else:
             self.optim.zero_grad(set_to_none=False)
             loss.backward()

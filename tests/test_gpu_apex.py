@@ -190,3 +190,42 @@ def test_fused_q_head_loss_matches_composition():
         rel = (g1.float() - g2.float()).abs().max() / \
             g2.float().abs().max().clamp_min(1e-8)
         assert rel < 0.05, (nm, rel)
+
+
+def test_direct_grads_match_backward():
+    """mp.direct_grads (autograd.grad + one cat into the flat buffer) must
+    produce the same flat compute grads as zero_grads + loss.backward."""
+    import copy
+
+    from distributed_rl_amd.algos.ape_x import ApexLearner
+    from distributed_rl_amd.config import Config, load_config
+
+    raw = copy.deepcopy(load_config("ape_x").raw)
+    raw["REPLAY_MEMORY_LEN"] = 2048
+    raw["BATCHSIZE"] = 64
+    learner = ApexLearner(Config(raw=raw), device=DEV, enable_tb=False)
+    _fill(learner, 1024)
+    data, idx, w = learner.replay.sample(64, learner.beta)
+    stats, prio = learner._fwd_bwd(data, w)  # direct path
+    g_direct = learner.mp.flat_cgrad.clone()
+
+    # reference: classic backward through the same math
+    s = data["state"].permute(0, 3, 1, 2)
+    sp = data["next_state"].permute(0, 3, 1, 2)
+    from distributed_rl_amd import ops as _ops
+
+    h_s = learner._fast_hidden(s)
+    with torch.no_grad():
+        h_on = learner._fast_hidden(sp)
+        h_tg = learner._target_hidden(sp)
+    wa, ba, wv, bv = learner._head_params
+    wat, bat, wvt, bvt = learner._thead_params
+    loss, _, _ = _ops.dueling_q_head_loss(
+        h_s, wa, ba, wv, bv, h_on, h_tg, wat, bat, wvt, bvt,
+        data["action"].long(), data["reward"], data["done"], w,
+        learner.gamma, learner.n_step, learner.alpha)
+    learner.mp.zero_grads()
+    loss.backward()
+    torch.cuda.synchronize()
+    g_ref = learner.mp.flat_cgrad
+    assert torch.equal(g_direct, g_ref), (g_direct - g_ref).abs().max()
