@@ -32,11 +32,8 @@ namespace {
 using bf16x8 = __attribute__((ext_vector_type(8))) __bf16;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
-constexpr int kPad = 8;  // LDS row pad: makes the per-row count of 16 B
-// blocks ODD ((K+8)/8), so 16 lanes' b128 reads at consecutive rows land
-// on all 8 bank groups. Round-2 PMC re-measure: the old pad-32 gave
-// (K+32)/8 % 8 == 4 -> only 2 groups -> a systematic 4 conflict-cycles
-// per MFMA on conv fwd (gpurun_out/*_counter_collection.csv).
+constexpr int kPad = 32;  // LDS row pad: (K+pad)*2 % 256 in {64,192} spreads the
+// b128 16-lane service quads across all 16 bank-row slots (PMC-verified)
 
 template <int H, int W, int C, int KH, int KW, int S, int COUT, bool U8IN>
 struct ConvGeom {

@@ -1953,7 +1953,7 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
   // serialized L2 round-trips per wave): the block's W slice (64 gate-rows
   // x H) is burst-staged into LDS ONCE and reused for all T steps; h is
   // burst-staged per step. Fragment reads then come from LDS.
-  constexpr int LROW = H + 8;  // odd 16 B-block row stride: full bank-group spread
+  constexpr int LROW = H + 32;  // (LROW*2) % 256 == 64: spreads b128 quads
   __shared__ __bf16 wlds[64][LROW];
   __shared__ __bf16 hlds[32][LROW];
   __shared__ float gbuf[4][32][16];
@@ -2087,8 +2087,8 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_bf16_kernel(
   // buffer, with the k-split re-drawn so each wave covers a quarter of
   // each phase (the union over waves x phases is the full K; partials
   // still fold 4-ways through LDS).
-  constexpr int LROWW = 4 * H + 8;   // odd-block strides (bank spread)
-  constexpr int LROWD = 2 * H + 8;
+  constexpr int LROWW = 4 * H + 32;
+  constexpr int LROWD = 2 * H + 32;
   __shared__ __bf16 wtlds[16][LROWW];
   __shared__ __bf16 dglds[32][LROWD];
   __shared__ float partial[4][32][16];
