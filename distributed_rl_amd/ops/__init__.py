@@ -397,8 +397,14 @@ class _DuelingDQNLossFn(torch.autograd.Function):
         ctx.save_for_backward(coef, actions)
         ctx.shape = (B, A)
         ctx.in_dtype = adv_s.dtype
-        ctx.mark_non_differentiable(prio)
-        return acc2[0], prio, acc2[1]
+        # every non-loss output must be non-differentiable: a grad_fn-
+        # carrying stat held in a graphed stepper's static_out keeps the
+        # whole autograd graph alive across steps, and the stale
+        # AccumulateGrad stream then breaks hipGraph capture (observed as
+        # a core dump at R2D2 capture time)
+        loss_out, qmean = acc2[0], acc2[1]
+        ctx.mark_non_differentiable(prio, qmean)
+        return loss_out, prio, qmean
 
     @staticmethod
     def backward(ctx, gout, _gprio, _gqm):
@@ -464,8 +470,9 @@ class _DuelingQLossFn(torch.autograd.Function):
             acc2[0:1], prio, coef, acc2[1:2],
         )
         ctx.save_for_backward(coef, actions, wa, wv, h_s)
-        ctx.mark_non_differentiable(prio)
-        return acc2[0], prio, acc2[1]
+        loss_out, qmean = acc2[0], acc2[1]
+        ctx.mark_non_differentiable(prio, qmean)
+        return loss_out, prio, qmean
 
     @staticmethod
     def backward(ctx, gout, _gprio, _gqm):
@@ -506,8 +513,13 @@ class _R2D2SeqLossFn(torch.autograd.Function):
         ext.r2d2_prio(td, float(alpha), float(eta), prio)
         ctx.save_for_backward(td, actions, weights)
         ctx.meta = (T, burn_in, tuple(q_train.shape))
-        ctx.mark_non_differentiable(prio)
-        return stats[0], prio, stats[1], stats[2]
+        # stats views must be non-differentiable: a grad_fn-carrying stat
+        # in a graphed stepper's static_out keeps the autograd graph alive
+        # across steps and the stale AccumulateGrad stream then breaks
+        # hipGraph capture (core dump at R2D2 capture, r2d2_live.log)
+        loss_out, value, td_abs = stats[0], stats[1], stats[2]
+        ctx.mark_non_differentiable(prio, value, td_abs)
+        return loss_out, prio, value, td_abs
 
     @staticmethod
     def backward(ctx, gout, _gp, _gv, _gt):
