@@ -746,6 +746,8 @@ void r2d2_loss_bwd(torch::Tensor td, torch::Tensor act, torch::Tensor w,
 void seq_transpose_rows(torch::Tensor src, torch::Tensor dst);
 void gather_rows(torch::Tensor idx, std::vector<torch::Tensor> srcs,
                  std::vector<torch::Tensor> dsts);
+void lstm_diag(torch::Tensor h_bfs, torch::Tensor w_bf, torch::Tensor sink,
+               torch::Tensor ctr, int64_t B, int64_t T, int64_t mode);
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   register_conv(m);
@@ -813,6 +815,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "(B,T,row) -> (T,B,row) whole-row block copy");
   m.def("gather_rows", &gather_rows,
         "fused multi-column replay row gather (one launch per sample)");
+  m.def("lstm_diag", &lstm_diag,
+        "persistent-step cost decomposition (barrier/stage/mfma)");
 }
 // appended: fused ReLU-mask backward (gout *= (out > 0)), bf16, one pass —
 // replaces the bool-compare + mul pair per conv layer in the fused-conv
@@ -2789,4 +2793,90 @@ void gather_rows(torch::Tensor idx, std::vector<torch::Tensor> srcs,
     chunks *= 2;
   hipLaunchKernelGGL(gather_rows_kernel, dim3(k, a.ncols, chunks), dim3(256),
                      0, cur_stream(), a, idx.data_ptr<int64_t>(), k, chunks);
+}
+
+// ===========================================================================
+// Diagnostic: persistent-kernel step-cost decomposition (tools/
+// gpu_dgrad_bench.py). mode 0 = grid barrier only; 1 = + h burst-stage;
+// 2 = + MFMA phase over LDS; isolates what the ~8 us/step persistent LSTM
+// floor is made of.
+// ===========================================================================
+namespace {
+template <int H>
+__global__ __launch_bounds__(256) void lstm_diag_kernel(
+    const __bf16* __restrict__ h_bfs, const __bf16* __restrict__ w_bf,
+    float* __restrict__ sink, unsigned int* __restrict__ ctr, int B, int T,
+    int mode) {
+  constexpr int LROW = H + 32;
+  __shared__ __bf16 wlds[64][LROW];
+  __shared__ __bf16 hlds[32][LROW];
+  const int tid = threadIdx.x;
+  const int lane = tid & 63;
+  const int wave = tid >> 6;
+  const int s16 = blockIdx.x * 16;
+  const int kbase = (lane >> 4) * 8;
+  const unsigned nblocks = gridDim.x;
+  if (mode >= 2) {
+    for (int base = tid * 8; base < 64 * H; base += 256 * 8) {
+      const int r = base / H;
+      const int g = r >> 4, c = r & 15;
+      *reinterpret_cast<bf16x8_k5*>(&wlds[r][base - r * H]) =
+          *reinterpret_cast<const bf16x8_k5*>(
+              w_bf + ((int64_t)g * H + s16 + c) * H + (base - r * H));
+    }
+  }
+  __syncthreads();
+  f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
+  for (int t = 0; t < T; ++t) {
+    if (mode >= 1) {
+      const __bf16* h_bf = h_bfs + (int64_t)(t & 1) * B * H;
+      for (int base = tid * 8; base < 32 * H; base += 256 * 8) {
+        const int r = base / H;
+        bf16x8_k5 v{};
+        if (r < B)
+          v = *reinterpret_cast<const bf16x8_k5*>(h_bf + (int64_t)r * H +
+                                                  (base - r * H));
+        *reinterpret_cast<bf16x8_k5*>(&hlds[r][base - r * H]) = v;
+      }
+      __syncthreads();
+    }
+    if (mode >= 2) {
+#pragma unroll 4
+      for (int kb = 0; kb < H; kb += 32) {
+        bf16x8_k5 bfrag = *reinterpret_cast<const bf16x8_k5*>(
+            &wlds[wave * 16 + (lane & 15)][kb + kbase]);
+        bf16x8_k5 a0 = *reinterpret_cast<const bf16x8_k5*>(
+            &hlds[lane & 15][kb + kbase]);
+        bf16x8_k5 a1 = *reinterpret_cast<const bf16x8_k5*>(
+            &hlds[16 + (lane & 15)][kb + kbase]);
+        acc0 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a0, bfrag, acc0, 0, 0, 0);
+        acc1 = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a1, bfrag, acc1, 0, 0, 0);
+      }
+    }
+    __syncthreads();
+    if (tid == 0) {
+      __builtin_amdgcn_fence(__ATOMIC_RELEASE, "agent");
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      __hip_atomic_fetch_add((gu32_t*)ctr, 1u, DRL_RLX_AGENT);
+      const unsigned target = nblocks * (unsigned)(t + 1);
+      unsigned spins = 0;
+      while (__hip_atomic_load((gu32_t*)ctr, DRL_RLX_AGENT) < target) {
+        __builtin_amdgcn_s_sleep(4);
+        if (++spins > 5000000u) break;
+      }
+      __builtin_amdgcn_fence(__ATOMIC_ACQUIRE, "agent");
+    }
+    __syncthreads();
+  }
+  if (tid == 0 && acc0[0] > 1e30f) sink[blockIdx.x] = acc0[0] + acc1[0];
+}
+}  // namespace
+
+void lstm_diag(torch::Tensor h_bfs, torch::Tensor w_bf, torch::Tensor sink,
+               torch::Tensor ctr, int64_t B, int64_t T, int64_t mode) {
+  hipLaunchKernelGGL(lstm_diag_kernel<512>, dim3(32), dim3(256), 0,
+                     cur_stream(), (const __bf16*)h_bfs.data_ptr(),
+                     (const __bf16*)w_bf.data_ptr(), sink.data_ptr<float>(),
+                     (unsigned int*)ctr.data_ptr(), (int)B, (int)T,
+                     (int)mode);
 }

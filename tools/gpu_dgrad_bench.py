@@ -63,5 +63,24 @@ def main():
         print(f"lstm seq80 fwd+bwd persistent={flag}: {ms:.3f} ms")
 
 
-if __name__ == "__main__":
+if __name__ == "__main__" and os.environ.get("DRL_DIAG") != "1":
     main()
+
+
+def lstm_diag():
+    ext = hip_ext(required=True)
+    B, H, T = 32, 512, 80
+    h = torch.randn(2, B, H, device=DEV).to(torch.bfloat16)
+    w = torch.randn(4 * H, H, device=DEV).to(torch.bfloat16)
+    sink = torch.zeros(32, device=DEV)
+    for mode, name in [(0, "barrier-only"), (1, "+h-stage"), (2, "+mfma")]:
+        def run():
+            ctr = torch.zeros(1, dtype=torch.int32, device=DEV)
+            ext.lstm_diag(h, w, sink, ctr, B, T, mode)
+        ms = t_ms(run, iters=30, warm=5)
+        print(f"lstm_diag {name}: {ms*1e3:7.1f} us total, "
+              f"{ms*1e3/T:6.2f} us/step")
+
+
+if __name__ == "__main__" and os.environ.get("DRL_DIAG") == "1":
+    lstm_diag()
