@@ -1961,6 +1961,9 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
   __shared__ __bf16 wlds[64][LROW];
   __shared__ __bf16 hlds[32][LROW];
   __shared__ float gbuf[4][32][16];
+  __shared__ float xplds[4][32][17];  // xp slice (scattered 64 B runs in
+                                      // global; the per-lane scalar reads
+                                      // in the park phase measured ~3 us)
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wave = tid >> 6;
@@ -1992,14 +1995,31 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
   for (int t = 0; t < T; ++t) {
     const __bf16* h_bf = h_bfs + (int64_t)t * B * H;
     const float* xp_t = xp + (int64_t)t * B * 4 * H;
-    // burst-stage h_t (zero-pad rows >= B)
-    for (int base = tid * 8; base < 32 * H; base += 256 * 8) {
+    // burst-stage h_t (zero-pad rows >= B), 16 elems per thread round
+    for (int base = tid * 16; base < 32 * H; base += 256 * 16) {
       const int r = base / H;
       const int k = base - r * H;
-      bf16x8_k5 v{};
-      if (r < B)
-        v = *reinterpret_cast<const bf16x8_k5*>(h_bf + (int64_t)r * H + k);
-      *reinterpret_cast<bf16x8_k5*>(&hlds[r][k]) = v;
+      bf16x8_k5 v0{}, v1{};
+      if (r < B) {
+        v0 = *reinterpret_cast<const bf16x8_k5*>(h_bf + (int64_t)r * H + k);
+        v1 = *reinterpret_cast<const bf16x8_k5*>(h_bf + (int64_t)r * H + k + 8);
+      }
+      *reinterpret_cast<bf16x8_k5*>(&hlds[r][k]) = v0;
+      *reinterpret_cast<bf16x8_k5*>(&hlds[r][k + 8]) = v1;
+    }
+    // burst-stage the xp slice: thread t covers half of (gate, row) run
+    {
+      const int run = tid >> 1;  // gate*32 + row
+      const int g_ = run >> 5;
+      const int row = run & 31;
+      const int half = (tid & 1) * 8;
+      const int rowc = row < B ? row : 0;
+      const float* src =
+          xp_t + (int64_t)rowc * 4 * H + g_ * H + s16 + half;
+      float4 lo = *reinterpret_cast<const float4*>(src);
+      float4 hi = *reinterpret_cast<const float4*>(src + 4);
+      *reinterpret_cast<float4*>(&xplds[g_][row][half]) = lo;
+      *reinterpret_cast<float4*>(&xplds[g_][row][half + 4]) = hi;
     }
     __syncthreads();
     f32x4 acc0 = {0.f, 0.f, 0.f, 0.f}, acc1 = {0.f, 0.f, 0.f, 0.f};
@@ -2017,12 +2037,11 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
     const int crow = (lane >> 4) * 4;
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int row0 = crow + r, row1 = 16 + crow + r;
-      int row0c = row0 < B ? row0 : 0, row1c = row1 < B ? row1 : 0;
+      const int row0 = crow + r, row1 = 16 + crow + r;
       gbuf[wave][row0][lane & 15] =
-          acc0[r] + xp_t[(int64_t)row0c * 4 * H + col];
+          acc0[r] + xplds[wave][row0][lane & 15];
       gbuf[wave][row1][lane & 15] =
-          acc1[r] + xp_t[(int64_t)row1c * 4 * H + col];
+          acc1[r] + xplds[wave][row1][lane & 15];
     }
     __syncthreads();
     float* h_out = hs + (int64_t)(t + 1) * B * H;
@@ -2127,14 +2146,17 @@ __global__ __launch_bounds__(256) void lstm_seq_bwd_bf16_kernel(
 #pragma unroll
       for (int ph = 0; ph < 2; ++ph) {
         __syncthreads();  // dglds reuse: prior phase reads must finish
-        for (int base = tid * 8; base < 32 * 2 * H; base += 256 * 8) {
+        for (int base = tid * 16; base < 32 * 2 * H; base += 256 * 16) {
           const int r = base / (2 * H);
           const int k = base - r * 2 * H;
-          bf16x8_k5 v{};
-          if (r < B)
-            v = *reinterpret_cast<const bf16x8_k5*>(
-                dg_prev + (int64_t)r * 4 * H + ph * 2 * H + k);
-          *reinterpret_cast<bf16x8_k5*>(&dglds[r][k]) = v;
+          bf16x8_k5 v0{}, v1{};
+          if (r < B) {
+            const __bf16* sp = dg_prev + (int64_t)r * 4 * H + ph * 2 * H + k;
+            v0 = *reinterpret_cast<const bf16x8_k5*>(sp);
+            v1 = *reinterpret_cast<const bf16x8_k5*>(sp + 8);
+          }
+          *reinterpret_cast<bf16x8_k5*>(&dglds[r][k]) = v0;
+          *reinterpret_cast<bf16x8_k5*>(&dglds[r][k + 8]) = v1;
         }
         __syncthreads();
         const int kw0 = wave * (H / 2);  // this wave's quarter of the phase
