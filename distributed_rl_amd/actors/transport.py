@@ -346,6 +346,19 @@ class TransportSession:
         )
         self.reward_ring(actor_idx).push_records(rec)
 
+    def drain_rewards_with_eps(self) -> List[tuple]:
+        """[(episode_reward, actor_eps)] — eps rides each record so the
+        learner can report the near-greedy mean (the reference's Reward
+        scalar gates on eps < 0.05, APE_X/Player.py:272-277)."""
+        out: List[tuple] = []
+        for i in range(self.num_rings):
+            r = self.reward_ring(i).pop_records()
+            if r is not None:
+                rec = r.view(self.reward_codec.np_dtype).reshape(-1)
+                out.extend((float(a), float(b))
+                           for a, b in zip(rec["reward"], rec["eps"]))
+        return out
+
     def drain_rewards(self) -> List[float]:
         out: List[float] = []
         for i in range(self.num_rings):
@@ -441,6 +454,9 @@ class LearnerEndpoint:
 
     def drain_rewards(self):
         return self.session.drain_rewards()
+
+    def drain_rewards_with_eps(self):
+        return self.session.drain_rewards_with_eps()
 
     def total_drops(self) -> int:
         """Backpressure telemetry: rows the actors pushed that the rings

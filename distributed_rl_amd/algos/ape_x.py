@@ -748,17 +748,27 @@ class ApexLearner(LearnerBase):
         return last_loss
 
     def _log_block(self, stats):
-        rewards = self.transport.drain_rewards() if self.transport else []
-        # no fabricated placeholder when nothing was drained (the reference
-        # writes -21.0, a Pong-specific constant — APE_X/Learner.py:231);
-        # skip the scalar instead, like the IMPALA learner does
+        # eps rides the reward records: report both the fleet mean and the
+        # near-greedy mean (the reference's Reward gates on eps < 0.05,
+        # APE_X/Player.py:272-277). No fabricated -21 placeholder when
+        # nothing was drained.
+        pairs = (self.transport.drain_rewards_with_eps()
+                 if self.transport is not None
+                 and hasattr(self.transport, "drain_rewards_with_eps")
+                 else [(r, 0.0) for r in (self.transport.drain_rewards()
+                                          if self.transport else [])])
+        rewards = [r for r, _ in pairs]
+        greedy = [r for r, e in pairs if e < 0.05]
         mean_r = float(np.mean(rewards)) if rewards else float("nan")
+        mean_g = float(np.mean(greedy)) if greedy else float("nan")
         timing = self.flush_timing()
         loss = float(stats["loss"])
         value = float(stats["value"])
         norm = float(self.model.calculateNorm())
+        if greedy:
+            self.log_scalar("Reward", mean_g)  # reference semantics
         if rewards:
-            self.log_scalar("Reward", mean_r)
+            self.log_scalar("Reward_all", mean_r)
         self.log_scalar("value", value)
         self.log_scalar("norm", norm)
         self.log_scalar("loss", loss)
@@ -766,7 +776,8 @@ class ApexLearner(LearnerBase):
             sps = self.LOG_EVERY / max(timing["wall"], 1e-9)
             print(
                 f"[APE_X] step={self.step_count} loss={loss:.5f} value={value:.3f} "
-                f"norm={norm:.2f} reward={mean_r:.1f} replay={len(self.replay)} "
+                f"norm={norm:.2f} reward={mean_r:.1f} "
+                f"greedy_reward={mean_g:.1f} replay={len(self.replay)} "
                 f"steps/s={sps:.1f} "
                 + " ".join(f"{k}={v:.2f}s" for k, v in timing.items()),
                 flush=True,
