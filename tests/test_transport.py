@@ -136,3 +136,50 @@ def test_learner_endpoint_ring_partitioning(tmp_path):
         assert sorted(c1["action"].tolist()) == [1, 3]
     finally:
         sess.close()
+
+
+def test_pack_rows_matches_numpy_unpack():
+    """C++ ingest pack (AoS shm records -> SoA staging) vs the numpy
+    structured-dtype reference — host code, runs on CPU."""
+    import numpy as np
+    import torch
+
+    from distributed_rl_amd.ops import hip_ext
+    from distributed_rl_amd.replay import make_apex_schema
+
+    ext = hip_ext(required=False)
+    if ext is None or not hasattr(ext, "pack_rows"):
+        import pytest
+
+        pytest.skip("extension not built")
+    from distributed_rl_amd.actors.transport import RecordCodec
+
+    codec = RecordCodec(make_apex_schema(), with_priority=True)
+    rng = np.random.default_rng(3)
+    n = 37
+    cols = {
+        "state": rng.integers(0, 255, (n, 4, 84, 84), dtype=np.uint8),
+        "action": rng.integers(0, 6, (n,), dtype=np.int32),
+        "reward": rng.random(n, dtype=np.float32),
+        "next_state": rng.integers(0, 255, (n, 4, 84, 84), dtype=np.uint8),
+        "done": rng.random(n, dtype=np.float32),
+    }
+    prio = rng.random(n, dtype=np.float32)
+    rec = codec.pack(cols, prio)
+    raw = torch.from_numpy(np.ascontiguousarray(rec).view(np.uint8).reshape(-1))
+
+    names = list(codec.schema) + ["priority"]
+    f = codec.np_dtype.fields
+    offs = [int(f[k][1]) for k in names]
+    sizes = [int(f[k][0].itemsize) for k in names]
+    dsts = [torch.zeros((64, *codec.schema[k][0]),
+                        dtype=dict(codec.schema)[k][1])
+            for k in codec.schema] + [torch.zeros(64)]
+    ext.pack_rows(raw, codec.record_size, offs, sizes, dsts, 5)
+
+    ref_cols, ref_prio = codec.unpack(rec)
+    for i, k in enumerate(codec.schema):
+        got = dsts[i][5 : 5 + n].numpy()
+        assert np.array_equal(got.reshape(ref_cols[k].shape), ref_cols[k]), k
+        assert not dsts[i][:5].any() and not dsts[i][5 + n :].any()
+    assert np.array_equal(dsts[-1][5 : 5 + n].numpy(), ref_prio)
