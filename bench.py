@@ -176,7 +176,8 @@ def main():
     torch.set_num_threads(min(16, os.cpu_count() or 16))
     rank, local_rank, world = init_distributed()
     has_cuda = torch.cuda.is_available()
-    device = f"cuda:{local_rank}" if has_cuda else "cpu"
+    dev_idx = local_rank % torch.cuda.device_count() if has_cuda else 0
+    device = f"cuda:{dev_idx}" if has_cuda else "cpu"
     if world == 1 and args.gpus > 1:
         # N>1 must come through torchrun (one rank per GPU); a single process
         # only ever drives one GPU — never multiply frames by an unused N.

@@ -32,9 +32,13 @@ def init_distributed(backend: Optional[str] = None) -> tuple[int, int, int]:
     local_rank = int(os.environ.get("LOCAL_RANK", "0"))
     if world > 1 and not dist.is_initialized():
         if backend is None:
-            backend = "nccl" if torch.cuda.is_available() else "gloo"
+            # DRL_DIST_BACKEND=gloo lets a multi-rank rehearsal share one
+            # GPU (RCCL refuses duplicate devices); default RCCL
+            backend = os.environ.get(
+                "DRL_DIST_BACKEND",
+                "nccl" if torch.cuda.is_available() else "gloo")
         if torch.cuda.is_available():
-            torch.cuda.set_device(local_rank)
+            torch.cuda.set_device(local_rank % torch.cuda.device_count())
         dist.init_process_group(backend=backend)
     return rank, local_rank, world
 
