@@ -634,7 +634,7 @@ void conv_wrw(torch::Tensor in, torch::Tensor gout, torch::Tensor gw_ws,
   // scale the m-grid with the batch: R2D2/IMPALA run 8-20x Ape-X's row
   // count through the same kernel (conv1 wrw measured 139 us at 1.02M
   // rows with 512 blocks = 250 serial chunks per block)
-  int mb = (n_chunks > 8192 ? 4096 : 512) / L->ktiles;
+  int mb = (n_chunks > 8192 ? 2048 : 512) / L->ktiles;  // 4096 A/B'd worse (atomics)
   if (mb > n_chunks) mb = n_chunks;
   if (mb < 1) mb = 1;
   hipLaunchKernelGGL(L->fn, dim3(L->ktiles, mb), dim3(256), L->lds_bytes,
