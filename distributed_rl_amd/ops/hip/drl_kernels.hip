@@ -1968,7 +1968,6 @@ __global__ __launch_bounds__(256) void lstm_seq_fwd_bf16_kernel(
   const int lane = tid & 63;
   const int wave = tid >> 6;
   const int s16 = blockIdx.x * 16;
-  const int col = wave * H + s16 + (lane & 15);  // gate column in 4H
   const int kbase = (lane >> 4) * 8;
   const unsigned nblocks = gridDim.x;
 
