@@ -166,6 +166,9 @@ def main():
                     help="spawn K synthetic-env CPU actor processes feeding "
                          "the replay DURING the timed region (BASELINE "
                          "config-2 whole-node mode); 0 = learner-only")
+    ap.add_argument("--envs-per-proc", type=int, default=1,
+                    help="virtual actors per fleet process (batched shared-"
+                         "model inference; Ape-X only)")
     ap.add_argument("--burn-in", type=int, default=None,
                     help="R2D2 burn-in override (BASELINE config 4 = 40)")
     ap.add_argument("--transport-dir", default=None)
@@ -222,7 +225,8 @@ def main():
         args._transport = LearnerEndpoint(session, rank=rank, world_size=world)
         if rank == 0:
             fleet = ActorFleet(args.cfg, args.with_actors, tdir,
-                               env_kind="synthetic", respawn_on_exit=False)
+                               env_kind="synthetic", respawn_on_exit=False,
+                               envs_per_proc=args.envs_per_proc)
 
     builders = {"APE_X": build_apex, "IMPALA": build_impala, "R2D2": build_r2d2}
     learner, frames_per_step = builders[cfg.alg](cfg, device, rank, world, args)
@@ -246,7 +250,7 @@ def main():
         dt_warm = time.perf_counter() - t_warm
         print(f"# fleet warm: {learner.ingested_total} rows in {dt_warm:.1f}s "
               f"({learner.ingested_total / max(dt_warm, 1e-9):.0f} rows/s), "
-              f"{fleet.alive_count()}/{args.with_actors} actors alive",
+              f"{fleet.alive_count()}/{fleet.num_procs} actor procs alive",
               file=sys.stderr)
         learner.stop_ingest_thread()  # capture below needs quiet streams
 

@@ -37,6 +37,17 @@ def get_player_cls(alg: str):
     raise ValueError(f"unknown algorithm {alg}")
 
 
+def get_vec_runner(alg: str):
+    """Vectorized multi-env actor loop for `alg`, or None if the algorithm
+    only supports one env per actor process (fleet envs_per_proc > 1
+    requires this)."""
+    if alg.upper() == "APE_X":
+        from .ape_x import run_apex_vec
+
+        return run_apex_vec
+    return None
+
+
 def get_wire_schema(cfg: Config) -> Tuple[Dict, bool]:
     """(schema, with_priority) for the transport record codec."""
     alg = cfg.alg

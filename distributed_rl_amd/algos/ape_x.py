@@ -1004,3 +1004,54 @@ class ApexPlayer:
                 self.local.clear()
                 state = self.env.reset()
         self._flush()
+
+
+def run_apex_vec(players: List["ApexPlayer"], max_env_steps: int = 1_000_000):
+    """Vectorized multi-env actor loop: M ApexPlayers in ONE process share
+    the lead player's model/target, and every env step all M eps-greedy
+    argmaxes come from a single batched forward instead of M single-frame
+    forwards (the reference pins one Ray process per env,
+    run_actor.py:46-55; at 256 actors that is host-process bound).
+
+    Per-virtual-actor semantics are preserved exactly: each player keeps
+    its own env, eps ladder slot, rng stream, n-step LocalBuffer,
+    transport ring, and reward telemetry; only the weights and the
+    argmax batch are shared. Weight pulls / the bounded-staleness gate
+    run on the lead player (one shared model = one version for all M)."""
+    assert players, "empty player list"
+    lead = players[0]
+    for p in players[1:]:
+        p.model = lead.model
+        p.target = lead.target
+    lead.pull_weights()
+    states = [p.env.reset() for p in players]
+    ep_rew = [0.0] * len(players)
+    while lead.env_steps < max_env_steps:
+        x = torch.from_numpy(np.stack(states)).float().div_(255.0)
+        with torch.no_grad():
+            greedy = lead.model.forward([x])[0].argmax(1).tolist()
+        for j, p in enumerate(players):
+            if p.rng.random() < p.eps:
+                action = int(p.rng.integers(0, p.action_n))
+            else:
+                action = int(greedy[j])
+            next_state, reward, done, info = p.env.step(action)
+            ep_rew[j] += reward
+            p.local.append(states[j], action, reward)
+            p.pending.extend(p.local.emit_ready(
+                next_state, done or info.get("pseudo_done", False)))
+            if len(p.pending) >= p.PUSH_BATCH or done:
+                p._flush()
+            states[j] = next_state
+            p.env_steps += 1
+            if done:
+                if p.eps < 0.05 or p.report_all_rewards:
+                    p.transport.push_reward(p.idx, ep_rew[j], p.eps)
+                ep_rew[j] = 0.0
+                p.local.clear()
+                states[j] = p.env.reset()
+        if lead.env_steps % ACTOR_PULL_EVERY == 0:
+            lead.pull_weights()
+        lead._staleness_gate()
+    for p in players:
+        p._flush()

@@ -32,6 +32,9 @@ def main():
                     help="auto | atari | synthetic")
     ap.add_argument("--max-env-steps", type=int, default=1 << 60)
     ap.add_argument("--no-respawn", action="store_true")
+    ap.add_argument("--envs-per-proc", type=int, default=1,
+                    help="virtual actors per OS process (batched shared-"
+                         "model inference; Ape-X only)")
     ap.add_argument("--tcp", default="",
                     help="learner host:port for TCP mode (multi-host)")
     args = ap.parse_args()
@@ -51,7 +54,8 @@ def main():
     fleet = ActorFleet(args.cfg or cfg.alg.lower(), n, tdir,
                        start_idx=args.start_idx, env_kind=args.env,
                        max_env_steps=args.max_env_steps,
-                       respawn_on_exit=not args.no_respawn, tcp=args.tcp)
+                       respawn_on_exit=not args.no_respawn, tcp=args.tcp,
+                       envs_per_proc=args.envs_per_proc)
     fleet.start()
     print(f"[run_actor] {n} actors running (start_idx={args.start_idx})",
           flush=True)
