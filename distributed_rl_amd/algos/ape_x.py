@@ -938,7 +938,9 @@ class ApexPlayer:
             return
         trans = self.pending
         self.pending = []
-        prio = self._priorities(trans)
+        self._push_trans(trans, self._priorities(trans))
+
+    def _push_trans(self, trans: List[tuple], prio: np.ndarray):
         cols = {
             "state": np.stack([t[0] for t in trans]),
             "action": np.array([t[1] for t in trans], np.int32),
@@ -1017,7 +1019,14 @@ def run_apex_vec(players: List["ApexPlayer"], max_env_steps: int = 1_000_000):
     its own env, eps ladder slot, rng stream, n-step LocalBuffer,
     transport ring, and reward telemetry; only the weights and the
     argmax batch are shared. Weight pulls / the bounded-staleness gate
-    run on the lead player (one shared model = one version for all M)."""
+    run on the lead player (one shared model = one version for all M).
+
+    Priority forwards are ALSO cross-player batched: players whose
+    pending buffer is flush-ready after a step are flushed together with
+    one _priorities pass over the concatenation (3 forwards of M*16
+    instead of 3M forwards of 16 — the per-player flushes were the
+    dominant serial cost in the first A/B, profiles/r02_vec_actor.md),
+    then split back to each player's own ring."""
     assert players, "empty player list"
     lead = players[0]
     for p in players[1:]:
@@ -1026,10 +1035,25 @@ def run_apex_vec(players: List["ApexPlayer"], max_env_steps: int = 1_000_000):
     lead.pull_weights()
     states = [p.env.reset() for p in players]
     ep_rew = [0.0] * len(players)
+
+    def _batched_flush(ready: List["ApexPlayer"]):
+        counts = []
+        all_trans: List[tuple] = []
+        for p in ready:
+            counts.append(len(p.pending))
+            all_trans.extend(p.pending)
+            p.pending = []
+        prio = lead._priorities(all_trans)
+        off = 0
+        for p, n in zip(ready, counts):
+            p._push_trans(all_trans[off : off + n], prio[off : off + n])
+            off += n
+
     while lead.env_steps < max_env_steps:
         x = torch.from_numpy(np.stack(states)).float().div_(255.0)
         with torch.no_grad():
             greedy = lead.model.forward([x])[0].argmax(1).tolist()
+        flush_ready = []
         for j, p in enumerate(players):
             if p.rng.random() < p.eps:
                 action = int(p.rng.integers(0, p.action_n))
@@ -1040,8 +1064,8 @@ def run_apex_vec(players: List["ApexPlayer"], max_env_steps: int = 1_000_000):
             p.local.append(states[j], action, reward)
             p.pending.extend(p.local.emit_ready(
                 next_state, done or info.get("pseudo_done", False)))
-            if len(p.pending) >= p.PUSH_BATCH or done:
-                p._flush()
+            if p.pending and (len(p.pending) >= p.PUSH_BATCH or done):
+                flush_ready.append(p)
             states[j] = next_state
             p.env_steps += 1
             if done:
@@ -1050,6 +1074,8 @@ def run_apex_vec(players: List["ApexPlayer"], max_env_steps: int = 1_000_000):
                 ep_rew[j] = 0.0
                 p.local.clear()
                 states[j] = p.env.reset()
+        if flush_ready:
+            _batched_flush(flush_ready)
         if lead.env_steps % ACTOR_PULL_EVERY == 0:
             lead.pull_weights()
         lead._staleness_gate()

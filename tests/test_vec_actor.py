@@ -131,12 +131,17 @@ def test_vec_fleet_process_integration(tmp_path):
         fleet.start()
         got = 0
         t0 = time.time()
-        while got < 64 and time.time() - t0 < 120:
+        # wait for BOTH processes: every ring must have pushed (the second
+        # proc can deliver 64 rows before the first finishes importing
+        # torch, so gating on `got` alone races process startup)
+        while time.time() - t0 < 150:
             got += learner.ingest()
+            pushed = [session.ring(i).head for i in range(4)]
+            if got >= 64 and all(n > 0 for n in pushed):
+                break
             time.sleep(0.05)
         assert got >= 64, f"only {got} transitions arrived"
-        pushed = [session.ring(i).head for i in range(4)]
-        assert sum(1 for n in pushed if n > 0) == 4, pushed
+        assert all(n > 0 for n in pushed), pushed
         for _ in range(3):
             stats = learner.step()
         assert float(stats["loss"]) == float(stats["loss"])  # finite
